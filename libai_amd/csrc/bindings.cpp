@@ -1,0 +1,321 @@
+// Torch bindings for the libai_amd gfx950 kernel library.
+//
+// Compiled by g++ (torch headers only); the kernels themselves live in
+// csrc/kernels/*.hip, compiled by hipcc for gfx950 and linked in.  The split
+// keeps torch headers out of hipcc (seconds-per-kernel compiles) and keeps the
+// device code pure HIP.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime_api.h>
+
+#include <tuple>
+#include <vector>
+
+namespace {
+
+hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+void check_launch(const char* name) {
+  hipError_t err = hipGetLastError();
+  TORCH_CHECK(err == hipSuccess, name, " launch failed: ", hipGetErrorString(err));
+}
+
+bool is_bf16(const torch::Tensor& t) {
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16 || t.scalar_type() == torch::kFloat32,
+              "libai_amd kernels support bf16/f32, got ", t.scalar_type());
+  return t.scalar_type() == torch::kBFloat16;
+}
+
+#define CHECK_IN(t)                                               \
+  TORCH_CHECK(t.is_cuda(), #t " must be on device");              \
+  TORCH_CHECK(t.is_contiguous(), #t " must be contiguous");
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// extern "C" launchers from csrc/kernels/*.hip
+// ---------------------------------------------------------------------------
+extern "C" {
+void ln_fwd_bf16(const void*, const void*, const void*, void*, float*, float*, int64_t,
+                 int, float, bool, hipStream_t);
+void ln_fwd_f32(const void*, const void*, const void*, void*, float*, float*, int64_t,
+                int, float, bool, hipStream_t);
+void ln_bwd_dx_bf16(const void*, const void*, const void*, const float*, const float*,
+                    void*, int64_t, int, bool, hipStream_t);
+void ln_bwd_dx_f32(const void*, const void*, const void*, const float*, const float*,
+                   void*, int64_t, int, bool, hipStream_t);
+void ln_bwd_wgrad_bf16(const void*, const void*, const float*, const float*, float*,
+                       float*, void*, void*, int64_t, int, int, bool, hipStream_t);
+void ln_bwd_wgrad_f32(const void*, const void*, const float*, const float*, float*,
+                      float*, void*, void*, int64_t, int, int, bool, hipStream_t);
+void bias_gelu_fwd_bf16(const void*, const void*, void*, int64_t, int, hipStream_t);
+void bias_gelu_fwd_f32(const void*, const void*, void*, int64_t, int, hipStream_t);
+void bias_gelu_bwd_bf16(const void*, const void*, const void*, void*, int64_t, int,
+                        hipStream_t);
+void bias_gelu_bwd_f32(const void*, const void*, const void*, void*, int64_t, int,
+                       hipStream_t);
+void bias_dropout_res_fwd_bf16(const void*, const void*, const void*, void*, int64_t,
+                               int, float, uint64_t, hipStream_t);
+void bias_dropout_res_fwd_f32(const void*, const void*, const void*, void*, int64_t,
+                              int, float, uint64_t, hipStream_t);
+void bias_dropout_res_bwd_bf16(const void*, void*, int64_t, int, float, uint64_t,
+                               hipStream_t);
+void bias_dropout_res_bwd_f32(const void*, void*, int64_t, int, float, uint64_t,
+                              hipStream_t);
+void colsum_bf16(const void*, float*, void*, int64_t, int, int, hipStream_t);
+void colsum_f32(const void*, float*, void*, int64_t, int, int, hipStream_t);
+void softmax_fwd_bf16(const void*, const uint8_t*, void*, int64_t, int, int, int, float,
+                      float, uint64_t, bool, hipStream_t);
+void softmax_fwd_f32(const void*, const uint8_t*, void*, int64_t, int, int, int, float,
+                     float, uint64_t, bool, hipStream_t);
+void softmax_bwd_bf16(const void*, const void*, const uint8_t*, void*, int64_t, int,
+                      int, int, float, float, uint64_t, bool, hipStream_t);
+void softmax_bwd_f32(const void*, const void*, const uint8_t*, void*, int64_t, int, int,
+                     int, float, float, uint64_t, bool, hipStream_t);
+void ce_fwd_bf16(const void*, const int64_t*, float*, float*, float*, int64_t, int64_t,
+                 int64_t, int64_t, hipStream_t);
+void ce_fwd_f32(const void*, const int64_t*, float*, float*, float*, int64_t, int64_t,
+                int64_t, int64_t, hipStream_t);
+void ce_bwd_bf16(const void*, const int64_t*, const float*, const float*, const float*,
+                 void*, int64_t, int64_t, int64_t, int64_t, hipStream_t);
+void ce_bwd_f32(const void*, const int64_t*, const float*, const float*, const float*,
+                void*, int64_t, int64_t, int64_t, int64_t, hipStream_t);
+int adamw_chunk_elems();
+void adamw_step_bf16(const void*, int, float, float, float, float, float, float, float,
+                     float, hipStream_t);
+void adamw_step_f32(const void*, int, float, float, float, float, float, float, float,
+                    float, hipStream_t);
+void l2norm_sq_bf16(const void*, int, float*, hipStream_t);
+void l2norm_sq_f32(const void*, int, float*, hipStream_t);
+}
+
+// ---------------------------------------------------------------------------
+// LayerNorm / RMSNorm
+// ---------------------------------------------------------------------------
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ln_fwd(
+    torch::Tensor x, torch::Tensor gamma, c10::optional<torch::Tensor> beta, bool rms,
+    double eps) {
+  CHECK_IN(x);
+  CHECK_IN(gamma);
+  const int H = (int)gamma.numel();
+  const int64_t R = x.numel() / H;
+  auto y = torch::empty_like(x);
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto mean = rms ? torch::empty({0}, opts) : torch::empty({R}, opts);
+  auto rstd = torch::empty({R}, opts);
+  const void* bptr = nullptr;
+  if (beta.has_value()) {
+    CHECK_IN(beta.value());
+    bptr = beta->data_ptr();
+  }
+  auto fn = is_bf16(x) ? ln_fwd_bf16 : ln_fwd_f32;
+  fn(x.data_ptr(), gamma.data_ptr(), bptr, y.data_ptr(),
+     rms ? nullptr : mean.data_ptr<float>(), rstd.data_ptr<float>(), R, H, (float)eps,
+     rms, cur_stream());
+  check_launch("ln_fwd");
+  return {y, mean, rstd};
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ln_bwd(
+    torch::Tensor dy, torch::Tensor x, torch::Tensor gamma, torch::Tensor mean,
+    torch::Tensor rstd, bool rms, bool need_dbeta) {
+  CHECK_IN(dy);
+  CHECK_IN(x);
+  const int H = (int)gamma.numel();
+  const int64_t R = x.numel() / H;
+  auto dx = torch::empty_like(x);
+  auto dgamma = torch::empty_like(gamma);
+  auto dbeta = need_dbeta ? torch::empty_like(gamma) : torch::Tensor();
+  const int P = (int)std::min<int64_t>(256, std::max<int64_t>(1, R / 4));
+  auto partial = torch::empty({(need_dbeta ? 2L : 1L) * P, (int64_t)H},
+                              x.options().dtype(torch::kFloat32));
+  float* pgamma = partial.data_ptr<float>();
+  float* pbeta = need_dbeta ? pgamma + (int64_t)P * H : nullptr;
+  auto fdx = is_bf16(x) ? ln_bwd_dx_bf16 : ln_bwd_dx_f32;
+  auto fw = is_bf16(x) ? ln_bwd_wgrad_bf16 : ln_bwd_wgrad_f32;
+  const float* meanp = rms ? nullptr : mean.data_ptr<float>();
+  fdx(dy.data_ptr(), x.data_ptr(), gamma.data_ptr(), meanp, rstd.data_ptr<float>(),
+      dx.data_ptr(), R, H, rms, cur_stream());
+  fw(dy.data_ptr(), x.data_ptr(), meanp, rstd.data_ptr<float>(), pgamma, pbeta,
+     dgamma.data_ptr(), need_dbeta ? dbeta.data_ptr() : nullptr, R, H, P, rms,
+     cur_stream());
+  check_launch("ln_bwd");
+  return {dx, dgamma, dbeta};
+}
+
+// ---------------------------------------------------------------------------
+// fused bias ops
+// ---------------------------------------------------------------------------
+torch::Tensor bias_gelu_fwd(torch::Tensor x, c10::optional<torch::Tensor> b) {
+  CHECK_IN(x);
+  auto y = torch::empty_like(x);
+  const int W = b.has_value() ? (int)b->numel() : (int)x.size(-1);
+  auto fn = is_bf16(x) ? bias_gelu_fwd_bf16 : bias_gelu_fwd_f32;
+  fn(x.data_ptr(), b.has_value() ? b->data_ptr() : nullptr, y.data_ptr(), x.numel(), W,
+     cur_stream());
+  check_launch("bias_gelu_fwd");
+  return y;
+}
+
+torch::Tensor bias_gelu_bwd(torch::Tensor x, c10::optional<torch::Tensor> b,
+                            torch::Tensor dy) {
+  CHECK_IN(x);
+  CHECK_IN(dy);
+  auto dx = torch::empty_like(x);
+  const int W = b.has_value() ? (int)b->numel() : (int)x.size(-1);
+  auto fn = is_bf16(x) ? bias_gelu_bwd_bf16 : bias_gelu_bwd_f32;
+  fn(x.data_ptr(), b.has_value() ? b->data_ptr() : nullptr, dy.data_ptr(), dx.data_ptr(),
+     x.numel(), W, cur_stream());
+  check_launch("bias_gelu_bwd");
+  return dx;
+}
+
+torch::Tensor colsum(torch::Tensor x, int64_t W) {
+  CHECK_IN(x);
+  const int64_t R = x.numel() / W;
+  const int P = (int)std::min<int64_t>(256, std::max<int64_t>(1, R / 4));
+  auto partial = torch::empty({P, W}, x.options().dtype(torch::kFloat32));
+  auto out = torch::empty({W}, x.options());
+  auto fn = is_bf16(x) ? colsum_bf16 : colsum_f32;
+  fn(x.data_ptr(), partial.data_ptr<float>(), out.data_ptr(), R, (int)W, P,
+     cur_stream());
+  check_launch("colsum");
+  return out;
+}
+
+torch::Tensor bias_dropout_res_fwd(torch::Tensor x, c10::optional<torch::Tensor> b,
+                                   c10::optional<torch::Tensor> res, double p,
+                                   int64_t seed) {
+  CHECK_IN(x);
+  auto y = torch::empty_like(x);
+  const int W = b.has_value() ? (int)b->numel() : (int)x.size(-1);
+  auto fn = is_bf16(x) ? bias_dropout_res_fwd_bf16 : bias_dropout_res_fwd_f32;
+  fn(x.data_ptr(), b.has_value() ? b->data_ptr() : nullptr,
+     res.has_value() ? res->data_ptr() : nullptr, y.data_ptr(), x.numel(), W, (float)p,
+     (uint64_t)seed, cur_stream());
+  check_launch("bias_dropout_res_fwd");
+  return y;
+}
+
+torch::Tensor bias_dropout_res_bwd(torch::Tensor dy, double p, int64_t seed) {
+  CHECK_IN(dy);
+  auto dx = torch::empty_like(dy);
+  auto fn = is_bf16(dy) ? bias_dropout_res_bwd_bf16 : bias_dropout_res_bwd_f32;
+  fn(dy.data_ptr(), dx.data_ptr(), dy.numel(), (int)dy.size(-1), (float)p,
+     (uint64_t)seed, cur_stream());
+  check_launch("bias_dropout_res_bwd");
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
+// fused scale+mask+softmax(+dropout)
+// ---------------------------------------------------------------------------
+torch::Tensor softmax_fwd(torch::Tensor s, c10::optional<torch::Tensor> pad_mask,
+                          double scale, double p, int64_t seed, bool causal) {
+  CHECK_IN(s);
+  TORCH_CHECK(s.dim() == 4, "softmax_fwd expects [B, NH, SQ, SK]");
+  auto out = torch::empty_like(s);
+  const uint8_t* mp = nullptr;
+  if (pad_mask.has_value()) {
+    CHECK_IN(pad_mask.value());
+    TORCH_CHECK(pad_mask->scalar_type() == torch::kUInt8 ||
+                pad_mask->scalar_type() == torch::kBool);
+    mp = (const uint8_t*)pad_mask->data_ptr();
+  }
+  auto fn = is_bf16(s) ? softmax_fwd_bf16 : softmax_fwd_f32;
+  fn(s.data_ptr(), mp, out.data_ptr(), s.size(0), (int)s.size(1), (int)s.size(2),
+     (int)s.size(3), (float)scale, (float)p, (uint64_t)seed, causal, cur_stream());
+  check_launch("softmax_fwd");
+  return out;
+}
+
+torch::Tensor softmax_bwd(torch::Tensor s, torch::Tensor dout,
+                          c10::optional<torch::Tensor> pad_mask, double scale, double p,
+                          int64_t seed, bool causal) {
+  CHECK_IN(s);
+  CHECK_IN(dout);
+  auto ds = torch::empty_like(s);
+  const uint8_t* mp = nullptr;
+  if (pad_mask.has_value()) mp = (const uint8_t*)pad_mask->data_ptr();
+  auto fn = is_bf16(s) ? softmax_bwd_bf16 : softmax_bwd_f32;
+  fn(s.data_ptr(), dout.data_ptr(), mp, ds.data_ptr(), s.size(0), (int)s.size(1),
+     (int)s.size(2), (int)s.size(3), (float)scale, (float)p, (uint64_t)seed, causal,
+     cur_stream());
+  check_launch("softmax_bwd");
+  return ds;
+}
+
+// ---------------------------------------------------------------------------
+// vocab-parallel cross entropy
+// ---------------------------------------------------------------------------
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ce_fwd(torch::Tensor logits,
+                                                               torch::Tensor targets,
+                                                               int64_t vocab_start,
+                                                               int64_t ignore_index) {
+  CHECK_IN(logits);
+  CHECK_IN(targets);
+  TORCH_CHECK(logits.dim() == 2, "ce_fwd expects [R, V_local]");
+  const int64_t R = logits.size(0), Vl = logits.size(1);
+  auto opts = logits.options().dtype(torch::kFloat32);
+  auto lmax = torch::empty({R}, opts);
+  auto lsumexp = torch::empty({R}, opts);
+  auto tlogit = torch::empty({R}, opts);
+  auto fn = is_bf16(logits) ? ce_fwd_bf16 : ce_fwd_f32;
+  fn(logits.data_ptr(), targets.data_ptr<int64_t>(), lmax.data_ptr<float>(),
+     lsumexp.data_ptr<float>(), tlogit.data_ptr<float>(), R, Vl, vocab_start,
+     ignore_index, cur_stream());
+  check_launch("ce_fwd");
+  return {lmax, lsumexp, tlogit};
+}
+
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets, torch::Tensor gmax,
+                     torch::Tensor gsumexp, torch::Tensor gscale, int64_t vocab_start,
+                     int64_t ignore_index) {
+  CHECK_IN(logits);
+  auto dlogits = torch::empty_like(logits);
+  auto fn = is_bf16(logits) ? ce_bwd_bf16 : ce_bwd_f32;
+  fn(logits.data_ptr(), targets.data_ptr<int64_t>(), gmax.data_ptr<float>(),
+     gsumexp.data_ptr<float>(), gscale.data_ptr<float>(), dlogits.data_ptr(),
+     logits.size(0), logits.size(1), vocab_start, ignore_index, cur_stream());
+  check_launch("ce_bwd");
+  return dlogits;
+}
+
+// ---------------------------------------------------------------------------
+// fused AdamW
+// ---------------------------------------------------------------------------
+void adamw_step(torch::Tensor chunks, bool bf16_params, double lr, double beta1,
+                double beta2, double eps, double wd, double bc1, double bc2,
+                double grad_scale) {
+  CHECK_IN(chunks);
+  TORCH_CHECK(chunks.scalar_type() == torch::kInt64 && chunks.size(1) == 6);
+  auto fn = bf16_params ? adamw_step_bf16 : adamw_step_f32;
+  fn(chunks.data_ptr(), (int)chunks.size(0), (float)lr, (float)beta1, (float)beta2,
+     (float)eps, (float)wd, (float)bc1, (float)bc2, (float)grad_scale, cur_stream());
+  check_launch("adamw_step");
+}
+
+void l2norm_sq(torch::Tensor chunks, bool bf16_grads, torch::Tensor out) {
+  CHECK_IN(chunks);
+  TORCH_CHECK(chunks.scalar_type() == torch::kInt64 && chunks.size(1) == 2);
+  auto fn = bf16_grads ? l2norm_sq_bf16 : l2norm_sq_f32;
+  fn(chunks.data_ptr(), (int)chunks.size(0), out.data_ptr<float>(), cur_stream());
+  check_launch("l2norm_sq");
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("ln_fwd", &ln_fwd);
+  m.def("ln_bwd", &ln_bwd);
+  m.def("bias_gelu_fwd", &bias_gelu_fwd);
+  m.def("bias_gelu_bwd", &bias_gelu_bwd);
+  m.def("colsum", &colsum);
+  m.def("bias_dropout_res_fwd", &bias_dropout_res_fwd);
+  m.def("bias_dropout_res_bwd", &bias_dropout_res_bwd);
+  m.def("softmax_fwd", &softmax_fwd);
+  m.def("softmax_bwd", &softmax_bwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
+  m.def("adamw_step", &adamw_step);
+  m.def("l2norm_sq", &l2norm_sq);
+  m.def("adamw_chunk_elems", &adamw_chunk_elems);
+}

@@ -1,0 +1,139 @@
+"""Column/row/data-parallel linear layers.
+
+Explicit-collective re-design of the reference's SBP-annotated Linear1D
+(reference: libai/layers/linear.py:78-168):
+
+  * ``parallel="col"``: weight sharded along the OUTPUT dim ([out/tp, in]).
+    Forward copies the (replicated) input into the TP region (identity fwd /
+    all-reduce bwd — the reference's grad_sbp pin at linear.py:131, C2 in
+    SURVEY.md §2.4) and produces a split output.
+  * ``parallel="row"``: weight sharded along the INPUT dim ([out, in/tp]).
+    The partial output is summed with an all-reduce (fwd) / identity (bwd)
+    (C1); the bias is added AFTER the reduction.
+  * ``parallel="data"``: plain replicated linear.
+
+``skip_bias_add=True`` returns (output, bias) so the caller can fuse the bias
+into a following elementwise kernel (bias_gelu / bias_dropout_add).
+"""
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from ..parallel.comm import (
+    copy_to_tensor_parallel_region,
+    reduce_from_tensor_parallel_region,
+)
+from ..utils import distributed as du
+
+__all__ = ["Linear1D", "Linear"]
+
+
+def init_tp_shard_(param, full_shape, init_method, shard_dim):
+    """Initialize a TP-sharded parameter as a SLICE of the full-tensor init.
+
+    All TP ranks draw the same full tensor (they run init under the same seed)
+    and keep their shard, so sharded == replicated numerics hold exactly —
+    the reference gets this from global-tensor init (libai/layers/linear.py:98-105).
+    """
+    dutil = du.get_dist_util()
+    tp, tpr = dutil.tensor_parallel_size, dutil.tensor_parallel_rank
+    full = torch.empty(*full_shape, dtype=param.dtype, device=param.device)
+    init_method(full)
+    if tp == 1:
+        with torch.no_grad():
+            param.copy_(full)
+        return
+    shard = full.chunk(tp, dim=shard_dim)[tpr]
+    with torch.no_grad():
+        param.copy_(shard)
+
+
+class Linear1D(nn.Module):
+    def __init__(
+        self,
+        in_features,
+        out_features,
+        bias=True,
+        parallel="data",
+        init_method=nn.init.xavier_normal_,
+        skip_bias_add=False,
+        *,
+        layer_idx=0,
+        dtype=None,
+    ):
+        super().__init__()
+        assert parallel in ("data", "col", "row"), parallel
+        self.in_features = in_features
+        self.out_features = out_features
+        self.parallel = parallel
+        self.skip_bias_add = skip_bias_add
+        self.layer_idx = layer_idx
+
+        dutil = du.get_dist_util()
+        tp = dutil.tensor_parallel_size
+        dtype = dtype or torch.get_default_dtype()
+
+        if parallel == "col":
+            assert out_features % tp == 0, (out_features, tp)
+            self.weight = nn.Parameter(
+                torch.empty(out_features // tp, in_features, dtype=dtype)
+            )
+            self.weight.tensor_parallel = True
+            init_tp_shard_(self.weight, (out_features, in_features), init_method, 0)
+            if bias:
+                self.bias = nn.Parameter(torch.zeros(out_features // tp, dtype=dtype))
+                self.bias.tensor_parallel = True
+            else:
+                self.register_parameter("bias", None)
+        elif parallel == "row":
+            assert in_features % tp == 0, (in_features, tp)
+            self.weight = nn.Parameter(
+                torch.empty(out_features, in_features // tp, dtype=dtype)
+            )
+            self.weight.tensor_parallel = True
+            init_tp_shard_(self.weight, (out_features, in_features), init_method, 1)
+            if bias:
+                self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype))
+            else:
+                self.register_parameter("bias", None)
+        else:
+            self.weight = nn.Parameter(torch.empty(out_features, in_features, dtype=dtype))
+            init_method(self.weight.data)
+            if bias:
+                self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype))
+            else:
+                self.register_parameter("bias", None)
+
+    def forward(self, x):
+        if self.parallel == "col":
+            x = copy_to_tensor_parallel_region(x)
+            out = F.linear(x, self.weight)
+            if self.skip_bias_add:
+                return out, self.bias
+            if self.bias is not None:
+                out = out + self.bias
+            return out
+        if self.parallel == "row":
+            out = F.linear(x, self.weight)
+            out = reduce_from_tensor_parallel_region(out)
+            if self.skip_bias_add:
+                return out, self.bias
+            if self.bias is not None:
+                out = out + self.bias
+            return out
+        out = F.linear(x, self.weight)
+        if self.skip_bias_add:
+            return out, self.bias
+        if self.bias is not None:
+            out = out + self.bias
+        return out
+
+    def extra_repr(self):
+        return (
+            f"in_features={self.in_features}, out_features={self.out_features}, "
+            f"parallel={self.parallel}, layer_idx={self.layer_idx}"
+        )
+
+
+Linear = Linear1D

@@ -1,0 +1,82 @@
+"""Fused bias+gelu and bias+dropout+residual ops.
+
+Replaces the reference's flow._C.fused_bias_add_gelu (libai/layers/mlp.py:95-97)
+and flow._C.fused_bias_add_dropout (+ the residual add that follows it in
+TransformerLayer; reference: libai/layers/attention.py:265-267,
+libai/layers/transformer_layer.py:170-232).
+"""
+
+import torch
+import torch.nn.functional as F
+
+from ._ext import draw_seed, ext, use_hip
+
+__all__ = ["bias_gelu", "bias_dropout_add"]
+
+
+class _BiasGeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, bias):
+        x = x.contiguous()
+        b = bias.contiguous() if bias is not None else None
+        y = ext().bias_gelu_fwd(x, b)
+        ctx.save_for_backward(x, *( (b,) if b is not None else () ))
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        if ctx.has_bias:
+            x, b = ctx.saved_tensors
+        else:
+            (x,) = ctx.saved_tensors
+            b = None
+        dy = dy.contiguous()
+        dx = ext().bias_gelu_bwd(x, b, dy)
+        dbias = ext().colsum(dx, x.shape[-1]) if ctx.has_bias else None
+        return dx, dbias
+
+
+def bias_gelu(x, bias=None):
+    """y = gelu(x + bias) with exact-erf gelu (torch default)."""
+    if use_hip(x):
+        return _BiasGeluFn.apply(x, bias)
+    return F.gelu(x + bias if bias is not None else x)
+
+
+class _BiasDropoutAddFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, bias, residual, p):
+        x = x.contiguous()
+        b = bias.contiguous() if bias is not None else None
+        r = residual.contiguous() if residual is not None else None
+        seed = draw_seed() if p > 0 else 0
+        y = ext().bias_dropout_res_fwd(x, b, r, p, seed)
+        ctx.p = p
+        ctx.seed = seed
+        ctx.has_bias = b is not None
+        ctx.has_res = r is not None
+        ctx.width = x.shape[-1]
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        dy = dy.contiguous()
+        if ctx.p > 0:
+            dx = ext().bias_dropout_res_bwd(dy, ctx.p, ctx.seed)
+        else:
+            dx = dy
+        dbias = ext().colsum(dx, ctx.width) if ctx.has_bias else None
+        dres = dy if ctx.has_res else None
+        return dx, dbias, dres, None
+
+
+def bias_dropout_add(x, bias=None, residual=None, p=0.0, training=True):
+    """y = residual + dropout(x + bias).  Philox-recomputed mask (no mask tensor)."""
+    if use_hip(x):
+        return _BiasDropoutAddFn.apply(x, bias, residual, p if training else 0.0)
+    out = x + bias if bias is not None else x
+    out = F.dropout(out, p=p, training=training)
+    if residual is not None:
+        out = out + residual
+    return out
