@@ -1,0 +1,226 @@
+"""Standard hooks (reference: libai/engine/hooks.py:46-417)."""
+
+import logging
+import math
+import time
+
+import torch
+
+from ..utils import distributed as du
+from ..utils.events import EventWriter
+from ..utils.timer import Timer
+from .trainer import HookBase
+
+__all__ = [
+    "CallbackHook",
+    "IterationTimer",
+    "PeriodicWriter",
+    "PeriodicCheckpointerHook",
+    "BestCheckpointer",
+    "EvalHook",
+    "LRScheduler",
+]
+
+
+class CallbackHook(HookBase):
+    def __init__(self, *, before_train=None, after_train=None, before_step=None,
+                 after_step=None):
+        self._before_train = before_train
+        self._after_train = after_train
+        self._before_step = before_step
+        self._after_step = after_step
+
+    def before_train(self):
+        if self._before_train:
+            self._before_train(self.trainer)
+
+    def after_train(self):
+        self.trainer = None
+        if self._after_train:
+            self._after_train()
+
+    def before_step(self):
+        if self._before_step:
+            self._before_step(self.trainer)
+
+    def after_step(self):
+        if self._after_step:
+            self._after_step(self.trainer)
+
+
+class IterationTimer(HookBase):
+    """Wall-time per step, excluding warmup (reference: hooks.py:81-145)."""
+
+    def __init__(self, warmup_iter=3):
+        self._warmup_iter = warmup_iter
+        self._step_timer = Timer()
+        self._start_time = time.perf_counter()
+        self._total_timer = Timer()
+
+    def before_train(self):
+        self._start_time = time.perf_counter()
+        self._total_timer.reset()
+        self._total_timer.pause()
+
+    def after_train(self):
+        logger = logging.getLogger(__name__)
+        total_time = time.perf_counter() - self._start_time
+        total_time_minus_hooks = self._total_timer.seconds()
+        hook_time = total_time - total_time_minus_hooks
+        num_iter = self.trainer.iter + 1 - self.trainer.start_iter - self._warmup_iter
+        if num_iter > 0 and total_time_minus_hooks > 0:
+            logger.info(
+                f"Total training time: {total_time_minus_hooks:.2f}s "
+                f"({total_time_minus_hooks / num_iter:.4f} s/iter, "
+                f"{hook_time:.2f}s in hooks)"
+            )
+
+    def before_step(self):
+        self._step_timer.reset()
+        self._total_timer.resume()
+
+    def after_step(self):
+        num_iter = self.trainer.iter - self.trainer.start_iter + 1
+        if num_iter > self._warmup_iter:
+            self.trainer.storage.put_scalar("time", self._step_timer.seconds())
+        else:
+            self._start_time = time.perf_counter()
+            self._total_timer.reset()
+        self._total_timer.pause()
+
+
+class PeriodicWriter(HookBase):
+    def __init__(self, writers, period=20):
+        self._writers = writers
+        for w in writers:
+            assert isinstance(w, EventWriter)
+        self._period = period
+
+    def after_step(self):
+        if (self.trainer.iter + 1) % self._period == 0 or (
+            self.trainer.iter == self.trainer.max_iter - 1
+        ):
+            for writer in self._writers:
+                writer.write()
+
+    def after_train(self):
+        for writer in self._writers:
+            writer.write()
+            writer.close()
+
+
+class PeriodicCheckpointerHook(HookBase):
+    def __init__(self, periodic_checkpointer):
+        self._pc = periodic_checkpointer
+
+    def before_train(self):
+        self._pc.max_iter = self.trainer.max_iter
+
+    def after_step(self):
+        self._pc.step(self.trainer.iter)
+
+
+class BestCheckpointer(HookBase):
+    """Track a validation metric and save model_best (reference: hooks.py:193-293)."""
+
+    def __init__(self, eval_period, checkpointer, val_metric, mode="max",
+                 file_prefix="model_best"):
+        self._period = eval_period
+        self._checkpointer = checkpointer
+        self._metric = val_metric
+        assert mode in ("max", "min")
+        self._compare = (lambda a, b: a > b) if mode == "max" else (lambda a, b: a < b)
+        self._prefix = file_prefix
+        self.best_value = None
+        self.best_iter = None
+        self.logger = logging.getLogger(__name__)
+
+    def _best_checking(self):
+        storage = self.trainer.storage
+        try:
+            latest = storage.latest()[self._metric]
+        except KeyError:
+            self.logger.warning(
+                f"no metric {self._metric!r} found for best-checkpointing"
+            )
+            return
+        value, it = latest
+        if value is None or math.isnan(value) or math.isinf(value):
+            return
+        if self.best_value is None or self._compare(value, self.best_value):
+            self.best_value, self.best_iter = value, it
+            self._checkpointer.save(self._prefix, iteration=it)
+            self.logger.info(
+                f"saved best model at iter {it} with {self._metric}={value:.4f}"
+            )
+
+    def after_step(self):
+        if self._period > 0 and (self.trainer.iter + 1) % self._period == 0 and (
+            self.trainer.iter != self.trainer.max_iter - 1
+        ):
+            self._best_checking()
+
+    def after_train(self):
+        if self.trainer.iter + 1 >= self.trainer.max_iter:
+            self._best_checking()
+
+
+class EvalHook(HookBase):
+    def __init__(self, eval_period, eval_function):
+        self._period = eval_period
+        self._func = eval_function
+
+    def _do_eval(self):
+        results = self._func()
+        if results:
+            flattened = {}
+
+            def _flat(d, prefix=""):
+                for k, v in d.items():
+                    if isinstance(v, dict):
+                        _flat(v, prefix + k + "/")
+                    else:
+                        try:
+                            flattened[prefix + k] = float(v)
+                        except (TypeError, ValueError):
+                            pass
+
+            _flat(results)
+            if du.is_main_process():
+                self.trainer.storage.put_scalars(**flattened, smoothing_hint=False)
+        du.synchronize()
+
+    def after_step(self):
+        if self._period > 0 and (self.trainer.iter + 1) % self._period == 0 and (
+            self.trainer.iter != self.trainer.max_iter - 1
+        ):
+            self._do_eval()
+
+    def after_train(self):
+        if self.trainer.iter + 1 >= self.trainer.max_iter:
+            self._do_eval()
+
+
+class LRScheduler(HookBase):
+    """Step the LR scheduler each iteration and log lr (reference: hooks.py:357-417)."""
+
+    def __init__(self, optimizer=None, scheduler=None):
+        self._optimizer = optimizer
+        self._scheduler = scheduler
+
+    def before_train(self):
+        self._optimizer = self._optimizer or self.trainer.optimizer
+        self._scheduler = self._scheduler or getattr(self.trainer, "lr_scheduler", None)
+        self._best_param_group_id = 0
+        largest = max(len(g["params"]) for g in self._optimizer.param_groups)
+        for i, g in enumerate(self._optimizer.param_groups):
+            if len(g["params"]) == largest:
+                self._best_param_group_id = i
+                break
+
+    def after_step(self):
+        lr = self._optimizer.param_groups[self._best_param_group_id]["lr"]
+        if du.is_main_process():
+            self.trainer.storage.put_scalar("lr", lr, smoothing_hint=False)
+        if self._scheduler is not None:
+            self._scheduler.step()

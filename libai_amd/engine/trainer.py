@@ -1,0 +1,201 @@
+"""Training loop core: HookBase, TrainerBase, EagerTrainer.
+
+Reference behavior: libai/engine/trainer.py:32-349.  The reference's
+GraphTrainer (OneFlow nn.Graph) has no analog here; the eager trainer plus
+the explicit pipeline scheduler cover both execution modes, and a
+hipGraph-captured steady-state step is the planned equivalent optimization.
+"""
+
+import logging
+import time
+import weakref
+
+import numpy as np
+import torch
+
+from ..utils import distributed as du
+from ..utils.events import EventStorage
+
+__all__ = ["HookBase", "TrainerBase", "EagerTrainer"]
+
+
+class HookBase:
+    """4-point lifecycle hooks (reference: trainer.py:32-87)."""
+
+    trainer = None
+
+    def before_train(self):
+        pass
+
+    def after_train(self):
+        pass
+
+    def before_step(self):
+        pass
+
+    def after_step(self):
+        pass
+
+
+class TrainerBase:
+    def __init__(self):
+        self._hooks = []
+        self.start_iter = 0
+        self.max_iter = 0
+        self.iter = 0
+        self.storage = None
+
+    def register_hooks(self, hooks):
+        hooks = [h for h in hooks if h is not None]
+        for h in hooks:
+            assert isinstance(h, HookBase)
+            h.trainer = weakref.proxy(self)
+        self._hooks.extend(hooks)
+
+    def train(self, start_iter, max_iter):
+        logger = logging.getLogger(__name__)
+        logger.info(f"Starting training from iteration {start_iter}")
+        self.iter = self.start_iter = start_iter
+        self.max_iter = max_iter
+        with EventStorage(start_iter) as self.storage:
+            try:
+                self.before_train()
+                for self.iter in range(start_iter, max_iter):
+                    self.before_step()
+                    self.run_step()
+                    self.after_step()
+                self.iter += 1
+            except Exception:
+                logger.exception("Exception during training:")
+                raise
+            finally:
+                self.after_train()
+
+    def before_train(self):
+        for h in self._hooks:
+            h.before_train()
+
+    def after_train(self):
+        if self.storage is not None:
+            self.storage.iter = self.iter
+        for h in self._hooks:
+            h.after_train()
+
+    def before_step(self):
+        self.storage.iter = self.iter
+        for h in self._hooks:
+            h.before_step()
+
+    def after_step(self):
+        for h in self._hooks:
+            h.after_step()
+
+    def run_step(self):
+        raise NotImplementedError
+
+    @staticmethod
+    def write_metrics(loss_dict, data_time, prefix=""):
+        """Log scalars, gathered/averaged over DP ranks on rank 0
+        (reference: trainer.py:179-219)."""
+        metrics = (
+            {k: v.detach().cpu().item() for k, v in loss_dict.items()}
+            if loss_dict is not None
+            else {}
+        )
+        metrics["data_time"] = data_time
+        import torch.distributed as dist
+
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            gathered = [None] * dist.get_world_size()
+            dist.all_gather_object(gathered, metrics)
+        else:
+            gathered = [metrics]
+        if du.is_main_process():
+            from ..utils.events import get_event_storage
+
+            storage = get_event_storage()
+            data_time = np.max([x.pop("data_time", 0.0) for x in gathered])
+            storage.put_scalar("data_time", data_time)
+            keys = set()
+            for x in gathered:
+                keys.update(x.keys())
+            averaged = {
+                k: np.mean([x[k] for x in gathered if k in x]) for k in keys
+            }
+            total = sum(averaged.values())
+            if not np.isfinite(total):
+                raise FloatingPointError(
+                    f"Loss became infinite or NaN at iteration {storage.iter}: {averaged}"
+                )
+            storage.put_scalar(f"{prefix}total_loss", total)
+            if len(averaged) > 1:
+                storage.put_scalars(**{f"{prefix}{k}": v for k, v in averaged.items()})
+
+
+class EagerTrainer(TrainerBase):
+    """Micro-batch accumulation loop (reference: trainer.py:263-286).
+
+    One run_step = grad_acc_steps micro-batches -> DP grad sync -> clip +
+    fused AdamW step -> zero.  Grad sync is one all-reduce per flat bucket.
+    """
+
+    def __init__(self, model, data_loader, optimizer, grad_acc_steps=1,
+                 pipeline_scheduler=None):
+        super().__init__()
+        model.train()
+        self.model = model
+        self.data_loader = data_loader
+        self._data_loader_iter = iter(data_loader)
+        self.optimizer = optimizer
+        self.grad_acc_steps = grad_acc_steps
+        self.pipeline_scheduler = pipeline_scheduler
+
+    def get_batch(self, data):
+        from ..data.structures import Instance
+
+        if isinstance(data, Instance):
+            data = data.to_dict()
+        device = du.get_device()
+        return {
+            k: (v.to(device, non_blocking=True) if torch.is_tensor(v) else v)
+            for k, v in data.items()
+        }
+
+    def _sync_dp_grads(self):
+        import torch.distributed as dist
+
+        dutil = du.get_dist_util()
+        if dutil.data_parallel_size == 1 or not dist.is_initialized():
+            return
+        for _, b in self.optimizer.buckets:
+            b.flat_grad.div_(dutil.data_parallel_size)
+            dist.all_reduce(b.flat_grad, group=dutil.data_parallel_group)
+
+    def run_step(self):
+        start = time.perf_counter()
+        if self.pipeline_scheduler is not None:
+            batches = []
+            for _ in range(self.grad_acc_steps):
+                batches.append(self.get_batch(next(self._data_loader_iter)))
+            data_time = time.perf_counter() - start
+            loss_dict = self.pipeline_scheduler.run_1f1b(batches)
+        else:
+            data_time = 0.0
+            loss_dict = None
+            for _ in range(self.grad_acc_steps):
+                t0 = time.perf_counter()
+                data = self.get_batch(next(self._data_loader_iter))
+                data_time += time.perf_counter() - t0
+                losses = self.model(**data)
+                losses = {k: v for k, v in losses.items() if v.requires_grad or v.is_floating_point()}
+                total = sum(losses.values()) / self.grad_acc_steps
+                total.backward()
+                if loss_dict is None:
+                    loss_dict = {k: v.detach() / self.grad_acc_steps for k, v in losses.items()}
+                else:
+                    for k, v in losses.items():
+                        loss_dict[k] += v.detach() / self.grad_acc_steps
+        self._sync_dp_grads()
+        self.optimizer.step()
+        self.optimizer.zero_grad()
+        self.write_metrics(loss_dict, data_time)

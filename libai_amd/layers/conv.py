@@ -35,16 +35,19 @@ class Conv1D(nn.Module):
             assert out_features % tp == 0
             self.weight = nn.Parameter(torch.empty(in_features, out_features // tp, dtype=dtype))
             self.weight.tensor_parallel = True
+            self.weight.tp_shard_dim = 1
             init_tp_shard_(self.weight, (in_features, out_features), init_method, 1)
             self.bias = (
                 nn.Parameter(torch.zeros(out_features // tp, dtype=dtype)) if bias else None
             )
             if self.bias is not None:
                 self.bias.tensor_parallel = True
+                self.bias.tp_shard_dim = 0
         elif parallel == "row":
             assert in_features % tp == 0
             self.weight = nn.Parameter(torch.empty(in_features // tp, out_features, dtype=dtype))
             self.weight.tensor_parallel = True
+            self.weight.tp_shard_dim = 0
             init_tp_shard_(self.weight, (in_features, out_features), init_method, 0)
             self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype)) if bias else None
         else:

@@ -60,6 +60,7 @@ class VocabEmbedding(nn.Module):
             torch.empty(self.vocab_local, embedding_dim, dtype=dtype)
         )
         self.weight.tensor_parallel = True
+        self.weight.tp_shard_dim = 0
         init_tp_shard_(self.weight, (num_embeddings, embedding_dim), init_method, 0)
 
     def forward(self, ids):

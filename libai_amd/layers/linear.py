@@ -80,10 +80,12 @@ class Linear1D(nn.Module):
                 torch.empty(out_features // tp, in_features, dtype=dtype)
             )
             self.weight.tensor_parallel = True
+            self.weight.tp_shard_dim = 0
             init_tp_shard_(self.weight, (out_features, in_features), init_method, 0)
             if bias:
                 self.bias = nn.Parameter(torch.zeros(out_features // tp, dtype=dtype))
                 self.bias.tensor_parallel = True
+                self.bias.tp_shard_dim = 0
             else:
                 self.register_parameter("bias", None)
         elif parallel == "row":
@@ -92,6 +94,7 @@ class Linear1D(nn.Module):
                 torch.empty(out_features, in_features // tp, dtype=dtype)
             )
             self.weight.tensor_parallel = True
+            self.weight.tp_shard_dim = 1
             init_tp_shard_(self.weight, (out_features, in_features), init_method, 1)
             if bias:
                 self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype))

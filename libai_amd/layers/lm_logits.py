@@ -28,6 +28,7 @@ class LMLogits(nn.Module):
 
             self.bias = nn.Parameter(torch.zeros(vocab_size // tp))
             self.bias.tensor_parallel = True
+            self.bias.tp_shard_dim = 0
         else:
             self.register_parameter("bias", None)
 

@@ -46,6 +46,8 @@ class _Bucket:
             n = p.numel()
             self.flat_param[off : off + n].copy_(p.data.reshape(-1))
             p.data = self.flat_param[off : off + n].view(p.shape)
+            if p.grad is not None:  # preserve grads accumulated before flattening
+                self.flat_grad[off : off + n].copy_(p.grad.reshape(-1))
             p.grad = self.flat_grad[off : off + n].view(p.shape)
             off += n
         self.flat_master = self.flat_param.float() if dtype != torch.float32 else self.flat_param
