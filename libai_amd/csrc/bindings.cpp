@@ -80,6 +80,14 @@ void ce_bwd_bf16(const void*, const int64_t*, const float*, const float*, const 
                  void*, int64_t, int64_t, int64_t, int64_t, hipStream_t);
 void ce_bwd_f32(const void*, const int64_t*, const float*, const float*, const float*,
                 void*, int64_t, int64_t, int64_t, int64_t, hipStream_t);
+void flash_fwd_bf16(const void*, const void*, const void*, void*, float*, int64_t,
+                    int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
+                    int64_t, int64_t, int64_t, int64_t, int, int, int, int, int, float,
+                    float, uint64_t, int, hipStream_t);
+void attn_dropout_apply_bf16(void*, int64_t, int64_t, int64_t, float, uint64_t,
+                             hipStream_t);
+void attn_dropout_apply_f32(void*, int64_t, int64_t, int64_t, float, uint64_t,
+                            hipStream_t);
 int adamw_chunk_elems();
 void adamw_step_bf16(const void*, int, float, float, float, float, float, float, float,
                      float, hipStream_t);
@@ -282,6 +290,44 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets, torch::Tensor 
 }
 
 // ---------------------------------------------------------------------------
+// flash attention
+// ---------------------------------------------------------------------------
+std::tuple<torch::Tensor, torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k,
+                                                   torch::Tensor v, double scale,
+                                                   double p_drop, int64_t seed,
+                                                   bool causal) {
+  // q/k/v: [B, S, H, D] (strided views into a fused qkv buffer are fine; the
+  // last dim must be contiguous and 16B-aligned)
+  TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "flash_fwd is bf16-only");
+  TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1);
+  const int B = (int)q.size(0), Sq = (int)q.size(1), H = (int)q.size(2),
+            D = (int)q.size(3);
+  const int Sk = (int)k.size(1);
+  TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+  TORCH_CHECK(Sk % 8 == 0, "Sk must be a multiple of 8");
+  auto o = torch::empty({B, Sq, H, D}, q.options());
+  auto lse = torch::empty({B, H, Sq}, q.options().dtype(torch::kFloat32));
+  flash_fwd_bf16(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                 lse.data_ptr<float>(), q.stride(0), q.stride(1), q.stride(2),
+                 k.stride(0), k.stride(1), k.stride(2), v.stride(0), v.stride(1),
+                 v.stride(2), o.stride(0), o.stride(1), o.stride(2), B, H, Sq, Sk, D,
+                 (float)scale, (float)p_drop, (uint64_t)seed, causal ? 1 : 0,
+                 cur_stream());
+  check_launch("flash_fwd");
+  return {o, lse};
+}
+
+void attn_dropout_apply(torch::Tensor x, int64_t Sq, int64_t Sk, double p,
+                        int64_t seed) {
+  CHECK_IN(x);
+  int64_t BH = x.numel() / (Sq * Sk);
+  auto fn = is_bf16(x) ? attn_dropout_apply_bf16 : attn_dropout_apply_f32;
+  fn(x.data_ptr(), BH, Sq, Sk, (float)p, (uint64_t)seed, cur_stream());
+  check_launch("attn_dropout_apply");
+}
+
+// ---------------------------------------------------------------------------
 // fused AdamW
 // ---------------------------------------------------------------------------
 void adamw_step(torch::Tensor chunks, bool bf16_params, double lr, double beta1,
@@ -313,6 +359,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_dropout_res_bwd", &bias_dropout_res_bwd);
   m.def("softmax_fwd", &softmax_fwd);
   m.def("softmax_bwd", &softmax_bwd);
+  m.def("flash_fwd", &flash_fwd);
+  m.def("attn_dropout_apply", &attn_dropout_apply);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("adamw_step", &adamw_step);

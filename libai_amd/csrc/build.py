@@ -60,7 +60,6 @@ def build(force=False, verbose=True):
                 "-O3",
                 "-std=c++17",
                 "-fPIC",
-                "-ffast-math",
                 "-c",
                 src,
                 "-o",

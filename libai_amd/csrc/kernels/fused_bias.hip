@@ -110,14 +110,27 @@ __global__ void colsum_partial_kernel(const typename E::T* __restrict__ in,
   partial[(int64_t)blockIdx.y * W + j] = s;
 }
 
+// fold: block = 64 cols x 8 p-lanes so the strided partial reads have TLP
+// (a [W]-thread fold is latency-bound: 4 blocks cannot hide 256 dependent
+// HBM/L2 round trips).
 template <class E>
 __global__ void colsum_fold_kernel(const float* __restrict__ partial,
                                    typename E::T* __restrict__ out, int P, int W) {
-  const int j = blockIdx.x * blockDim.x + threadIdx.x;
-  if (j >= W) return;
+  __shared__ float lds[8][64];
+  const int j = blockIdx.x * 64 + (int)(threadIdx.x % 64);
+  const int pl = threadIdx.x / 64;  // 8 p-lanes
   float s = 0.f;
-  for (int p = 0; p < P; ++p) s += partial[(int64_t)p * W + j];
-  out[j] = E::from_f(s);
+  if (j < W) {
+    for (int p = pl; p < P; p += 8) s += partial[(int64_t)p * W + j];
+  }
+  lds[pl][threadIdx.x % 64] = s;
+  __syncthreads();
+  if (pl == 0 && j < W) {
+    float acc = 0.f;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) acc += lds[q][threadIdx.x % 64];
+    out[j] = E::from_f(acc);
+  }
 }
 
 inline int64_t ew_grid(int64_t nvec) {
@@ -161,7 +174,7 @@ inline int64_t ew_grid(int64_t nvec) {
                                 int W, int P, hipStream_t stream) {                      \
     colsum_partial_kernel<ETYPE><<<dim3(CDIV(W, 256), P), dim3(256), 0, stream>>>(       \
         (const ETYPE::T*)in, partial, R, W);                                             \
-    colsum_fold_kernel<ETYPE><<<dim3(CDIV(W, 256)), dim3(256), 0, stream>>>(             \
+    colsum_fold_kernel<ETYPE><<<dim3(CDIV(W, 64)), dim3(512), 0, stream>>>(              \
         partial, (ETYPE::T*)out, P, W);                                                  \
   }
 

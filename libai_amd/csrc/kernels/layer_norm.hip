@@ -171,20 +171,36 @@ __global__ void norm_bwd_wgrad_partial_kernel(const typename E::T* __restrict__ 
   if (pbeta) pbeta[(int64_t)blockIdx.y * H + j] = db;
 }
 
+// fold over P row-group partials: 64 cols x 8 p-lanes per block (TLP over the
+// strided reads; a [H]-thread fold is latency-bound at H~1k).
 template <class E>
 __global__ void wgrad_fold_kernel(const float* __restrict__ pgamma,
                                   const float* __restrict__ pbeta,
                                   typename E::T* __restrict__ dgamma,
                                   typename E::T* __restrict__ dbeta, int P, int H) {
-  const int j = blockIdx.x * blockDim.x + threadIdx.x;
-  if (j >= H) return;
+  __shared__ float lds[2][8][64];
+  const int j = blockIdx.x * 64 + (int)(threadIdx.x % 64);
+  const int pl = threadIdx.x / 64;
   float dg = 0.f, db = 0.f;
-  for (int p = 0; p < P; ++p) {
-    dg += pgamma[(int64_t)p * H + j];
-    if (pbeta) db += pbeta[(int64_t)p * H + j];
+  if (j < H) {
+    for (int p = pl; p < P; p += 8) {
+      dg += pgamma[(int64_t)p * H + j];
+      if (pbeta) db += pbeta[(int64_t)p * H + j];
+    }
   }
-  dgamma[j] = E::from_f(dg);
-  if (dbeta) dbeta[j] = E::from_f(db);
+  lds[0][pl][threadIdx.x % 64] = dg;
+  lds[1][pl][threadIdx.x % 64] = db;
+  __syncthreads();
+  if (pl == 0 && j < H) {
+    float sdg = 0.f, sdb = 0.f;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) {
+      sdg += lds[0][q][threadIdx.x % 64];
+      sdb += lds[1][q][threadIdx.x % 64];
+    }
+    dgamma[j] = E::from_f(sdg);
+    if (dbeta) dbeta[j] = E::from_f(sdb);
+  }
 }
 
 inline int row_block_threads(int H, int vec) {
@@ -236,7 +252,7 @@ inline int row_block_threads(int H, int vec) {
     else                                                                                \
       norm_bwd_wgrad_partial_kernel<ETYPE, false><<<grid1, dim3(256), 0, stream>>>(     \
           (const ETYPE::T*)dy, (const ETYPE::T*)x, mean, rstd, pgamma, pbeta, R, H);    \
-    wgrad_fold_kernel<ETYPE><<<dim3(CDIV(H, 256)), dim3(256), 0, stream>>>(             \
+    wgrad_fold_kernel<ETYPE><<<dim3(CDIV(H, 64)), dim3(512), 0, stream>>>(              \
         pgamma, pbeta, (ETYPE::T*)dgamma, (ETYPE::T*)dbeta, P, H);                      \
   }
 

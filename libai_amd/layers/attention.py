@@ -105,6 +105,36 @@ class MultiheadAttention(nn.Module):
         If ``residual`` is given, returns hidden + dropout(out + bias) (the
         fused TransformerLayer path); otherwise applies bias+dropout only.
         """
+        # fused flash path: consumes the qkv buffer with ZERO copies (strided
+        # [b, s, nh, hs] views), O lands directly in [b, s, h] layout.
+        if (
+            not self.is_cross_attention
+            and past_key_value is None
+            and not use_cache
+        ):
+            from ..ops.attention import flash_attention, flash_attention_available
+
+            b, s, _ = hidden_states.shape
+            if flash_attention_available(
+                self.head_size, hidden_states.dtype, hidden_states.device, s, s,
+                attention_mask,
+            ):
+                qkv = self.query_key_value(hidden_states)
+                qkv5 = qkv.view(b, s, self.num_heads_local, 3, self.head_size)
+                o = flash_attention(
+                    qkv5[..., 0, :], qkv5[..., 1, :], qkv5[..., 2, :],
+                    scale=self.norm_factor * (self.coeff if self.coeff else 1.0),
+                    p_drop=self.attention_dropout_prob,
+                    causal=self.attn_mask_type == AttnMaskType.causal,
+                    training=self.training,
+                )
+                context = o.reshape(b, s, self.num_heads_local * self.head_size)
+                out, bias = self.dense(context)
+                return bias_dropout_add(
+                    out, bias=bias, residual=residual, p=self.output_dropout_prob,
+                    training=self.training,
+                )
+
         if self.is_cross_attention:
             q = self._split_heads(self.query(hidden_states), 1)[0]
             if past_key_value is not None:

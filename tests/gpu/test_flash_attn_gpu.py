@@ -1,0 +1,140 @@
+"""Flash-attention kernel vs plain PyTorch fp32 reference."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _setup():
+    from libai_amd.utils import distributed as du
+
+    du.setup_dist_util({})
+    yield
+
+
+def _ref_attention(q, k, v, scale, causal):
+    # fp32 reference on [B, S, H, D] inputs
+    qh = q.float().permute(0, 2, 1, 3)
+    kh = k.float().permute(0, 2, 1, 3)
+    vh = v.float().permute(0, 2, 1, 3)
+    s = torch.matmul(qh, kh.transpose(-1, -2)) * scale
+    if causal:
+        sq, sk = s.shape[-2], s.shape[-1]
+        cm = torch.ones(sq, sk, dtype=torch.bool, device=q.device).tril_(sk - sq)
+        s = s.masked_fill(~cm, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    o = torch.matmul(p, vh)
+    return o.permute(0, 2, 1, 3)
+
+
+@pytest.mark.parametrize("d", [64, 128])
+@pytest.mark.parametrize("causal", [True, False])
+@pytest.mark.parametrize("s", [128, 384, 1024])
+def test_flash_fwd_matches_reference(d, causal, s):
+    from libai_amd.ops.attention import flash_attention
+
+    torch.manual_seed(0)
+    b, h = 2, 4
+    q = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / math.sqrt(d)
+    o = flash_attention(q, k, v, scale, p_drop=0.0, causal=causal)
+    ref = _ref_attention(q, k, v, scale, causal)
+    err = (o.float() - ref).abs().max().item()
+    assert err < 2e-2, f"flash fwd max err {err}"
+
+
+def test_flash_fwd_strided_qkv_views():
+    """The fused-qkv strided views must give identical results."""
+    from libai_amd.ops.attention import flash_attention
+
+    torch.manual_seed(0)
+    b, s, h, d = 2, 256, 4, 64
+    qkv = torch.randn(b, s, h, 3, d, device="cuda", dtype=torch.bfloat16)
+    q, k, v = qkv[..., 0, :], qkv[..., 1, :], qkv[..., 2, :]
+    o1 = flash_attention(q, k, v, 0.125, causal=True)
+    o2 = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(), 0.125,
+                         causal=True)
+    assert torch.equal(o1, o2)
+
+
+@pytest.mark.parametrize("causal", [True, False])
+def test_flash_backward_matches_reference(causal):
+    from libai_amd.ops.attention import flash_attention
+
+    torch.manual_seed(0)
+    b, s, h, d = 2, 256, 4, 64
+    q = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    scale = 1.0 / math.sqrt(d)
+    o = flash_attention(q, k, v, scale, p_drop=0.0, causal=causal)
+    g = torch.randn_like(o)
+    o.backward(g)
+
+    qr = q.detach().float().requires_grad_(True)
+    kr = k.detach().float().requires_grad_(True)
+    vr = v.detach().float().requires_grad_(True)
+    ref = _ref_attention(qr, kr, vr, scale, causal)
+    ref.backward(g.float())
+
+    for name, got, want in [("dq", q.grad, qr.grad), ("dk", k.grad, kr.grad),
+                            ("dv", v.grad, vr.grad)]:
+        rel = (got.float() - want).abs().max() / (want.abs().max() + 1e-6)
+        assert rel < 5e-2, f"{name} rel err {rel}"
+
+
+def test_flash_dropout_statistics():
+    from libai_amd.ops.attention import flash_attention
+
+    torch.manual_seed(3)
+    b, s, h, d = 2, 256, 4, 64
+    q = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn_like(q)
+    v = torch.ones_like(q)
+    p = 0.5
+    # with V = ones, each output elem = sum of kept P / (1-p); mean over many
+    # rows ~ 1.0 (since sum P = 1)
+    o = flash_attention(q, k, v, 0.125, p_drop=p, causal=False, training=True)
+    mean = o.float().mean().item()
+    assert abs(mean - 1.0) < 0.05, f"dropout-scaled mean {mean} != 1"
+    # eval mode: no dropout
+    o2 = flash_attention(q, k, v, 0.125, p_drop=p, causal=False, training=False)
+    assert abs(o2.float().mean().item() - 1.0) < 1e-2
+    assert (o2.float() - 1.0).abs().max() < 0.05
+
+
+def test_flash_dropout_backward_runs_and_finite():
+    from libai_amd.ops.attention import flash_attention
+
+    torch.manual_seed(0)
+    b, s, h, d = 2, 128, 2, 64
+    q = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn_like(q).requires_grad_(True)
+    v = torch.randn_like(q).requires_grad_(True)
+    o = flash_attention(q, k, v, 0.125, p_drop=0.1, causal=True)
+    o.sum().backward()
+    for t in (q.grad, k.grad, v.grad):
+        assert torch.isfinite(t.float()).all()
+
+
+def test_attention_layer_uses_flash_and_matches_unfused():
+    """MultiheadAttention flash path vs its own unfused path (eval)."""
+    from libai_amd import layers
+
+    torch.manual_seed(0)
+    attn = layers.MultiheadAttention(256, 4, attn_mask_type="causal")
+    attn = attn.to(torch.bfloat16).cuda().eval()
+    x = torch.randn(2, 128, 256, device="cuda", dtype=torch.bfloat16)
+    y_flash = attn(x)
+    # force unfused path by pretending there's a cache request
+    out_unfused, _ = attn(x, use_cache=True)
+    rel = (y_flash.float() - out_unfused.float()).abs().max() / (
+        out_unfused.float().abs().max() + 1e-6
+    )
+    assert rel < 3e-2, f"flash vs unfused layer mismatch {rel}"
