@@ -84,6 +84,13 @@ void flash_fwd_bf16(const void*, const void*, const void*, void*, float*, int64_
                     int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
                     int64_t, int64_t, int64_t, int64_t, int, int, int, int, int, float,
                     float, uint64_t, int, hipStream_t);
+void flash_bwd_bf16(const void*, const void*, const void*, const void*, const void*,
+                    const float*, float*, void*, void*, void*, int64_t, int64_t,
+                    int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
+                    int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
+                    int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
+                    int64_t, int, int, int, int, int, float, float, uint64_t, int,
+                    hipStream_t);
 void attn_dropout_apply_bf16(void*, int64_t, int64_t, int64_t, float, uint64_t,
                              hipStream_t);
 void attn_dropout_apply_f32(void*, int64_t, int64_t, int64_t, float, uint64_t,
@@ -318,6 +325,40 @@ std::tuple<torch::Tensor, torch::Tensor> flash_fwd(torch::Tensor q, torch::Tenso
   return {o, lse};
 }
 
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> flash_bwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor o,
+    torch::Tensor dout, torch::Tensor lse, double scale, double p_drop, int64_t seed,
+    bool causal, c10::optional<torch::Tensor> dqkv) {
+  const int B = (int)q.size(0), Sq = (int)q.size(1), H = (int)q.size(2),
+            D = (int)q.size(3);
+  const int Sk = (int)k.size(1);
+  TORCH_CHECK(dout.is_contiguous() && o.is_contiguous());
+  torch::Tensor dq, dk, dv;
+  if (dqkv.has_value()) {
+    // [B, S, H, 3, D] fused grad buffer: write q/k/v grads in place
+    auto g = dqkv.value();
+    dq = g.select(3, 0);
+    dk = g.select(3, 1);
+    dv = g.select(3, 2);
+  } else {
+    dq = torch::empty_like(q, q.options());
+    dk = torch::empty({B, Sk, H, D}, q.options());
+    dv = torch::empty({B, Sk, H, D}, q.options());
+  }
+  auto drow = torch::empty({B, H, Sq}, q.options().dtype(torch::kFloat32));
+  flash_bwd_bf16(
+      q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(), dout.data_ptr(),
+      lse.data_ptr<float>(), drow.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
+      dv.data_ptr(), q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),
+      k.stride(2), v.stride(0), v.stride(1), v.stride(2), o.stride(0), o.stride(1),
+      o.stride(2), dout.stride(0), dout.stride(1), dout.stride(2), dq.stride(0),
+      dq.stride(1), dq.stride(2), dk.stride(0), dk.stride(1), dk.stride(2),
+      dv.stride(0), dv.stride(1), dv.stride(2), B, H, Sq, Sk, D, (float)scale,
+      (float)p_drop, (uint64_t)seed, causal ? 1 : 0, cur_stream());
+  check_launch("flash_bwd");
+  return {dq, dk, dv};
+}
+
 void attn_dropout_apply(torch::Tensor x, int64_t Sq, int64_t Sk, double p,
                         int64_t seed) {
   CHECK_IN(x);
@@ -360,6 +401,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_fwd", &softmax_fwd);
   m.def("softmax_bwd", &softmax_bwd);
   m.def("flash_fwd", &flash_fwd);
+  m.def("flash_bwd", &flash_bwd);
   m.def("attn_dropout_apply", &attn_dropout_apply);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);

@@ -142,6 +142,20 @@ __device__ __forceinline__ float u32_to_uniform(uint32_t x) {
   return (x >> 8) * (1.0f / 16777216.0f);
 }
 
+// Cheap per-element counter RNG (splitmix64 finalizer): used by attention
+// dropout where fwd and bwd kernels index elements in DIFFERENT lane layouts,
+// so a per-element (not per-4-group) generator keeps regeneration cheap on
+// both sides.
+__device__ __forceinline__ uint32_t rnd_hash(uint64_t seed, uint64_t idx) {
+  uint64_t x = seed + idx * 0x9E3779B97F4A7C15ull;
+  x ^= x >> 30;
+  x *= 0xBF58476D1CE4E5B9ull;
+  x ^= x >> 27;
+  x *= 0x94D049BB133111EBull;
+  x ^= x >> 31;
+  return (uint32_t)x;
+}
+
 // ---------------------------------------------------------------------------
 // gelu (tanh approximation, matching torch.nn.functional.gelu(approximate="tanh"))
 // and its derivative; plus exact-erf gelu to match torch's default.

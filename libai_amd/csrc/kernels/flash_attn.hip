@@ -1,27 +1,31 @@
-// Fused flash-style scaled-dot-product attention for gfx950 (bf16, MFMA).
+// Fused flash-style scaled-dot-product attention for gfx950 (bf16, MFMA):
+// forward + full backward.
 //
 // Replaces the reference's unfused QK^T -> fused_scale_tril_softmax -> PV
-// chain (reference libai/layers/attention.py:211-253) with a
-// single online-softmax kernel: the S x S score matrix never touches HBM.
+// chain (reference libai/layers/attention.py:211-253) with
+// online-softmax kernels: the S x S score matrix never touches HBM in either
+// direction.
 //
-// Structure (v1, per the CDNA4 guide's 8-warp attention recipe, simplified):
-//   * workgroup = 4 waves x 64 = 256 threads; each wave owns QB=32 query rows
-//     -> 128 q rows per workgroup; K/V tiles of KVB=64 keys staged in LDS and
-//     shared by the 4 waves.
-//   * QK^T via v_mfma_f32_32x32x16_bf16 with SWAPPED operands
-//     (S[kv][q] = K x Q^T), so each lane holds score slices of ONE query
-//     (col = lane&31) and the online-softmax row reduce is 16 in-lane f32 ops
-//     + one shfl_xor(32).
-//   * P -> bf16 B-fragments in-register via pack + permlane32_swap (guide
-//     T12); PV uses V B-fragments delivered transposed by ds_read_b64_tr_b16
-//     from a [kv/4][d/16][4][16]-blocked LDS image (guide T10; mapping
-//     verified empirically in csrc/tools/frag_probe.cpp).
-//   * K LDS tile is XOR-swizzled (byte ^= (row&15)<<4) for conflict-free
-//     ds_read_b128 A-fragments (guide T2/G4).
-//   * causal masking is an in-kernel predicate; attention dropout is philox
-//     keyed on (batch*head, q, kv) so the backward regenerates the mask.
+// Shared machinery (fragment layouts verified empirically by
+// csrc/tools/frag_probe.cpp on MI355X):
+//   * v_mfma_f32_32x32x16_bf16 everywhere.  A-frag: lane i=l&31, k=(l>>5)*8+j.
+//     B-frag: lane j=l&31, same k split.  D-frag: lane col j=l&31,
+//     row=(r&3)+8*(r>>2)+4*(l>>5).
+//   * row images: [ROWS][D] bf16 in LDS, 16B chunks XOR-swizzled by
+//     (row&15)<<4 -> conflict-free ds_read_b128 row fragments (guide T2/G4).
+//   * tr images: [r/4][d/16][4][16]-blocked bf16 in LDS; ds_read_b64_tr_b16
+//     delivers transposed fragments (lane gets column d, 4 rows per read).
+//   * D-layout f32 regs -> contiguous-k bf16 fragments via pack +
+//     permlane32_swap (guide T12).
+//   * attention dropout = per-element splitmix hash on (bh, q, kv): fwd and
+//     both bwd kernels regenerate identical masks in their own lane layouts.
 //
-// Saved for backward: O and per-row logsumexp (lse = m + log(l)).
+// Forward saves O and per-row logsumexp.  Backward (flash-attention-2 style):
+//   Drow = rowsum(dO*O);  P = exp(scale*S - lse)
+//   dV = Pd^T dO;  dPd = dO V^T;  dS = scale * P (dPd*mask - Drow)
+//   dK = dS^T Q;   dQ = dS K
+// bwd_kv computes dK,dV (grid over kv tiles); bwd_q computes dQ (grid over
+// q tiles); both recompute S on the fly.
 #include "common.h"
 
 namespace {
@@ -31,18 +35,11 @@ typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 typedef __bf16 bf16x4_t __attribute__((ext_vector_type(4)));
 typedef float f32x16_t __attribute__((ext_vector_type(16)));
 
-#define QB 32    // q rows per wave
+#define QB 32    // rows per wave
 #define NW 4     // waves per workgroup
 #define QBLK (QB * NW)
-#define KVB 64   // kv rows per LDS tile
+#define KVB 64   // kv rows per LDS tile (forward)
 
-// LDS K tile: [KVB][D] row-major bf16, 16-byte chunks XOR-swizzled by row.
-__device__ __forceinline__ int k_lds_off(int row, int col /*bf16 units*/) {
-  int byte = row * 128 /*64*2B... D template handles*/ + col * 2;
-  return byte;  // swizzle applied by caller (needs D)
-}
-
-// pack two f32 into one u32 of two bf16 (compiler emits v_cvt_pk_bf16_f32)
 __device__ __forceinline__ uint32_t pack_bf16x2(float lo, float hi) {
   union {
     uint32_t u;
@@ -53,46 +50,145 @@ __device__ __forceinline__ uint32_t pack_bf16x2(float lo, float hi) {
   return r.u;
 }
 
-template <int D>  // head dim: 64 or 128
+// D-layout regs (8 consecutive: rows {0..3}+4hi and {8..11}+4hi of a 16-row
+// chunk) -> one contiguous-k bf16x8 fragment.
+__device__ __forceinline__ bf16x8_t repack_chunk(const float* sv8) {
+  uint32_t x0 = pack_bf16x2(sv8[0], sv8[1]);
+  uint32_t z0 = pack_bf16x2(sv8[2], sv8[3]);
+  uint32_t y0 = pack_bf16x2(sv8[4], sv8[5]);
+  uint32_t w0 = pack_bf16x2(sv8[6], sv8[7]);
+  auto rx = __builtin_amdgcn_permlane32_swap(x0, y0, false, false);
+  auto rz = __builtin_amdgcn_permlane32_swap(z0, w0, false, false);
+  uint32_t words[4] = {(uint32_t)rx[0], (uint32_t)rz[0], (uint32_t)rx[1],
+                       (uint32_t)rz[1]};
+  bf16x8_t out;
+  memcpy(&out, words, 16);
+  return out;
+}
+
+// ---- LDS images -----------------------------------------------------------
+
+// row image write: 16B chunk of row `row` at col chunk col8*8, swizzled
+template <int D>
+__device__ __forceinline__ void row_img_write(bf16_t* lds, int row, int col8,
+                                              bf16x8_t val) {
+  int byte = (row * D + col8 * 8) * 2;
+  byte ^= (row & 15) << 4;
+  *(bf16x8_t*)((char*)lds + byte) = val;
+}
+
+// row fragment: element [row][c*16 + hi*8 .. +8]
+template <int D>
+__device__ __forceinline__ bf16x8_t row_img_frag(const bf16_t* lds, int row, int c,
+                                                 int hi) {
+  int byte = (row * D + c * 16 + hi * 8) * 2;
+  byte ^= (row & 15) << 4;
+  return *(const bf16x8_t*)((const char*)lds + byte);
+}
+
+// tr image write: 16B chunk (row, 8 cols at d0)
+template <int D>
+__device__ __forceinline__ void tr_img_write(bf16_t* lds, int row, int d0,
+                                             bf16x8_t val) {
+  int idx = (((row >> 2) * (D / 16) + (d0 >> 4)) * 64) + (row & 3) * 16 + (d0 & 15);
+  *(bf16x8_t*)(lds + idx) = val;
+}
+
+// transposed fragment: lane gets [rbase + hi*8 + j][dt*32 + l31] for j=0..7
+template <int D>
+__device__ __forceinline__ bf16x8_t tr_img_frag(const bf16_t* lds, int rbase, int dt,
+                                                int lane) {
+  const int hi = lane >> 5;
+  const int l31 = lane & 31;
+  const int r0 = rbase + hi * 8;
+  const int d0 = dt * 32 + (l31 & ~15);
+  const bf16_t* p0 =
+      lds + (((r0 >> 2) * (D / 16) + (d0 >> 4)) * 64) + (lane & 15) * 4;
+  const bf16_t* p1 =
+      lds + ((((r0 + 4) >> 2) * (D / 16) + (d0 >> 4)) * 64) + (lane & 15) * 4;
+  bf16x4_t a = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4_t*)p0);
+  bf16x4_t b = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4_t*)p1);
+  bf16x8_t out;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    out[j] = a[j];
+    out[4 + j] = b[j];
+  }
+  return out;
+}
+
+// cooperative staging of a [ROWS][D] global tile into row and/or tr images
+template <int ROWS, int D, bool ROW_IMG, bool TR_IMG>
+__device__ __forceinline__ void stage_tile(bf16_t* row_lds, bf16_t* tr_lds,
+                                           const bf16_t* gp, int row0, int bound,
+                                           int64_t stride, int tid) {
+  constexpr int CHUNKS_TOTAL = ROWS * D / 8;
+  constexpr int ITER = CDIV(CHUNKS_TOTAL, 256);
+#pragma unroll
+  for (int cc = 0; cc < ITER; ++cc) {
+    const int flat = tid + cc * 256;
+    if (flat >= CHUNKS_TOTAL) break;
+    const int row = flat / (D / 8);
+    const int col8 = flat % (D / 8);
+    const int gr = row0 + row;
+    bf16x8_t val =
+        (gr < bound) ? *(const bf16x8_t*)(gp + (int64_t)gr * stride + col8 * 8)
+                     : bf16x8_t{};
+    if (ROW_IMG) row_img_write<D>(row_lds, row, col8, val);
+    if (TR_IMG) tr_img_write<D>(tr_lds, row, col8 * 8, val);
+  }
+}
+
+// dropout keep factor for score element (bh, q, kv)
+__device__ __forceinline__ float drop_keep(uint64_t seed, uint64_t bh, int64_t Sq,
+                                           int64_t Sk, int q, int kv, float p,
+                                           float ks) {
+  uint64_t idx = (bh * Sq + q) * Sk + kv;
+  return (u32_to_uniform(rnd_hash(seed, idx)) > p) ? ks : 0.f;
+}
+
+// ===========================================================================
+// forward
+// ===========================================================================
+template <int D>
 __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, bf16_t* __restrict__ o, float* __restrict__ lse,
-    int64_t q_sb, int64_t q_ss, int64_t q_sh,  // elem strides: batch, seq, head
-    int64_t k_sb, int64_t k_ss, int64_t k_sh, int64_t v_sb, int64_t v_ss,
-    int64_t v_sh, int64_t o_sb, int64_t o_ss, int64_t o_sh, int H, int Sq, int Sk,
-    float scale, float p_drop, uint64_t seed, int causal) {
-  constexpr int DT = D / 32;      // 32-wide d tiles (O accum tiles)
-  constexpr int KC = D / 16;      // 16-deep k-chunks per QK^T mfma chain
-  static_assert(D == 64 || D == 128, "head dim 64/128 only");
+    int64_t q_sb, int64_t q_ss, int64_t q_sh, int64_t k_sb, int64_t k_ss,
+    int64_t k_sh, int64_t v_sb, int64_t v_ss, int64_t v_sh, int64_t o_sb,
+    int64_t o_ss, int64_t o_sh, int H, int Sq, int Sk, float scale, float p_drop,
+    uint64_t seed, int causal) {
+  constexpr int DT = D / 32;
+  constexpr int KC = D / 16;
 
   __shared__ __align__(16) bf16_t k_lds[KVB * D];
-  __shared__ __align__(16) bf16_t v_lds[KVB * D];  // [kv/4][d/16][4][16] blocks
+  __shared__ __align__(16) bf16_t v_lds[KVB * D];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
   const int l31 = lane & 31;
-  const int hi = lane >> 5;  // k-half selector in A/B fragments
+  const int hi = lane >> 5;
 
-  const int bh = blockIdx.y;  // b * H + h
+  const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
   const int q_block = blockIdx.x * QBLK;
-  const int q_base = q_block + wave * QB;  // this wave's first q row
+  const int q_base = q_block + wave * QB;
 
   const bf16_t* qp = q + b * q_sb + h * q_sh;
   const bf16_t* kp = k + b * k_sb + h * k_sh;
   const bf16_t* vp = v + b * v_sb + h * v_sh;
 
-  // ---- load this wave's Q B-fragments (lane: q row = l31, k = hi*8+0..7) --
   bf16x8_t qfrag[KC];
   {
-    const int qrow = q_base + l31;
+    const int qrow = min(q_base + l31, Sq - 1);
 #pragma unroll
     for (int c = 0; c < KC; ++c)
       qfrag[c] = *(const bf16x8_t*)(qp + (int64_t)qrow * q_ss + c * 16 + hi * 8);
   }
 
-  // ---- accumulators ----
   f32x16_t oacc[DT];
 #pragma unroll
   for (int t = 0; t < DT; ++t) oacc[t] = f32x16_t{};
@@ -100,72 +196,40 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
 
   const int kv_end = causal ? min(Sk, q_block + QBLK) : Sk;
   const int n_tiles = CDIV(kv_end, KVB);
-  const float keep_scale = (p_drop > 0.f) ? 1.0f / (1.0f - p_drop) : 1.0f;
+  const float ks = (p_drop > 0.f) ? 1.0f / (1.0f - p_drop) : 1.0f;
+  const int qg = q_base + l31;
 
   for (int tile = 0; tile < n_tiles; ++tile) {
     const int kv0 = tile * KVB;
-    // ---- stage K and V tiles (all 256 threads) ----
-    // thread t: row = t>>2 (+32 for second half when 256*16B < tile bytes)
     __syncthreads();
-    {
-      // KVB * D bf16 = KVB*D*2 bytes; each thread writes (KVB*D/8)/256 chunks
-      constexpr int CHUNKS = (KVB * D / 8) / 256;
-#pragma unroll
-      for (int cc = 0; cc < CHUNKS; ++cc) {
-        const int flat = tid + cc * 256;         // 16B chunk index
-        const int row = flat / (D / 8);          // kv row
-        const int col8 = flat % (D / 8);         // 8-elem col chunk
-        const int kvr = kv0 + row;
-        bf16x8_t kv8 = (kvr < Sk)
-                           ? *(const bf16x8_t*)(kp + (int64_t)kvr * k_ss + col8 * 8)
-                           : bf16x8_t{};
-        // K: row-major with (row&15)<<4 byte XOR swizzle
-        int kbyte = (row * D + col8 * 8) * 2;
-        kbyte ^= (row & 15) << 4;
-        *(bf16x8_t*)((char*)k_lds + kbyte) = kv8;
-        bf16x8_t vv8 = (kvr < Sk)
-                           ? *(const bf16x8_t*)(vp + (int64_t)kvr * v_ss + col8 * 8)
-                           : bf16x8_t{};
-        // V: [kv>>2][d>>4][kv&3][d&15] blocked image (for tr16 reads)
-        const int d0 = col8 * 8;
-        int vidx = (((row >> 2) * (D / 16) + (d0 >> 4)) * 64) + (row & 3) * 16 +
-                   (d0 & 15);
-        *(bf16x8_t*)(v_lds + vidx) = vv8;
-      }
-    }
+    stage_tile<KVB, D, true, false>(k_lds, nullptr, kp, kv0, Sk, k_ss, tid);
+    stage_tile<KVB, D, false, true>(nullptr, v_lds, vp, kv0, Sk, v_ss, tid);
     __syncthreads();
 
-    if (causal && kv0 > q_base + QB - 1) continue;  // tile beyond this wave
+    if (causal && kv0 > q_base + QB - 1) continue;
 
-    // ---- S = K x Q^T : two 32x32 tiles (kv halves) ----
+    // S = K x Q^T : two 32x32 tiles
     f32x16_t s0{}, s1{};
 #pragma unroll
     for (int c = 0; c < KC; ++c) {
-      // A-frag: K[kv = half*32 + l31][hs = c*16 + hi*8 + 0..7]
-      int byte0 = ((l31)*D + c * 16 + hi * 8) * 2;
-      byte0 ^= (l31 & 15) << 4;
-      bf16x8_t ka = *(const bf16x8_t*)((char*)k_lds + byte0);
+      bf16x8_t ka = row_img_frag<D>(k_lds, l31, c, hi);
       s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[c], s0, 0, 0, 0);
-      int byte1 = ((l31 + 32) * D + c * 16 + hi * 8) * 2;
-      byte1 ^= ((l31 + 32) & 15) << 4;
-      bf16x8_t kb = *(const bf16x8_t*)((char*)k_lds + byte1);
+      bf16x8_t kb = row_img_frag<D>(k_lds, l31 + 32, c, hi);
       s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb, qfrag[c], s1, 0, 0, 0);
     }
 
-    // ---- online softmax (lane owns query q_base + l31) ----
-    const int qg = q_base + l31;  // this lane's global q row
     float sv[2][16];
     float pmax = -3.0e38f;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int kva = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
       int kvb = kva + 32;
-      float a = s0[r] * scale, bvl = s1[r] * scale;
+      float a = s0[r] * scale, bb = s1[r] * scale;
       if (kva >= Sk || (causal && kva > qg)) a = -3.0e38f;
-      if (kvb >= Sk || (causal && kvb > qg)) bvl = -3.0e38f;
+      if (kvb >= Sk || (causal && kvb > qg)) bb = -3.0e38f;
       sv[0][r] = a;
-      sv[1][r] = bvl;
-      pmax = fmaxf(pmax, fmaxf(a, bvl));
+      sv[1][r] = bb;
+      pmax = fmaxf(pmax, fmaxf(a, bb));
     }
     pmax = fmaxf(pmax, __shfl_xor(pmax, 32));
 
@@ -183,86 +247,31 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
       }
     lsum += __shfl_xor(lsum, 32);
     l_run = l_run * alpha + lsum;
-    // rescale O accumulators (each lane's accum belongs to its own query)
 #pragma unroll
     for (int t = 0; t < DT; ++t)
 #pragma unroll
       for (int r = 0; r < 16; ++r) oacc[t][r] *= alpha;
 
-    // ---- dropout on P (philox on (bh, q, kv)) ----
     if (p_drop > 0.f) {
 #pragma unroll
-      for (int t = 0; t < 2; ++t) {
+      for (int t = 0; t < 2; ++t)
 #pragma unroll
-        for (int r4 = 0; r4 < 4; ++r4) {
-          // 4 consecutive regs share a philox call; counter from first kv
-          uint32_t rnd[4];
-          int kvf = kv0 + t * 32 + (r4 * 8) + 4 * hi;  // kv of reg r4*4
-          // unique counter per (bh, q, kv group of 4): kv pattern within a
-          // group r: (r&3) consecutive -> use kv base of the 4-reg run
-          uint64_t ctr =
-              (((uint64_t)bh * Sq + qg) * (uint64_t)Sk + kvf) >> 2;
-          philox4(seed, ctr, rnd);
-#pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            float kp_ = (u32_to_uniform(rnd[j]) > p_drop) ? keep_scale : 0.f;
-            sv[t][r4 * 4 + j] *= kp_;
-          }
+        for (int r = 0; r < 16; ++r) {
+          int kv = kv0 + t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          sv[t][r] *= drop_keep(seed, bh, Sq, Sk, qg, kv, p_drop, ks);
         }
-      }
     }
 
-    // ---- P -> bf16 B-fragments via pack + permlane32_swap ----
-    // chunk c covers kv = kv0 + t*32 + c16*16; B word w: lanes<32 kv pairs
-    // (2w, 2w+1), lanes>=32 kv pairs (8+2w, 8+2w+1) of the chunk.
+    // PV: O^T[d][q] += V^T x P
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
-      uint32_t pw[8];  // two chunks x 4 words
 #pragma unroll
       for (int c16 = 0; c16 < 2; ++c16) {
-        // regs for kv offsets {0,1,2,3} of this chunk: r = c16*8 + {0..3}
-        // regs for kv offsets {8..11}: r = c16*8 + {4..7}
-        uint32_t x0 = pack_bf16x2(sv[t][c16 * 8 + 0], sv[t][c16 * 8 + 1]);
-        uint32_t z0 = pack_bf16x2(sv[t][c16 * 8 + 2], sv[t][c16 * 8 + 3]);
-        uint32_t y0 = pack_bf16x2(sv[t][c16 * 8 + 4], sv[t][c16 * 8 + 5]);
-        uint32_t w0 = pack_bf16x2(sv[t][c16 * 8 + 6], sv[t][c16 * 8 + 7]);
-        auto rx = __builtin_amdgcn_permlane32_swap(x0, y0, false, false);
-        auto rz = __builtin_amdgcn_permlane32_swap(z0, w0, false, false);
-        pw[c16 * 4 + 0] = rx[0];
-        pw[c16 * 4 + 1] = rz[0];
-        pw[c16 * 4 + 2] = rx[1];
-        pw[c16 * 4 + 3] = rz[1];
-      }
-      // ---- PV: O[q][d] += P x V, B-frag of V via tr16 reads ----
-#pragma unroll
-      for (int c16 = 0; c16 < 2; ++c16) {
-        bf16x8_t pfrag;
-        memcpy(&pfrag, &pw[c16 * 4], 16);
-        const int kvc = t * 32 + c16 * 16;  // chunk kv base within tile
+        bf16x8_t pfrag = repack_chunk(&sv[t][c16 * 8]);
+        const int kvc = t * 32 + c16 * 16;
 #pragma unroll
         for (int dt = 0; dt < DT; ++dt) {
-          // lane needs V[kv = kvc + hi*8 + j'][d = dt*32 + l31]
-          // image block of (kv row group, d group): two tr reads (4 kv each)
-          bf16x4_t va, vb2;
-          {
-            const int kvr = kvc + hi * 8;          // rows kvr..kvr+3
-            const int d0 = dt * 32 + (l31 & ~15);  // 16-d group
-            int base = (((kvr >> 2) * (D / 16) + (d0 >> 4)) * 64);
-            va = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                (__attribute__((address_space(3))) bf16x4_t*)(v_lds + base +
-                                                              (lane & 15) * 4));
-            const int kvr2 = kvr + 4;
-            int base2 = (((kvr2 >> 2) * (D / 16) + (d0 >> 4)) * 64);
-            vb2 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                (__attribute__((address_space(3))) bf16x4_t*)(v_lds + base2 +
-                                                              (lane & 15) * 4));
-          }
-          bf16x8_t vfrag;
-#pragma unroll
-          for (int j = 0; j < 4; ++j) {
-            vfrag[j] = va[j];
-            vfrag[4 + j] = vb2[j];
-          }
+          bf16x8_t vfrag = tr_img_frag<D>(v_lds, kvc, dt, lane);
           oacc[dt] =
               __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag, pfrag, oacc[dt], 0, 0, 0);
         }
@@ -270,25 +279,325 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
     }
   }
 
-  // ---- epilogue: O = oacc / l, write [q][d]; lse = m + log(l) ----
-  const int qg = q_base + l31;
   const float inv_l = (l_run > 0.f) ? 1.0f / l_run : 0.f;
   bf16_t* op = o + b * o_sb + h * o_sh + (int64_t)qg * o_ss;
   if (qg < Sq) {
 #pragma unroll
-    for (int dt = 0; dt < DT; ++dt) {
+    for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
       for (int r4 = 0; r4 < 4; ++r4) {
-        // regs r4*4..r4*4+3 are d = dt*32 + {0..3} + 8*r4 + 4*hi  (contiguous)
-        u16x4 pack;
+        u16x4 pk;
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
-          pack[j] = f2bf(oacc[dt][r4 * 4 + j] * inv_l);
-        *(u16x4*)(op + dt * 32 + 8 * r4 + 4 * hi) = pack;
+        for (int j = 0; j < 4; ++j) pk[j] = f2bf(oacc[dt][r4 * 4 + j] * inv_l);
+        *(u16x4*)(op + dt * 32 + 8 * r4 + 4 * hi) = pk;
       }
-    }
     if (hi == 0 && lse != nullptr)
       lse[((int64_t)bh * Sq) + qg] = m_run + __logf(l_run > 0.f ? l_run : 1.f);
+  }
+}
+
+// ===========================================================================
+// Drow = rowsum(dO * O): one wave per 8 rows (8 lanes x 8 elems for D=64)
+// out layout [B*H, Sq] (matches lse); inputs [B, Sq, H, D] strided
+// ===========================================================================
+template <int D>
+__global__ void rowdot_kernel(const bf16_t* __restrict__ dout,
+                              const bf16_t* __restrict__ o, float* __restrict__ drow,
+                              int64_t sb, int64_t ss, int64_t sh, int H, int Sq,
+                              int64_t nrows) {
+  constexpr int LPR = D / 8;   // lanes per row
+  constexpr int RPW = 64 / LPR;  // rows per wave
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / 64;
+  const int lane = threadIdx.x & 63;
+  const int64_t row = wave_id * RPW + lane / LPR;  // flat (b*Sq+q)*H+h
+  if (row >= nrows) return;
+  const int64_t BSH = (int64_t)Sq * H;
+  const int64_t b = row / BSH;
+  const int64_t rem = row % BSH;
+  const int64_t qi = rem / H;
+  const int64_t h = rem % H;
+  const bf16_t* dp = dout + b * sb + qi * ss + h * sh + (lane % LPR) * 8;
+  const bf16_t* op = o + b * sb + qi * ss + h * sh + (lane % LPR) * 8;
+  u16x8 dv = *(const u16x8*)dp;
+  u16x8 ov = *(const u16x8*)op;
+  float s = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) s += bf2f(dv[j]) * bf2f(ov[j]);
+#pragma unroll
+  for (int off = LPR / 2; off > 0; off >>= 1) s += __shfl_xor(s, off, 64);
+  if (lane % LPR == 0) drow[(b * H + h) * Sq + qi] = s;
+}
+
+// ===========================================================================
+// backward over kv tiles: dK, dV   (block = 128 kv rows, wave owns 32)
+// ===========================================================================
+template <int D>
+__global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    bf16_t* __restrict__ dk, bf16_t* __restrict__ dv, int64_t q_sb, int64_t q_ss,
+    int64_t q_sh, int64_t k_sb, int64_t k_ss, int64_t k_sh, int64_t v_sb,
+    int64_t v_ss, int64_t v_sh, int64_t do_sb, int64_t do_ss, int64_t do_sh,
+    int64_t dk_sb, int64_t dk_ss, int64_t dk_sh, int64_t dv_sb, int64_t dv_ss,
+    int64_t dv_sh, int H, int Sq, int Sk, float scale, float p_drop, uint64_t seed,
+    int causal) {
+  constexpr int DT = D / 32;
+  constexpr int KC = D / 16;
+
+  __shared__ __align__(16) bf16_t q_row[QB * D];
+  __shared__ __align__(16) bf16_t q_tr[QB * D];
+  __shared__ __align__(16) bf16_t do_row[QB * D];
+  __shared__ __align__(16) bf16_t do_tr[QB * D];
+  __shared__ float lse_lds[QB];
+  __shared__ float drow_lds[QB];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
+
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int kv_block = blockIdx.x * QBLK;   // 128 kv rows per block
+  const int kv_base = kv_block + wave * QB; // this wave's 32 kv rows
+  const int kvg = kv_base + l31;            // lane's kv row
+
+  const bf16_t* qp = q + b * q_sb + h * q_sh;
+  const bf16_t* kp = k + b * k_sb + h * k_sh;
+  const bf16_t* vp = v + b * v_sb + h * v_sh;
+  const bf16_t* dop = dout + b * do_sb + h * do_sh;
+
+  // persistent per-lane K and V rows (B-fragments: lane j = kv)
+  bf16x8_t kf[KC], vf[KC];
+  {
+    const int kvr = min(kvg, Sk - 1);
+#pragma unroll
+    for (int c = 0; c < KC; ++c) {
+      kf[c] = *(const bf16x8_t*)(kp + (int64_t)kvr * k_ss + c * 16 + hi * 8);
+      vf[c] = *(const bf16x8_t*)(vp + (int64_t)kvr * v_ss + c * 16 + hi * 8);
+    }
+  }
+
+  f32x16_t dk_acc[DT], dv_acc[DT];
+#pragma unroll
+  for (int t = 0; t < DT; ++t) {
+    dk_acc[t] = f32x16_t{};
+    dv_acc[t] = f32x16_t{};
+  }
+
+  const float ks = (p_drop > 0.f) ? 1.0f / (1.0f - p_drop) : 1.0f;
+  const int qt_start = causal ? (kv_block / QB) : 0;
+  const int n_qtiles = CDIV(Sq, QB);
+
+  for (int qt = qt_start; qt < n_qtiles; ++qt) {
+    const int q0 = qt * QB;
+    __syncthreads();
+    stage_tile<QB, D, true, true>(q_row, q_tr, qp, q0, Sq, q_ss, tid);
+    stage_tile<QB, D, true, true>(do_row, do_tr, dop, q0, Sq, do_ss, tid);
+    if (tid < QB) {
+      int qr = min(q0 + tid, Sq - 1);
+      lse_lds[tid] = lse[(int64_t)bh * Sq + qr];
+      drow_lds[tid] = drow[(int64_t)bh * Sq + qr];
+    }
+    __syncthreads();
+
+    if (causal && q0 + QB - 1 < kv_base) continue;  // tile entirely above diag
+
+    // S[q][kv] = Q x K^T  (lane col = kv)
+    f32x16_t s{};
+    f32x16_t dpd{};
+#pragma unroll
+    for (int c = 0; c < KC; ++c) {
+      bf16x8_t qa = row_img_frag<D>(q_row, l31, c, hi);
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kf[c], s, 0, 0, 0);
+      bf16x8_t da = row_img_frag<D>(do_row, l31, c, hi);
+      dpd = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vf[c], dpd, 0, 0, 0);
+    }
+
+    // per-reg: q = q0 + pattern(r); kv = lane's kvg
+    float pd[16], ds[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const bool valid = qrow < Sq && kvg < Sk && (!causal || qrow >= kvg);
+      float p = 0.f;
+      if (valid) p = __expf(s[r] * scale - lse_lds[qrow - q0]);
+      float keep = (p_drop > 0.f && valid)
+                       ? drop_keep(seed, bh, Sq, Sk, qrow, kvg, p_drop, ks)
+                       : (valid ? 1.f : 0.f);
+      pd[r] = p * keep;
+      ds[r] = scale * p * (dpd[r] * keep - drow_lds[qrow - q0]);
+      if (!valid) ds[r] = 0.f;
+    }
+
+    // dV^T[d][kv] += dO^T x Pd ;  dK^T[d][kv] += Q^T x dS
+#pragma unroll
+    for (int c16 = 0; c16 < 2; ++c16) {
+      bf16x8_t pdf = repack_chunk(&pd[c16 * 8]);
+      bf16x8_t dsf = repack_chunk(&ds[c16 * 8]);
+      const int qc = c16 * 16;
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        bf16x8_t dof = tr_img_frag<D>(do_tr, qc, dt, lane);
+        dv_acc[dt] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, pdf, dv_acc[dt], 0, 0, 0);
+        bf16x8_t qtf = tr_img_frag<D>(q_tr, qc, dt, lane);
+        dk_acc[dt] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(qtf, dsf, dk_acc[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  // write dK, dV rows (lane owns kv row kvg; d in 4-element runs)
+  if (kvg < Sk) {
+    bf16_t* dkp = dk + b * dk_sb + h * dk_sh + (int64_t)kvg * dk_ss;
+    bf16_t* dvp = dv + b * dv_sb + h * dv_sh + (int64_t)kvg * dv_ss;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+      for (int r4 = 0; r4 < 4; ++r4) {
+        u16x4 a, c;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          a[j] = f2bf(dk_acc[dt][r4 * 4 + j]);
+          c[j] = f2bf(dv_acc[dt][r4 * 4 + j]);
+        }
+        *(u16x4*)(dkp + dt * 32 + 8 * r4 + 4 * hi) = a;
+        *(u16x4*)(dvp + dt * 32 + 8 * r4 + 4 * hi) = c;
+      }
+  }
+}
+
+// ===========================================================================
+// backward over q tiles: dQ   (block = 128 q rows, wave owns 32; fwd-like)
+// ===========================================================================
+template <int D>
+__global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    bf16_t* __restrict__ dq, int64_t q_sb, int64_t q_ss, int64_t q_sh, int64_t k_sb,
+    int64_t k_ss, int64_t k_sh, int64_t v_sb, int64_t v_ss, int64_t v_sh,
+    int64_t do_sb, int64_t do_ss, int64_t do_sh, int64_t dq_sb, int64_t dq_ss,
+    int64_t dq_sh, int H, int Sq, int Sk, float scale, float p_drop, uint64_t seed,
+    int causal) {
+  constexpr int DT = D / 32;
+  constexpr int KC = D / 16;
+
+  __shared__ __align__(16) bf16_t k_row[QB * D];
+  __shared__ __align__(16) bf16_t k_tr[QB * D];
+  __shared__ __align__(16) bf16_t v_row[QB * D];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
+
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int q_block = blockIdx.x * QBLK;
+  const int q_base = q_block + wave * QB;
+  const int qg = q_base + l31;
+
+  const bf16_t* qp = q + b * q_sb + h * q_sh;
+  const bf16_t* kp = k + b * k_sb + h * k_sh;
+  const bf16_t* vp = v + b * v_sb + h * v_sh;
+  const bf16_t* dop = dout + b * do_sb + h * do_sh;
+
+  // persistent per-lane Q and dO rows (B-fragments: lane j = q)
+  bf16x8_t qf[KC], dof[KC];
+  {
+    const int qr = min(qg, Sq - 1);
+#pragma unroll
+    for (int c = 0; c < KC; ++c) {
+      qf[c] = *(const bf16x8_t*)(qp + (int64_t)qr * q_ss + c * 16 + hi * 8);
+      dof[c] = *(const bf16x8_t*)(dop + (int64_t)qr * do_ss + c * 16 + hi * 8);
+    }
+  }
+  const float lse_lane = lse[(int64_t)bh * Sq + min(qg, Sq - 1)];
+  const float drow_lane = drow[(int64_t)bh * Sq + min(qg, Sq - 1)];
+
+  f32x16_t dq_acc[DT];
+#pragma unroll
+  for (int t = 0; t < DT; ++t) dq_acc[t] = f32x16_t{};
+
+  const float ks = (p_drop > 0.f) ? 1.0f / (1.0f - p_drop) : 1.0f;
+  const int kv_end = causal ? min(Sk, q_block + QBLK) : Sk;
+  const int n_tiles = CDIV(kv_end, QB);
+
+  for (int tile = 0; tile < n_tiles; ++tile) {
+    const int kv0 = tile * QB;
+    __syncthreads();
+    stage_tile<QB, D, true, true>(k_row, k_tr, kp, kv0, Sk, k_ss, tid);
+    stage_tile<QB, D, true, false>(v_row, nullptr, vp, kv0, Sk, v_ss, tid);
+    __syncthreads();
+
+    if (causal && kv0 > q_base + QB - 1) continue;
+
+    // S^T[kv][q] = K x Q^T ;  dPd^T[kv][q] = V x dO^T   (lane col = q)
+    f32x16_t st{};
+    f32x16_t dpdt{};
+#pragma unroll
+    for (int c = 0; c < KC; ++c) {
+      bf16x8_t ka = row_img_frag<D>(k_row, l31, c, hi);
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qf[c], st, 0, 0, 0);
+      bf16x8_t va = row_img_frag<D>(v_row, l31, c, hi);
+      dpdt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dof[c], dpdt, 0, 0, 0);
+    }
+
+    float ds[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const bool valid = kv < Sk && qg < Sq && (!causal || kv <= qg);
+      float p = valid ? __expf(st[r] * scale - lse_lane) : 0.f;
+      float keep = (p_drop > 0.f && valid)
+                       ? drop_keep(seed, bh, Sq, Sk, qg, kv, p_drop, ks)
+                       : (valid ? 1.f : 0.f);
+      ds[r] = valid ? scale * p * (dpdt[r] * keep - drow_lane) : 0.f;
+    }
+
+    // dQ^T[d][q] += K^T x dS^T
+#pragma unroll
+    for (int c16 = 0; c16 < 2; ++c16) {
+      bf16x8_t dsf = repack_chunk(&ds[c16 * 8]);
+      const int kvc = c16 * 16;
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        bf16x8_t ktf = tr_img_frag<D>(k_tr, kvc, dt, lane);
+        dq_acc[dt] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(ktf, dsf, dq_acc[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  if (qg < Sq) {
+    bf16_t* dqp = dq + b * dq_sb + h * dq_sh + (int64_t)qg * dq_ss;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+      for (int r4 = 0; r4 < 4; ++r4) {
+        u16x4 a;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) a[j] = f2bf(dq_acc[dt][r4 * 4 + j]);
+        *(u16x4*)(dqp + dt * 32 + 8 * r4 + 4 * hi) = a;
+      }
+  }
+}
+
+// dropout-mask application for external recompute paths / tests
+template <class E>
+__global__ void attn_dropout_apply_kernel(typename E::T* __restrict__ x, int64_t BH,
+                                          int64_t Sq, int64_t Sk, float p,
+                                          float keep_scale, uint64_t seed) {
+  const int64_t total = BH * Sq * Sk;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float kp = (u32_to_uniform(rnd_hash(seed, (uint64_t)i)) > p) ? keep_scale : 0.f;
+    x[i] = E::from_f(E::to_f(x[i]) * kp);
   }
 }
 
@@ -315,37 +624,61 @@ extern "C" void flash_fwd_bf16(const void* q, const void* k, const void* v, void
         scale, p_drop, seed, causal);
 }
 
-// ---------------------------------------------------------------------------
-// apply the SAME philox attention-dropout mask the flash forward used, to an
-// arbitrary [B*H, Sq, Sk] tensor (used by the recompute backward):
-//   x *= keep(seed, bh, q, kv) / (1-p)
-// ---------------------------------------------------------------------------
-namespace {
-template <class E>
-__global__ void attn_dropout_apply_kernel(typename E::T* __restrict__ x, int64_t BH,
-                                          int64_t Sq, int64_t Sk, float p,
-                                          float keep_scale, uint64_t seed) {
-  const int64_t total4 = BH * Sq * (Sk >> 2);
-  for (int64_t i4 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i4 < total4;
-       i4 += (int64_t)gridDim.x * blockDim.x) {
-    uint32_t rnd[4];
-    philox4(seed, (uint64_t)i4, rnd);
-    typename E::T* xp = x + i4 * 4;
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      float kp = (u32_to_uniform(rnd[j]) > p) ? keep_scale : 0.f;
-      xp[j] = E::from_f(E::to_f(xp[j]) * kp);
-    }
+extern "C" void flash_bwd_bf16(
+    const void* q, const void* k, const void* v, const void* o, const void* dout,
+    const float* lse, float* drow_ws, void* dq, void* dk, void* dv, int64_t q_sb,
+    int64_t q_ss, int64_t q_sh, int64_t k_sb, int64_t k_ss, int64_t k_sh,
+    int64_t v_sb, int64_t v_ss, int64_t v_sh, int64_t o_sb, int64_t o_ss,
+    int64_t o_sh, int64_t do_sb, int64_t do_ss, int64_t do_sh, int64_t dq_sb,
+    int64_t dq_ss, int64_t dq_sh, int64_t dk_sb, int64_t dk_ss, int64_t dk_sh,
+    int64_t dv_sb, int64_t dv_ss, int64_t dv_sh, int B, int H, int Sq, int Sk, int D,
+    float scale, float p_drop, uint64_t seed, int causal, hipStream_t stream) {
+  // Drow = rowsum(dO * O)  (dO and O share layout; use dO's strides for both:
+  // the wrapper guarantees o was allocated with the same layout)
+  {
+    int rpw = 64 / (D / 8);
+    int64_t rows = (int64_t)B * Sq * H;
+    int64_t waves = CDIV(rows, rpw);
+    uint32_t blocks = (uint32_t)CDIV(waves * 64, 256);
+    if (D == 64)
+      rowdot_kernel<64><<<dim3(blocks), 256, 0, stream>>>(
+          (const bf16_t*)dout, (const bf16_t*)o, drow_ws, do_sb, do_ss, do_sh, H, Sq,
+          rows);
+    else
+      rowdot_kernel<128><<<dim3(blocks), 256, 0, stream>>>(
+          (const bf16_t*)dout, (const bf16_t*)o, drow_ws, do_sb, do_ss, do_sh, H, Sq,
+          rows);
+  }
+#define BWD_ARGS_KV                                                                  \
+  (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (const bf16_t*)dout, lse,    \
+      drow_ws, (bf16_t*)dk, (bf16_t*)dv, q_sb, q_ss, q_sh, k_sb, k_ss, k_sh, v_sb,   \
+      v_ss, v_sh, do_sb, do_ss, do_sh, dk_sb, dk_ss, dk_sh, dv_sb, dv_ss, dv_sh, H,  \
+      Sq, Sk, scale, p_drop, seed, causal
+#define BWD_ARGS_Q                                                                   \
+  (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (const bf16_t*)dout, lse,    \
+      drow_ws, (bf16_t*)dq, q_sb, q_ss, q_sh, k_sb, k_ss, k_sh, v_sb, v_ss, v_sh,    \
+      do_sb, do_ss, do_sh, dq_sb, dq_ss, dq_sh, H, Sq, Sk, scale, p_drop, seed,      \
+      causal
+  dim3 block(256);
+  if (D == 64) {
+    flash_bwd_kv_kernel<64>
+        <<<dim3(CDIV(Sk, QBLK), B * H), block, 0, stream>>>(BWD_ARGS_KV);
+    flash_bwd_q_kernel<64>
+        <<<dim3(CDIV(Sq, QBLK), B * H), block, 0, stream>>>(BWD_ARGS_Q);
+  } else if (D == 128) {
+    flash_bwd_kv_kernel<128>
+        <<<dim3(CDIV(Sk, QBLK), B * H), block, 0, stream>>>(BWD_ARGS_KV);
+    flash_bwd_q_kernel<128>
+        <<<dim3(CDIV(Sq, QBLK), B * H), block, 0, stream>>>(BWD_ARGS_Q);
   }
 }
-}  // namespace
 
 #define ATTN_DROP_LAUNCHER(SUFF, ETYPE)                                            \
   extern "C" void attn_dropout_apply_##SUFF(void* x, int64_t BH, int64_t Sq,       \
                                             int64_t Sk, float p, uint64_t seed,    \
                                             hipStream_t stream) {                  \
-    int64_t n4 = BH * Sq * (Sk >> 2);                                              \
-    int64_t g = CDIV(n4, 256);                                                     \
+    int64_t n = BH * Sq * Sk;                                                      \
+    int64_t g = CDIV(n, 256);                                                      \
     if (g > 4096) g = 4096;                                                        \
     attn_dropout_apply_kernel<ETYPE><<<dim3((uint32_t)g), dim3(256), 0, stream>>>( \
         (ETYPE::T*)x, BH, Sq, Sk, p, 1.0f / (1.0f - p), seed);                     \

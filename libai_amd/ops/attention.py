@@ -43,45 +43,10 @@ class _FlashAttnFn(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         scale, p_drop, seed, causal = ctx.meta
-        B, Sq, H, D = q.shape
-        Sk = k.shape[1]
-
-        # [B, H, S, D] views for the batched GEMMs
-        qh = q.permute(0, 2, 1, 3)
-        kh = k.permute(0, 2, 1, 3)
-        vh = v.permute(0, 2, 1, 3)
-        doh = do.permute(0, 2, 1, 3)
-
-        s = torch.matmul(qh, kh.transpose(-1, -2))  # [B,H,Sq,Sk] bf16
-        p = torch.exp(s.float() * scale - lse.unsqueeze(-1))
-        if causal:
-            cm = torch.ones(Sq, Sk, dtype=torch.bool, device=q.device).tril_(Sk - Sq)
-            p = p.masked_fill(~cm, 0.0)
-        p = p.to(q.dtype)
-        if p_drop > 0:
-            pd = p.clone()
-            ext().attn_dropout_apply(pd, Sq, Sk, p_drop, seed)
-        else:
-            pd = p
-
-        dv = torch.matmul(pd.transpose(-1, -2), doh)  # [B,H,Sk,D]
-        dpd = torch.matmul(doh, vh.transpose(-1, -2))  # [B,H,Sq,Sk]
-        if p_drop > 0:
-            ext().attn_dropout_apply(dpd, Sq, Sk, p_drop, seed)
-        # rowsum(dP*P) == rowsum(dO*O) (holds with dropout; see flash-attn)
-        Drow = (doh.float() * o.permute(0, 2, 1, 3).float()).sum(-1)  # [B,H,Sq]
-        ds = (p.float() * (dpd.float() - Drow.unsqueeze(-1)) * scale).to(q.dtype)
-        dq = torch.matmul(ds, kh)  # [B,H,Sq,D]
-        dk = torch.matmul(ds.transpose(-1, -2), qh)  # [B,H,Sk,D]
-
-        return (
-            dq.permute(0, 2, 1, 3),
-            dk.permute(0, 2, 1, 3),
-            dv.permute(0, 2, 1, 3),
-            None,
-            None,
-            None,
+        dq, dk, dv = ext().flash_bwd(
+            q, k, v, o, do.contiguous(), lse, scale, p_drop, seed, causal, None
         )
+        return dq, dk, dv, None, None, None
 
 
 def flash_attention(q, k, v, scale, p_drop=0.0, causal=True, training=True):
