@@ -44,8 +44,27 @@ def parse_args():
     return p.parse_args()
 
 
+def _enable_tuned_gemms():
+    """Load the committed hipBLASLt TunableOp algo table (gfx950) if present:
+    ~3% on the 345M step, no tuning cost at runtime."""
+    table = os.path.join(os.path.dirname(os.path.abspath(__file__)), "libai_amd",
+                         "data", "tunableop_gfx950.csv")
+    if os.path.exists(table) and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ:
+        import shutil
+        import tempfile
+
+        # torch inserts the device ordinal before .csv; provide all 8
+        d = tempfile.mkdtemp(prefix="tunableop_")
+        for i in range(8):
+            shutil.copy(table, os.path.join(d, f"tunableop{i}.csv"))
+        os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
+        os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
+        os.environ["PYTORCH_TUNABLEOP_FILENAME"] = os.path.join(d, "tunableop.csv")
+
+
 def main():
     args = parse_args()
+    _enable_tuned_gemms()
     world = int(os.environ.get("WORLD_SIZE", 1))
     assert world == args.gpus or world == 1, (
         f"WORLD_SIZE {world} != --gpus {args.gpus}; launch N>1 with torchrun"

@@ -345,12 +345,12 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
   constexpr int DT = D / 32;
   constexpr int KC = D / 16;
 
-  __shared__ __align__(16) bf16_t q_row[QB * D];
-  __shared__ __align__(16) bf16_t q_tr[QB * D];
-  __shared__ __align__(16) bf16_t do_row[QB * D];
-  __shared__ __align__(16) bf16_t do_tr[QB * D];
-  __shared__ float lse_lds[QB];
-  __shared__ float drow_lds[QB];
+  __shared__ __align__(16) bf16_t q_row[2 * QB * D];
+  __shared__ __align__(16) bf16_t q_tr[2 * QB * D];
+  __shared__ __align__(16) bf16_t do_row[2 * QB * D];
+  __shared__ __align__(16) bf16_t do_tr[2 * QB * D];
+  __shared__ float lse_lds[2 * QB];
+  __shared__ float drow_lds[2 * QB];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -388,64 +388,71 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
   }
 
   const float ks = (p_drop > 0.f) ? 1.0f / (1.0f - p_drop) : 1.0f;
-  const int qt_start = causal ? (kv_block / QB) : 0;
-  const int n_qtiles = CDIV(Sq, QB);
+  const int qt_start = causal ? (kv_block / (2 * QB)) : 0;
+  const int n_qtiles = CDIV(Sq, 2 * QB);
 
   for (int qt = qt_start; qt < n_qtiles; ++qt) {
-    const int q0 = qt * QB;
+    const int q0 = qt * 2 * QB;  // 64 q rows staged per tile
     __syncthreads();
-    stage_tile<QB, D, true, true>(q_row, q_tr, qp, q0, Sq, q_ss, tid);
-    stage_tile<QB, D, true, true>(do_row, do_tr, dop, q0, Sq, do_ss, tid);
-    if (tid < QB) {
+    stage_tile<2 * QB, D, true, true>(q_row, q_tr, qp, q0, Sq, q_ss, tid);
+    stage_tile<2 * QB, D, true, true>(do_row, do_tr, dop, q0, Sq, do_ss, tid);
+    if (tid < 2 * QB) {
       int qr = min(q0 + tid, Sq - 1);
       lse_lds[tid] = lse[(int64_t)bh * Sq + qr];
       drow_lds[tid] = drow[(int64_t)bh * Sq + qr];
     }
     __syncthreads();
 
-    if (causal && q0 + QB - 1 < kv_base) continue;  // tile entirely above diag
+    if (causal && q0 + 2 * QB - 1 < kv_base) continue;  // entirely above diag
 
-    // S[q][kv] = Q x K^T  (lane col = kv)
-    f32x16_t s{};
-    f32x16_t dpd{};
 #pragma unroll
-    for (int c = 0; c < KC; ++c) {
-      bf16x8_t qa = row_img_frag<D>(q_row, l31, c, hi);
-      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kf[c], s, 0, 0, 0);
-      bf16x8_t da = row_img_frag<D>(do_row, l31, c, hi);
-      dpd = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vf[c], dpd, 0, 0, 0);
-    }
+    for (int sub = 0; sub < 2; ++sub) {
+      const int q0s = q0 + sub * QB;
+      if (causal && q0s + QB - 1 < kv_base) continue;
+      const int ro = sub * QB;  // row offset inside the staged images
 
-    // per-reg: q = q0 + pattern(r); kv = lane's kvg
-    float pd[16], ds[16];
+      // S[q][kv] = Q x K^T  (lane col = kv)
+      f32x16_t s{};
+      f32x16_t dpd{};
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-      const bool valid = qrow < Sq && kvg < Sk && (!causal || qrow >= kvg);
-      float p = 0.f;
-      if (valid) p = __expf(s[r] * scale - lse_lds[qrow - q0]);
-      float keep = (p_drop > 0.f && valid)
-                       ? drop_keep(seed, bh, Sq, Sk, qrow, kvg, p_drop, ks)
-                       : (valid ? 1.f : 0.f);
-      pd[r] = p * keep;
-      ds[r] = scale * p * (dpd[r] * keep - drow_lds[qrow - q0]);
-      if (!valid) ds[r] = 0.f;
-    }
+      for (int c = 0; c < KC; ++c) {
+        bf16x8_t qa = row_img_frag<D>(q_row, ro + l31, c, hi);
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kf[c], s, 0, 0, 0);
+        bf16x8_t da = row_img_frag<D>(do_row, ro + l31, c, hi);
+        dpd = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vf[c], dpd, 0, 0, 0);
+      }
 
-    // dV^T[d][kv] += dO^T x Pd ;  dK^T[d][kv] += Q^T x dS
+      // per-reg: q = q0s + pattern(r); kv = lane's kvg
+      float pd[16], ds[16];
 #pragma unroll
-    for (int c16 = 0; c16 < 2; ++c16) {
-      bf16x8_t pdf = repack_chunk(&pd[c16 * 8]);
-      bf16x8_t dsf = repack_chunk(&ds[c16 * 8]);
-      const int qc = c16 * 16;
+      for (int r = 0; r < 16; ++r) {
+        const int qrow = q0s + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const bool valid = qrow < Sq && kvg < Sk && (!causal || qrow >= kvg);
+        float p = 0.f;
+        if (valid) p = __expf(s[r] * scale - lse_lds[qrow - q0]);
+        float keep = (p_drop > 0.f && valid)
+                         ? drop_keep(seed, bh, Sq, Sk, qrow, kvg, p_drop, ks)
+                         : (valid ? 1.f : 0.f);
+        pd[r] = p * keep;
+        ds[r] = scale * p * (dpd[r] * keep - drow_lds[qrow - q0]);
+        if (!valid) ds[r] = 0.f;
+      }
+
+      // dV^T[d][kv] += dO^T x Pd ;  dK^T[d][kv] += Q^T x dS
 #pragma unroll
-      for (int dt = 0; dt < DT; ++dt) {
-        bf16x8_t dof = tr_img_frag<D>(do_tr, qc, dt, lane);
-        dv_acc[dt] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, pdf, dv_acc[dt], 0, 0, 0);
-        bf16x8_t qtf = tr_img_frag<D>(q_tr, qc, dt, lane);
-        dk_acc[dt] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(qtf, dsf, dk_acc[dt], 0, 0, 0);
+      for (int c16 = 0; c16 < 2; ++c16) {
+        bf16x8_t pdf = repack_chunk(&pd[c16 * 8]);
+        bf16x8_t dsf = repack_chunk(&ds[c16 * 8]);
+        const int qc = ro + c16 * 16;
+#pragma unroll
+        for (int dt = 0; dt < DT; ++dt) {
+          bf16x8_t dof = tr_img_frag<D>(do_tr, qc, dt, lane);
+          dv_acc[dt] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, pdf, dv_acc[dt], 0, 0, 0);
+          bf16x8_t qtf = tr_img_frag<D>(q_tr, qc, dt, lane);
+          dk_acc[dt] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(qtf, dsf, dk_acc[dt], 0, 0, 0);
+        }
       }
     }
   }
@@ -486,9 +493,9 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
   constexpr int DT = D / 32;
   constexpr int KC = D / 16;
 
-  __shared__ __align__(16) bf16_t k_row[QB * D];
-  __shared__ __align__(16) bf16_t k_tr[QB * D];
-  __shared__ __align__(16) bf16_t v_row[QB * D];
+  __shared__ __align__(16) bf16_t k_row[2 * QB * D];
+  __shared__ __align__(16) bf16_t k_tr[2 * QB * D];
+  __shared__ __align__(16) bf16_t v_row[2 * QB * D];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -526,50 +533,57 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
 
   const float ks = (p_drop > 0.f) ? 1.0f / (1.0f - p_drop) : 1.0f;
   const int kv_end = causal ? min(Sk, q_block + QBLK) : Sk;
-  const int n_tiles = CDIV(kv_end, QB);
+  const int n_tiles = CDIV(kv_end, 2 * QB);
 
   for (int tile = 0; tile < n_tiles; ++tile) {
-    const int kv0 = tile * QB;
+    const int kv0 = tile * 2 * QB;  // 64 kv rows staged per tile
     __syncthreads();
-    stage_tile<QB, D, true, true>(k_row, k_tr, kp, kv0, Sk, k_ss, tid);
-    stage_tile<QB, D, true, false>(v_row, nullptr, vp, kv0, Sk, v_ss, tid);
+    stage_tile<2 * QB, D, true, true>(k_row, k_tr, kp, kv0, Sk, k_ss, tid);
+    stage_tile<2 * QB, D, true, false>(v_row, nullptr, vp, kv0, Sk, v_ss, tid);
     __syncthreads();
 
     if (causal && kv0 > q_base + QB - 1) continue;
 
-    // S^T[kv][q] = K x Q^T ;  dPd^T[kv][q] = V x dO^T   (lane col = q)
-    f32x16_t st{};
-    f32x16_t dpdt{};
 #pragma unroll
-    for (int c = 0; c < KC; ++c) {
-      bf16x8_t ka = row_img_frag<D>(k_row, l31, c, hi);
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qf[c], st, 0, 0, 0);
-      bf16x8_t va = row_img_frag<D>(v_row, l31, c, hi);
-      dpdt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dof[c], dpdt, 0, 0, 0);
-    }
+    for (int sub = 0; sub < 2; ++sub) {
+      const int kv0s = kv0 + sub * QB;
+      if (kv0s >= kv_end || (causal && kv0s > q_base + QB - 1)) continue;
+      const int ro = sub * QB;
 
-    float ds[16];
+      // S^T[kv][q] = K x Q^T ;  dPd^T[kv][q] = V x dO^T   (lane col = q)
+      f32x16_t st{};
+      f32x16_t dpdt{};
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int kv = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-      const bool valid = kv < Sk && qg < Sq && (!causal || kv <= qg);
-      float p = valid ? __expf(st[r] * scale - lse_lane) : 0.f;
-      float keep = (p_drop > 0.f && valid)
-                       ? drop_keep(seed, bh, Sq, Sk, qg, kv, p_drop, ks)
-                       : (valid ? 1.f : 0.f);
-      ds[r] = valid ? scale * p * (dpdt[r] * keep - drow_lane) : 0.f;
-    }
+      for (int c = 0; c < KC; ++c) {
+        bf16x8_t ka = row_img_frag<D>(k_row, ro + l31, c, hi);
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qf[c], st, 0, 0, 0);
+        bf16x8_t va = row_img_frag<D>(v_row, ro + l31, c, hi);
+        dpdt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dof[c], dpdt, 0, 0, 0);
+      }
 
-    // dQ^T[d][q] += K^T x dS^T
+      float ds[16];
 #pragma unroll
-    for (int c16 = 0; c16 < 2; ++c16) {
-      bf16x8_t dsf = repack_chunk(&ds[c16 * 8]);
-      const int kvc = c16 * 16;
+      for (int r = 0; r < 16; ++r) {
+        const int kv = kv0s + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const bool valid = kv < Sk && qg < Sq && (!causal || kv <= qg);
+        float p = valid ? __expf(st[r] * scale - lse_lane) : 0.f;
+        float keep = (p_drop > 0.f && valid)
+                         ? drop_keep(seed, bh, Sq, Sk, qg, kv, p_drop, ks)
+                         : (valid ? 1.f : 0.f);
+        ds[r] = valid ? scale * p * (dpdt[r] * keep - drow_lane) : 0.f;
+      }
+
+      // dQ^T[d][q] += K^T x dS^T
 #pragma unroll
-      for (int dt = 0; dt < DT; ++dt) {
-        bf16x8_t ktf = tr_img_frag<D>(k_tr, kvc, dt, lane);
-        dq_acc[dt] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(ktf, dsf, dq_acc[dt], 0, 0, 0);
+      for (int c16 = 0; c16 < 2; ++c16) {
+        bf16x8_t dsf = repack_chunk(&ds[c16 * 8]);
+        const int kvc = ro + c16 * 16;
+#pragma unroll
+        for (int dt = 0; dt < DT; ++dt) {
+          bf16x8_t ktf = tr_img_frag<D>(k_tr, kvc, dt, lane);
+          dq_acc[dt] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(ktf, dsf, dq_acc[dt], 0, 0, 0);
+        }
       }
     }
   }
