@@ -153,10 +153,7 @@ def main():
         return {k: v.to(device, non_blocking=True) for k, v in b.items()}
 
     def sync_dp_grads():
-        if dp > 1 and dist.is_initialized():
-            for _, bk in optimizer.buckets:
-                bk.flat_grad.div_(dp)
-                dist.all_reduce(bk.flat_grad, group=dutil.data_parallel_group)
+        optimizer.grad_sync()
 
     step_i = 0
 

@@ -95,6 +95,22 @@ void attn_dropout_apply_bf16(void*, int64_t, int64_t, int64_t, float, uint64_t,
                              hipStream_t);
 void attn_dropout_apply_f32(void*, int64_t, int64_t, int64_t, float, uint64_t,
                             hipStream_t);
+void rope_fwd_bf16(const void*, void*, const float*, const float*, int64_t, int64_t,
+                   int64_t, int64_t, int64_t, int64_t, int, int, int, int, int,
+                   hipStream_t);
+void rope_fwd_f32(const void*, void*, const float*, const float*, int64_t, int64_t,
+                  int64_t, int64_t, int64_t, int64_t, int, int, int, int, int,
+                  hipStream_t);
+void rope_bwd_bf16(const void*, void*, const float*, const float*, int64_t, int64_t,
+                   int64_t, int64_t, int64_t, int64_t, int, int, int, int, int,
+                   hipStream_t);
+void rope_bwd_f32(const void*, void*, const float*, const float*, int64_t, int64_t,
+                  int64_t, int64_t, int64_t, int64_t, int, int, int, int, int,
+                  hipStream_t);
+void swiglu_fwd_bf16(const void*, void*, int64_t, int, hipStream_t);
+void swiglu_fwd_f32(const void*, void*, int64_t, int, hipStream_t);
+void swiglu_bwd_bf16(const void*, const void*, void*, int64_t, int, hipStream_t);
+void swiglu_bwd_f32(const void*, const void*, void*, int64_t, int, hipStream_t);
 int adamw_chunk_elems();
 void adamw_step_bf16(const void*, int, float, float, float, float, float, float, float,
                      float, hipStream_t);
@@ -368,6 +384,50 @@ void attn_dropout_apply(torch::Tensor x, int64_t Sq, int64_t Sk, double p,
   check_launch("attn_dropout_apply");
 }
 
+
+// ---------------------------------------------------------------------------
+// RoPE / SwiGLU
+// ---------------------------------------------------------------------------
+torch::Tensor rope(torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t,
+                   int64_t pos0, bool bwd) {
+  // x: [B, S, NH, HS] (strided view ok, last dim contiguous)
+  TORCH_CHECK(x.dim() == 4 && x.stride(3) == 1);
+  auto out = torch::empty(x.sizes(), x.options());
+  const int B = (int)x.size(0), S = (int)x.size(1), NH = (int)x.size(2),
+            HS = (int)x.size(3);
+  auto fn = bwd ? (is_bf16(x) ? rope_bwd_bf16 : rope_bwd_f32)
+                : (is_bf16(x) ? rope_fwd_bf16 : rope_fwd_f32);
+  fn(x.data_ptr(), out.data_ptr(), cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
+     x.stride(0), x.stride(1), x.stride(2), out.stride(0), out.stride(1),
+     out.stride(2), B, S, NH, HS, (int)pos0, cur_stream());
+  check_launch("rope");
+  return out;
+}
+
+torch::Tensor swiglu_fwd(torch::Tensor x) {
+  CHECK_IN(x);
+  const int64_t F = x.size(-1) / 2;
+  auto sizes = x.sizes().vec();
+  sizes.back() = F;
+  auto y = torch::empty(sizes, x.options());
+  auto fn = is_bf16(x) ? swiglu_fwd_bf16 : swiglu_fwd_f32;
+  fn(x.data_ptr(), y.data_ptr(), x.numel() / (2 * F), (int)F, cur_stream());
+  check_launch("swiglu_fwd");
+  return y;
+}
+
+torch::Tensor swiglu_bwd(torch::Tensor x, torch::Tensor dy) {
+  CHECK_IN(x);
+  CHECK_IN(dy);
+  const int64_t F = x.size(-1) / 2;
+  auto dx = torch::empty_like(x);
+  auto fn = is_bf16(x) ? swiglu_bwd_bf16 : swiglu_bwd_f32;
+  fn(x.data_ptr(), dy.data_ptr(), dx.data_ptr(), x.numel() / (2 * F), (int)F,
+     cur_stream());
+  check_launch("swiglu_bwd");
+  return dx;
+}
+
 // ---------------------------------------------------------------------------
 // fused AdamW
 // ---------------------------------------------------------------------------
@@ -405,6 +465,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_dropout_apply", &attn_dropout_apply);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
+  m.def("rope", &rope);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
   m.def("adamw_step", &adamw_step);
   m.def("l2norm_sq", &l2norm_sq);
   m.def("adamw_chunk_elems", &adamw_chunk_elems);

@@ -59,7 +59,7 @@ class SyntheticBertDataset(torch.utils.data.Dataset):
         ns_label = torch.randint(0, 2, (1,), generator=g)[0]
         return Instance(
             input_ids=DistTensorData(input_ids.long()),
-            attention_mask=DistTensorData(torch.zeros(self.seq_length, dtype=torch.uint8)),
+            attention_mask=DistTensorData(torch.ones(self.seq_length, dtype=torch.uint8)),
             tokentype_ids=DistTensorData(torch.zeros(self.seq_length, dtype=torch.long)),
             ns_labels=DistTensorData(ns_label.long(), placement_idx=-1),
             lm_labels=DistTensorData(lm_labels.long(), placement_idx=-1),

@@ -199,7 +199,13 @@ class DefaultTrainer(TrainerBase):
 
     @classmethod
     def build_optimizer(cls, cfg, model):
-        return build_optimizer(cfg.optim, model)
+        opt = build_optimizer(cfg.optim, model)
+        from ..parallel.zero import zero_stage_from_config
+
+        stage = zero_stage_from_config(cfg)
+        if stage and hasattr(opt, "zero_stage"):
+            opt.zero_stage = stage  # buckets are built lazily, so this applies
+        return opt
 
     @classmethod
     def build_lr_scheduler(cls, cfg, optimizer):

@@ -162,14 +162,18 @@ class EagerTrainer(TrainerBase):
         }
 
     def _sync_dp_grads(self):
+        if hasattr(self.optimizer, "grad_sync"):
+            self.optimizer.grad_sync()  # all-reduce, or ZeRO-2 reduce-scatter
+            return
         import torch.distributed as dist
 
         dutil = du.get_dist_util()
         if dutil.data_parallel_size == 1 or not dist.is_initialized():
             return
-        for _, b in self.optimizer.buckets:
-            b.flat_grad.div_(dutil.data_parallel_size)
-            dist.all_reduce(b.flat_grad, group=dutil.data_parallel_group)
+        for p in self.model.parameters():
+            if p.grad is not None:
+                p.grad.div_(dutil.data_parallel_size)
+                dist.all_reduce(p.grad, group=dutil.data_parallel_group)
 
     def run_step(self):
         start = time.perf_counter()

@@ -1,4 +1,16 @@
-from .gpt_model import GPTForPreTraining, GPTModel
+from .bert_model import BertForPreTraining, BertModel
 from .build import build_model
+from .gpt_model import GPTForPreTraining, GPTModel
+from .llama import LlamaForCausalLM, LlamaModel
+from .vision_transformer import VisionTransformer
 
-__all__ = ["GPTModel", "GPTForPreTraining", "build_model"]
+__all__ = [
+    "GPTModel",
+    "GPTForPreTraining",
+    "BertModel",
+    "BertForPreTraining",
+    "LlamaModel",
+    "LlamaForCausalLM",
+    "VisionTransformer",
+    "build_model",
+]

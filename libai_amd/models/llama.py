@@ -1,0 +1,321 @@
+"""Llama model family (reference: projects/Llama/llama.py).
+
+MI355X-native hot path: RMSNorm HIP kernel, RoPE HIP kernel on strided qkv
+views, flash attention (head_dim 128 for 7B), fused SwiGLU MLP, untied
+vocab-parallel output head + vocab-parallel CE.
+"""
+
+import math
+
+import torch
+from torch import nn
+from torch.utils.checkpoint import checkpoint as act_checkpoint
+
+from ..config import configurable
+from ..layers import Linear1D, LMLogits, ParallelCrossEntropyLoss, RMSLayerNorm, VocabEmbedding
+from ..ops.attention import flash_attention, flash_attention_available
+from ..ops.fused_bias import bias_dropout_add
+from ..ops.rope import apply_rotary_pos_emb
+from ..ops.softmax import fused_scale_mask_softmax
+from ..ops.swiglu import swiglu
+from ..utils import distributed as du
+from .utils.weight_init import init_method_normal, scaled_init_method_normal
+
+__all__ = ["LlamaModel", "LlamaForCausalLM"]
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, hidden_size, num_heads, max_position_embeddings,
+                 init_method, output_init_method, rope_theta=10000.0, *, layer_idx=0):
+        super().__init__()
+        self.hidden_size = hidden_size
+        self.num_heads = num_heads
+        self.head_dim = hidden_size // num_heads
+        dutil = du.get_dist_util()
+        self.num_heads_local = num_heads // dutil.tensor_parallel_size
+        self.max_pos = max_position_embeddings
+        self.rope_theta = rope_theta
+        self.layer_idx = layer_idx
+        self.query_key_value = Linear1D(hidden_size, 3 * hidden_size, bias=False,
+                                        parallel="col", init_method=init_method,
+                                        layer_idx=layer_idx)
+        self.o_proj = Linear1D(hidden_size, hidden_size, bias=False, parallel="row",
+                               init_method=output_init_method, skip_bias_add=True,
+                               layer_idx=layer_idx)
+        self.scale = 1.0 / math.sqrt(self.head_dim)
+
+    def forward(self, hidden_states, past_key_value=None, use_cache=False,
+                residual=None):
+        b, s, _ = hidden_states.shape
+        qkv = self.query_key_value(hidden_states)
+        qkv5 = qkv.view(b, s, self.num_heads_local, 3, self.head_dim)
+        pos0 = past_key_value[0].shape[2] if past_key_value is not None else 0
+        q = apply_rotary_pos_emb(qkv5[..., 0, :], self.max_pos, self.rope_theta, pos0)
+        k = apply_rotary_pos_emb(qkv5[..., 1, :], self.max_pos, self.rope_theta, pos0)
+        v = qkv5[..., 2, :]
+
+        if (
+            past_key_value is None
+            and not use_cache
+            and flash_attention_available(self.head_dim, q.dtype, q.device, s, s, None)
+        ):
+            o = flash_attention(q, k, v, self.scale, p_drop=0.0, causal=True,
+                                training=self.training)
+            context = o.reshape(b, s, self.num_heads_local * self.head_dim)
+        else:
+            # unfused path (decode / CPU): [b, nh, s, hs]
+            qh = q.permute(0, 2, 1, 3)
+            kh = k.permute(0, 2, 1, 3)
+            vh = v.permute(0, 2, 1, 3)
+            if past_key_value is not None:
+                pk, pv = past_key_value
+                kh = torch.cat([pk, kh], dim=2)
+                vh = torch.cat([pv, vh], dim=2)
+            present = (kh, vh) if use_cache else None
+            scores = torch.matmul(qh, kh.transpose(-1, -2))
+            causal = past_key_value is None
+            probs = fused_scale_mask_softmax(scores, scale=self.scale, causal=causal,
+                                             training=self.training)
+            ctx = torch.matmul(probs, vh)
+            context = ctx.permute(0, 2, 1, 3).reshape(
+                b, s, self.num_heads_local * self.head_dim
+            )
+            out, _ = self.o_proj(context)
+            out = out + residual if residual is not None else out
+            if use_cache:
+                return out, present
+            return out
+
+        out, _ = self.o_proj(context)
+        return out + residual if residual is not None else out
+
+
+class LlamaMLP(nn.Module):
+    """Gated MLP: fused [gate|up] col projection -> SwiGLU -> row projection."""
+
+    def __init__(self, hidden_size, intermediate_size, init_method,
+                 output_init_method, *, layer_idx=0):
+        super().__init__()
+        self.gate_up_proj = Linear1D(hidden_size, 2 * intermediate_size, bias=False,
+                                     parallel="col", init_method=init_method,
+                                     layer_idx=layer_idx)
+        self.down_proj = Linear1D(intermediate_size, hidden_size, bias=False,
+                                  parallel="row", init_method=output_init_method,
+                                  skip_bias_add=True, layer_idx=layer_idx)
+
+    def forward(self, x, residual=None):
+        gu = self.gate_up_proj(x)
+        inter = swiglu(gu)
+        out, _ = self.down_proj(inter)
+        return out + residual if residual is not None else out
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, hidden_size, intermediate_size, num_heads,
+                 max_position_embeddings, rms_norm_eps, init_method,
+                 output_init_method, rope_theta=10000.0, *, layer_idx=0):
+        super().__init__()
+        self.layer_idx = layer_idx
+        self.input_layernorm = RMSLayerNorm(hidden_size, eps=rms_norm_eps,
+                                            layer_idx=layer_idx)
+        self.self_attn = LlamaAttention(hidden_size, num_heads,
+                                        max_position_embeddings, init_method,
+                                        output_init_method, rope_theta,
+                                        layer_idx=layer_idx)
+        self.post_attention_layernorm = RMSLayerNorm(hidden_size, eps=rms_norm_eps,
+                                                     layer_idx=layer_idx)
+        self.mlp = LlamaMLP(hidden_size, intermediate_size, init_method,
+                            output_init_method, layer_idx=layer_idx)
+
+    def forward(self, hidden_states, past_key_value=None, use_cache=False):
+        ln1 = self.input_layernorm(hidden_states)
+        attn_out = self.self_attn(ln1, past_key_value=past_key_value,
+                                  use_cache=use_cache, residual=hidden_states)
+        if use_cache:
+            attn_out, present = attn_out
+        h = attn_out
+        ln2 = self.post_attention_layernorm(h)
+        out = self.mlp(ln2, residual=h)
+        if use_cache:
+            return out, present
+        return out
+
+
+class LlamaModel(nn.Module):
+    @configurable
+    def __init__(
+        self,
+        hidden_layers,
+        vocab_size,
+        hidden_size,
+        intermediate_size,
+        num_attention_heads,
+        max_position_embeddings=2048,
+        rms_norm_eps=1e-5,
+        initializer_range=0.02,
+        use_scaled_init_for_output_weights=False,
+        tie_word_embeddings=False,
+        rope_theta=10000.0,
+        amp_enabled=False,
+    ):
+        super().__init__()
+        init_method = init_method_normal(initializer_range)
+        output_init_method = (
+            scaled_init_method_normal(initializer_range, hidden_layers)
+            if use_scaled_init_for_output_weights
+            else init_method
+        )
+        self.embed_tokens = VocabEmbedding(vocab_size, hidden_size,
+                                           init_method=init_method, layer_idx=0)
+        self.layers = nn.ModuleList(
+            [
+                LlamaDecoderLayer(
+                    hidden_size, intermediate_size, num_attention_heads,
+                    max_position_embeddings, rms_norm_eps, init_method,
+                    output_init_method, rope_theta, layer_idx=i,
+                )
+                for i in range(hidden_layers)
+            ]
+        )
+        self.norm = RMSLayerNorm(hidden_size, eps=rms_norm_eps, layer_idx=-1)
+        self.hidden_layers = hidden_layers
+        self.checkpoint_activations = False
+
+    @classmethod
+    def from_config(cls, cfg):
+        return {
+            "hidden_layers": cfg.hidden_layers,
+            "vocab_size": cfg.vocab_size,
+            "hidden_size": cfg.hidden_size,
+            "intermediate_size": cfg.intermediate_size,
+            "num_attention_heads": cfg.num_attention_heads,
+            "max_position_embeddings": cfg.get("max_position_embeddings", 2048),
+            "rms_norm_eps": cfg.get("rms_norm_eps", 1e-5),
+            "initializer_range": cfg.get("initializer_range", 0.02),
+            "use_scaled_init_for_output_weights": cfg.get(
+                "use_scaled_init_for_output_weights", False
+            ),
+            "tie_word_embeddings": cfg.get("tie_word_embeddings", False),
+            "rope_theta": cfg.get("rope_theta", 10000.0),
+            "amp_enabled": cfg.get("amp_enabled", False),
+        }
+
+    def _run_layer(self, layer, h, past=None, use_cache=False):
+        if self.checkpoint_activations and self.training and not use_cache:
+            return act_checkpoint(layer, h, use_reentrant=False)
+        return layer(h, past_key_value=past, use_cache=use_cache)
+
+    def forward(self, input_ids, past_key_values=None, use_cache=False):
+        h = self.embed_tokens(input_ids)
+        presents = [] if use_cache else None
+        for i, layer in enumerate(self.layers):
+            past = past_key_values[i] if past_key_values is not None else None
+            h = self._run_layer(layer, h, past, use_cache)
+            if use_cache:
+                h, p = h
+                presents.append(p)
+        h = self.norm(h)
+        if use_cache:
+            return h, presents
+        return h
+
+    def set_activation_checkpoint(self, enabled=True):
+        self.checkpoint_activations = enabled
+
+
+class _LMHeadWeight(nn.Module):
+    """Vocab-sharded untied output-projection weight (module-wrapped so the
+    pipeline prune places it on the last stage only)."""
+
+    def __init__(self, vocab_size, hidden_size, init_method):
+        super().__init__()
+        dutil = du.get_dist_util()
+        tp = dutil.tensor_parallel_size
+        self.weight = nn.Parameter(torch.empty(vocab_size // tp, hidden_size))
+        self.weight.tensor_parallel = True
+        self.weight.tp_shard_dim = 0
+        from ..layers.linear import init_tp_shard_
+
+        init_tp_shard_(self.weight, (vocab_size, hidden_size), init_method, 0)
+
+
+class LlamaLoss(nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.lm_loss = ParallelCrossEntropyLoss()
+
+    def forward(self, logits, labels):
+        return {"lm_loss": self.lm_loss(logits, labels).mean()}
+
+
+class LlamaForCausalLM(nn.Module):
+    @configurable
+    def __init__(self, cfg=None, **kwargs):
+        super().__init__()
+        self.model = LlamaModel(cfg) if cfg is not None else LlamaModel(**kwargs)
+        vocab = cfg.vocab_size if cfg is not None else kwargs["vocab_size"]
+        hidden = cfg.hidden_size if cfg is not None else kwargs["hidden_size"]
+        tie = (cfg.get("tie_word_embeddings", False) if cfg is not None
+               else kwargs.get("tie_word_embeddings", False))
+        self.tie_word_embeddings = tie
+        if not tie:
+            init_range = (cfg.get("initializer_range", 0.02) if cfg is not None
+                          else kwargs.get("initializer_range", 0.02))
+            self.lm_head = _LMHeadWeight(vocab, hidden,
+                                         init_method_normal(init_range))
+        self.lm_logits = LMLogits(vocab, bias=False, layer_idx=-1)
+        self.loss_func = LlamaLoss()
+
+    @classmethod
+    def from_config(cls, cfg):
+        return {"cfg": cfg}
+
+    def _head_weight(self):
+        return (self.model.embed_tokens.weight if self.tie_word_embeddings
+                else self.lm_head.weight)
+
+    def forward(self, input_ids, labels=None, past_key_values=None, use_cache=False):
+        h = self.model(input_ids, past_key_values=past_key_values, use_cache=use_cache)
+        if use_cache:
+            h, presents = h
+        logits = self.lm_logits(h, self._head_weight())
+        if labels is not None:
+            return self.loss_func(logits, labels)
+        if use_cache:
+            return {"prediction_scores": logits, "past_key_values": presents}
+        return {"prediction_scores": logits}
+
+    def set_activation_checkpoint(self, enabled=True):
+        self.model.set_activation_checkpoint(enabled)
+
+    # -- pipeline protocol --------------------------------------------------
+
+    def pipeline_units(self):
+        units = [(0, "embed_tokens", lambda h, b: self.model.embed_tokens(b["input_ids"]))]
+        for i, layer in enumerate(self.model.layers):
+            units.append(
+                (i, f"layer_{i}",
+                 (lambda lyr: lambda h, b: self.model._run_layer(lyr, h))(layer))
+            )
+
+        def head(h, b):
+            h = self.model.norm(h)
+            logits = self.lm_logits(h, self._head_weight())
+            if b.get("labels") is not None:
+                return self.loss_func(logits, b["labels"])
+            return {"prediction_scores": logits}
+
+        units.append((-1, "head", head))
+        return units
+
+    def pipeline_stage_modules(self):
+        m = {0: [self.model.embed_tokens]}
+        for i, layer in enumerate(self.model.layers):
+            m.setdefault(i, []).append(layer)
+        last = [self.model.norm, self.lm_logits, self.loss_func]
+        if self.tie_word_embeddings:
+            last.append(self.model.embed_tokens)
+        else:
+            last.append(self.lm_head)
+        m.setdefault(-1, []).extend(last)
+        return m

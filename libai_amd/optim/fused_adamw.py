@@ -1,26 +1,27 @@
-"""Flat-buffer fused AdamW for MI355X.
+"""Flat-buffer fused AdamW for MI355X, with ZeRO-1/2 sharding.
 
 Design (replaces the reference's OneFlow fused model update + param-group
-clip_grad, reference: libai/models/utils/graph_base.py:74-76,
+clip_grad + enable_zero, reference: libai/models/utils/graph_base.py:69-76,
 libai/optim/build.py:86-126):
 
   * Parameters are grouped into buckets by (dtype, weight-decay on/off,
     tp-sharded) and each bucket's storage is FLATTENED: ``p.data`` becomes a
     view into one contiguous bf16 buffer, and ``p.grad`` a view into a flat
-    grad buffer.  DP all-reduce, grad zeroing, clipping and the AdamW update
-    each touch a handful of contiguous buffers instead of hundreds of
-    tensors.
+    grad buffer.  DP all-reduce / reduce-scatter, grad zeroing, clipping and
+    the AdamW update each touch a handful of contiguous buffers.
   * bf16 training keeps fp32 master weights; the HIP kernel updates master
     and rewrites the bf16 copy in one pass (csrc/kernels/adamw.hip).
-  * Gradient clipping computes the global L2 norm with a multi-tensor HIP
-    kernel; TP-sharded buckets contribute from every TP rank, replicated
-    buckets only from tp_rank 0, then the squared norm is all-reduced over
-    the model-parallel (TP x PP) axes — SURVEY.md §7 hard part 7.
-  * On CPU (tests) the same class runs a pure-torch reference update with
-    identical flat-buffer semantics.
+  * ZeRO (flat buckets make DP slices contiguous):
+      - stage 1: optimizer state (master/m/v) exists only for this DP rank's
+        1/dp slice; grads are all-reduced in full; after the slice update the
+        bf16 params are all-gathered over DP.
+      - stage 2: gradients are reduce-scattered so each rank only receives
+        its own slice's reduction (C6 in SURVEY.md §2.4).
+  * Gradient clipping computes the global L2 norm (TP-aware: replicated
+    buckets counted once per TP group; under ZeRO the slice norms are
+    additionally reduced over DP) and applies the clip scale inside the
+    fused update — one pass over grads total.
 """
-
-import math
 
 import torch
 import torch.distributed as dist
@@ -31,16 +32,27 @@ from ..utils import distributed as du
 __all__ = ["FusedAdamW"]
 
 
+def _pad_to(n, mult):
+    return (n + mult - 1) // mult * mult
+
+
 class _Bucket:
-    def __init__(self, params, dtype, device, weight_decay_on, tp_sharded):
+    def __init__(self, params, dtype, device, weight_decay_on, tp_sharded, dp_size,
+                 dp_rank, zero_stage):
         self.params = params
         self.dtype = dtype
         self.weight_decay_on = weight_decay_on
         self.tp_sharded = tp_sharded
-        self.numel = sum(p.numel() for p in params)
-        self.flat_param = torch.empty(self.numel, dtype=dtype, device=device)
+        self.zero = zero_stage
+        self.dp_size = dp_size
+        self.dp_rank = dp_rank
+        raw = sum(p.numel() for p in params)
+        self.numel = _pad_to(raw, dp_size * 64) if zero_stage > 0 else raw
+        self.shard = self.numel // dp_size if zero_stage > 0 else self.numel
+        self.shard_off = self.shard * dp_rank if zero_stage > 0 else 0
+
+        self.flat_param = torch.zeros(self.numel, dtype=dtype, device=device)
         self.flat_grad = torch.zeros(self.numel, dtype=dtype, device=device)
-        # flatten param storage and attach grad views
         off = 0
         for p in params:
             n = p.numel()
@@ -50,45 +62,61 @@ class _Bucket:
                 self.flat_grad[off : off + n].copy_(p.grad.reshape(-1))
             p.grad = self.flat_grad[off : off + n].view(p.shape)
             off += n
-        self.flat_master = self.flat_param.float() if dtype != torch.float32 else self.flat_param
-        self.exp_avg = torch.zeros(self.numel, dtype=torch.float32, device=device)
-        self.exp_avg_sq = torch.zeros(self.numel, dtype=torch.float32, device=device)
+
+        # optimizer state exists only for the local shard under ZeRO
+        owned = self.flat_param[self.shard_off : self.shard_off + self.shard]
+        self.flat_master = (
+            owned.float() if dtype != torch.float32 else owned
+        ) if zero_stage > 0 else (
+            self.flat_param.float() if dtype != torch.float32 else self.flat_param
+        )
+        state_n = self.shard if zero_stage > 0 else self.numel
+        self.exp_avg = torch.zeros(state_n, dtype=torch.float32, device=device)
+        self.exp_avg_sq = torch.zeros(state_n, dtype=torch.float32, device=device)
         self._adam_desc = None
         self._norm_desc = None
+
+    # region updated by the fused kernel (local shard under ZeRO)
+    def _upd_param(self):
+        return self.flat_param[self.shard_off : self.shard_off + self.shard] \
+            if self.zero > 0 else self.flat_param
+
+    def _upd_grad(self):
+        return self.flat_grad[self.shard_off : self.shard_off + self.shard] \
+            if self.zero > 0 else self.flat_grad
 
     def adam_desc(self, chunk):
         if self._adam_desc is None:
             esz = self.flat_param.element_size()
+            up, ug = self._upd_param(), self._upd_grad()
+            n_total = up.numel()
             rows = []
-            for off in range(0, self.numel, chunk):
-                n = min(chunk, self.numel - off)
+            for off in range(0, n_total, chunk):
+                n = min(chunk, n_total - off)
                 rows.append(
                     [
-                        self.flat_param.data_ptr() + off * esz
-                        if self.dtype != torch.float32
-                        else 0,
+                        up.data_ptr() + off * esz if self.dtype != torch.float32 else 0,
                         self.flat_master.data_ptr() + off * 4,
-                        self.flat_grad.data_ptr() + off * esz,
+                        ug.data_ptr() + off * esz,
                         self.exp_avg.data_ptr() + off * 4,
                         self.exp_avg_sq.data_ptr() + off * 4,
                         n,
                     ]
                 )
-            self._adam_desc = torch.tensor(
-                rows, dtype=torch.int64, device=self.flat_param.device
-            )
+            self._adam_desc = torch.tensor(rows, dtype=torch.int64,
+                                           device=self.flat_param.device)
         return self._adam_desc
 
     def norm_desc(self, chunk):
         if self._norm_desc is None:
             esz = self.flat_grad.element_size()
+            ug = self._upd_grad()
             rows = []
-            for off in range(0, self.numel, chunk):
-                n = min(chunk, self.numel - off)
-                rows.append([self.flat_grad.data_ptr() + off * esz, n])
-            self._norm_desc = torch.tensor(
-                rows, dtype=torch.int64, device=self.flat_grad.device
-            )
+            for off in range(0, ug.numel(), chunk):
+                n = min(chunk, ug.numel() - off)
+                rows.append([ug.data_ptr() + off * esz, n])
+            self._norm_desc = torch.tensor(rows, dtype=torch.int64,
+                                           device=self.flat_grad.device)
         return self._norm_desc
 
 
@@ -100,19 +128,25 @@ class FusedAdamW(torch.optim.Optimizer):
     """AdamW over flat buckets.  Extra kwargs:
 
     clip_grad: max global grad norm (0 = off), applied inside step()
+    zero_stage: 0 (off) / 1 (state sharding) / 2 (+grad reduce-scatter)
     """
 
     def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
-                 weight_decay=0.01, clip_grad=0.0):
+                 weight_decay=0.01, clip_grad=0.0, zero_stage=0):
         defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
         super().__init__(params, defaults)
         self.clip_grad = clip_grad
+        self.zero_stage = zero_stage
         self._step = 0
-        self._buckets = None  # list of (group_idx, _Bucket)
+        self._buckets = None
 
     # -- bucket construction (lazy: after model is on its final device) -----
 
     def _build_buckets(self):
+        dutil = du.get_dist_util()
+        dp, dpr = dutil.data_parallel_size, dutil.data_parallel_rank
+        zero = self.zero_stage if dp > 1 else 0
+        self._zero_eff = zero
         self._buckets = []
         for gi, group in enumerate(self.param_groups):
             by_key = {}
@@ -124,7 +158,8 @@ class FusedAdamW(torch.optim.Optimizer):
             wd_on = group["weight_decay"] > 0
             for (dtype, device, tp_sharded), plist in by_key.items():
                 self._buckets.append(
-                    (gi, _Bucket(plist, dtype, device, wd_on, tp_sharded))
+                    (gi, _Bucket(plist, dtype, device, wd_on, tp_sharded, dp, dpr,
+                                 zero))
                 )
 
     @property
@@ -134,9 +169,35 @@ class FusedAdamW(torch.optim.Optimizer):
         return self._buckets
 
     def zero_grad(self, set_to_none=False):
-        # grads are persistent flat views -> zero in place, never free
         for _, b in self.buckets:
             b.flat_grad.zero_()
+
+    # -- DP gradient communication (all-reduce / ZeRO reduce-scatter) --------
+
+    def grad_sync(self):
+        """Average gradients over the DP group; with ZeRO-2, reduce-scatter so
+        only the local slice is received."""
+        dutil = du.get_dist_util()
+        dp = dutil.data_parallel_size
+        if dp == 1 or not dist.is_initialized():
+            return
+        group = dutil.data_parallel_group
+        for _, b in self.buckets:
+            b.flat_grad.div_(dp)
+            if self._zero_eff >= 2:
+                out = b._upd_grad()
+                dist.reduce_scatter_tensor(out, b.flat_grad, group=group)
+            else:
+                dist.all_reduce(b.flat_grad, group=group)
+
+    def _gather_params(self):
+        dutil = du.get_dist_util()
+        if self._zero_eff == 0 or dutil.data_parallel_size == 1:
+            return
+        group = dutil.data_parallel_group
+        for _, b in self.buckets:
+            dist.all_gather_into_tensor(b.flat_param, b._upd_param().contiguous(),
+                                        group=group)
 
     # -- grad norm / clip ---------------------------------------------------
 
@@ -152,8 +213,10 @@ class FusedAdamW(torch.optim.Optimizer):
             if use_hip:
                 ext().l2norm_sq(b.norm_desc(chunk), b.dtype == torch.bfloat16, total)
             else:
-                total += b.flat_grad.float().pow(2).sum()
+                total += b._upd_grad().float().pow(2).sum()
         if dist.is_initialized():
+            if self._zero_eff > 0 and dutil.data_parallel_size > 1:
+                dist.all_reduce(total, group=dutil.data_parallel_group)
             if dutil.tensor_parallel_size > 1:
                 dist.all_reduce(total, group=dutil.tensor_parallel_group)
             if dutil.pipeline_parallel_size > 1:
@@ -172,12 +235,9 @@ class FusedAdamW(torch.optim.Optimizer):
         t = self._step
 
         grad_scale = 1.0
-        clip_coef_t = None
         if self.clip_grad and self.clip_grad > 0:
             norm = self._grad_norm_sq().sqrt()
-            # compute on host only on CPU; on GPU keep device-side (one sync ok)
-            clip_coef_t = (self.clip_grad / (norm + 1e-6)).clamp(max=1.0)
-            grad_scale = float(clip_coef_t.item())
+            grad_scale = float((self.clip_grad / (norm + 1e-6)).clamp(max=1.0).item())
 
         for gi, b in self.buckets:
             group = self.param_groups[gi]
@@ -194,7 +254,7 @@ class FusedAdamW(torch.optim.Optimizer):
                     eps, wd, bc1, bc2, grad_scale,
                 )
             else:
-                g = b.flat_grad.float() * grad_scale
+                g = b._upd_grad().float() * grad_scale
                 b.exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
                 b.exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
                 mhat = b.exp_avg / bc1
@@ -203,22 +263,40 @@ class FusedAdamW(torch.optim.Optimizer):
                     mhat / (vhat.sqrt() + eps) + wd * b.flat_master, alpha=-lr
                 )
                 if b.dtype != torch.float32:
-                    b.flat_param.copy_(b.flat_master.to(b.dtype))
+                    b._upd_param().copy_(b.flat_master.to(b.dtype))
+                elif self._zero_eff > 0:
+                    b._upd_param().copy_(b.flat_master)
+        self._gather_params()
         return loss
 
     # -- state dict (topology-independent per-param tensors) ----------------
 
+    def _full_state(self, b):
+        """(master, exp_avg, exp_avg_sq) covering the WHOLE bucket (gathers
+        the DP shards under ZeRO)."""
+        dutil = du.get_dist_util()
+        if self._zero_eff == 0 or dutil.data_parallel_size == 1:
+            return b.flat_master, b.exp_avg, b.exp_avg_sq
+        group = dutil.data_parallel_group
+        outs = []
+        for t in (b.flat_master, b.exp_avg, b.exp_avg_sq):
+            full = torch.empty(b.numel, dtype=torch.float32, device=t.device)
+            dist.all_gather_into_tensor(full, t.contiguous(), group=group)
+            outs.append(full)
+        return outs
+
     def state_dict(self):
         per_param = []
         for gi, b in self.buckets:
+            master, m, v = self._full_state(b)
             off = 0
             for p in b.params:
                 n = p.numel()
                 per_param.append(
                     {
-                        "master": b.flat_master[off : off + n].clone(),
-                        "exp_avg": b.exp_avg[off : off + n].clone(),
-                        "exp_avg_sq": b.exp_avg_sq[off : off + n].clone(),
+                        "master": master[off : off + n].clone(),
+                        "exp_avg": m[off : off + n].clone(),
+                        "exp_avg_sq": v[off : off + n].clone(),
                         "shape": list(p.shape),
                     }
                 )
@@ -237,14 +315,25 @@ class FusedAdamW(torch.optim.Optimizer):
             g.update(saved)
         idx = 0
         for gi, b in self.buckets:
+            device = b.flat_param.device
+            master = torch.zeros(b.numel, dtype=torch.float32, device=device)
+            m = torch.zeros_like(master)
+            v = torch.zeros_like(master)
             off = 0
             for p in b.params:
                 n = p.numel()
                 entry = state_dict["per_param"][idx]
-                b.flat_master[off : off + n].copy_(entry["master"])
-                b.exp_avg[off : off + n].copy_(entry["exp_avg"])
-                b.exp_avg_sq[off : off + n].copy_(entry["exp_avg_sq"])
+                master[off : off + n].copy_(entry["master"])
+                m[off : off + n].copy_(entry["exp_avg"])
+                v[off : off + n].copy_(entry["exp_avg_sq"])
                 idx += 1
                 off += n
+            sl = slice(b.shard_off, b.shard_off + b.shard) if self._zero_eff > 0 \
+                else slice(0, b.numel)
+            b.flat_master.copy_(master[sl])
+            b.exp_avg.copy_(m[sl])
+            b.exp_avg_sq.copy_(v[sl])
             if b.dtype != torch.float32:
-                b.flat_param.copy_(b.flat_master.to(b.dtype))
+                b.flat_param.copy_(master.to(b.dtype))
+            else:
+                b.flat_param.copy_(master)
