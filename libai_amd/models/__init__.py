@@ -2,6 +2,10 @@ from .bert_model import BertForPreTraining, BertModel
 from .build import build_model
 from .gpt_model import GPTForPreTraining, GPTModel
 from .llama import LlamaForCausalLM, LlamaModel
+from .resmlp import ResMLP
+from .roberta_model import RobertaForCausalLM, RobertaForPreTraining, RobertaModel
+from .swin_transformer import SwinTransformer
+from .t5_model import T5ForPreTraining, T5Model
 from .vision_transformer import VisionTransformer
 
 __all__ = [
@@ -9,8 +13,15 @@ __all__ = [
     "GPTForPreTraining",
     "BertModel",
     "BertForPreTraining",
+    "T5Model",
+    "T5ForPreTraining",
+    "RobertaModel",
+    "RobertaForPreTraining",
+    "RobertaForCausalLM",
     "LlamaModel",
     "LlamaForCausalLM",
     "VisionTransformer",
+    "SwinTransformer",
+    "ResMLP",
     "build_model",
 ]

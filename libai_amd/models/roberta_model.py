@@ -1,0 +1,108 @@
+"""RoBERTa model (reference: libai/models/roberta_model.py).
+
+Structurally BERT without the NSP head, with RoBERTa's padding-aware
+position ids (positions start at padding_idx+1 and padding tokens keep
+padding_idx) and an MLM head (dense + gelu + LN + tied logits).
+"""
+
+import torch
+from torch import nn
+
+from ..config import configurable
+from ..layers import LMLogits, ParallelCrossEntropyLoss
+from .bert_model import BertLMPredictionHead, BertModel
+from .utils.weight_init import init_method_normal
+
+__all__ = ["RobertaModel", "RobertaForPreTraining", "RobertaForCausalLM"]
+
+
+def create_position_ids_from_input_ids(input_ids, padding_idx=1):
+    mask = input_ids.ne(padding_idx).long()
+    incremental = torch.cumsum(mask, dim=1) * mask
+    return incremental + padding_idx
+
+
+class RobertaModel(BertModel):
+    """BERT trunk with RoBERTa position-id semantics."""
+
+    @configurable
+    def __init__(self, pad_token_id=1, **kwargs):
+        super(RobertaModel, self).__init__(**kwargs)
+        self.pad_token_id = pad_token_id
+
+    @classmethod
+    def from_config(cls, cfg):
+        base = BertModel.from_config(cfg)
+        base["pad_token_id"] = cfg.get("pad_token_id", 1)
+        return base
+
+    def forward(self, input_ids, attention_mask=None, tokentype_ids=None):
+        from .bert_model import extended_attn_mask
+
+        mask = extended_attn_mask(attention_mask)
+        pos_ids = create_position_ids_from_input_ids(input_ids, self.pad_token_id)
+        emb = self.embeddings.vocab_embeddings(input_ids)
+        emb = emb + self.embeddings.position_embeddings(
+            pos_ids.clamp(max=self.embeddings.position_embeddings.num_embeddings - 1)
+        )
+        if self.embeddings.tokentype_embeddings is not None:
+            if tokentype_ids is None:
+                tokentype_ids = torch.zeros_like(input_ids)
+            emb = emb + self.embeddings.tokentype_embeddings(tokentype_ids)
+        h = self.embeddings.embedding_dropout(emb)
+        for layer in self.layers:
+            h = self._run_layer(layer, h, mask)
+        h = self.final_layernorm(h)
+        pooled = self.pooler(h) if self.pooler is not None else None
+        return h, pooled
+
+
+class RobertaLoss(nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.lm_loss = ParallelCrossEntropyLoss()
+
+    def forward(self, logits, lm_labels, loss_mask):
+        loss = self.lm_loss(logits, lm_labels.clamp(min=0)).view(-1)
+        lm = loss * loss_mask.float().view(-1)
+        return {"lm_loss": lm.sum() / loss_mask.float().sum().clamp(min=1.0)}
+
+
+class RobertaForPreTraining(nn.Module):
+    @configurable
+    def __init__(self, cfg=None, **kwargs):
+        super().__init__()
+        if cfg is not None:
+            self.roberta = RobertaModel(cfg)
+            hidden = cfg.hidden_size
+            vocab = cfg.vocab_size
+            init_method = init_method_normal(cfg.get("initializer_range", 0.02))
+            eps = cfg.get("layernorm_eps", 1e-5)
+        else:
+            self.roberta = RobertaModel(**kwargs)
+            hidden = kwargs["hidden_size"]
+            vocab = kwargs["vocab_size"]
+            init_method = init_method_normal(kwargs.get("initializer_range", 0.02))
+            eps = kwargs.get("layernorm_eps", 1e-5)
+        self.lm_head = BertLMPredictionHead(hidden, init_method, eps, layer_idx=-1)
+        self.lm_logits = LMLogits(vocab, bias=True, layer_idx=-1)
+        self.loss_func = RobertaLoss()
+
+    @classmethod
+    def from_config(cls, cfg):
+        return {"cfg": cfg}
+
+    def forward(self, input_ids, attention_mask=None, tokentype_ids=None,
+                lm_labels=None, loss_mask=None):
+        seq_out, _ = self.roberta(input_ids, attention_mask, tokentype_ids)
+        h = self.lm_head(seq_out)
+        logits = self.lm_logits(h, self.roberta.word_embeddings_weight)
+        if lm_labels is not None and loss_mask is not None:
+            return self.loss_func(logits, lm_labels, loss_mask)
+        return {"prediction_scores": logits}
+
+    def set_activation_checkpoint(self, enabled=True):
+        self.roberta.set_activation_checkpoint(enabled)
+
+
+RobertaForCausalLM = RobertaForPreTraining  # reference exposes a CLM variant alias

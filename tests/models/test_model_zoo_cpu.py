@@ -98,3 +98,62 @@ def test_swiglu_reference():
     assert torch.allclose(y, ref, atol=1e-6)
     y.sum().backward()
     assert torch.isfinite(x.grad).all()
+
+
+def test_t5_fwd_bwd_and_masks():
+    from libai_amd.models import T5ForPreTraining
+
+    torch.manual_seed(0)
+    m = T5ForPreTraining(
+        vocab_size=128, hidden_size=32, hidden_layers=2, num_attention_heads=4,
+        intermediate_size=64, max_position_embeddings=32,
+        hidden_dropout_prob=0.0, attention_probs_dropout_prob=0.0,
+        embedding_dropout_prob=0.0,
+    )
+    b, se, sd = 2, 12, 8
+    enc = torch.randint(0, 128, (b, se))
+    dec = torch.randint(0, 128, (b, sd))
+    enc_mask = torch.ones(b, se, dtype=torch.uint8)
+    enc_mask[:, -2:] = 0
+    labels = torch.randint(0, 128, (b, sd))
+    loss_mask = torch.ones(b, sd, dtype=torch.long)
+    out = m(encoder_input_ids=enc, decoder_input_ids=dec, encoder_attn_mask=enc_mask,
+            lm_labels=labels, loss_mask=loss_mask)
+    assert torch.isfinite(out["masked_lm_loss"])
+    out["masked_lm_loss"].backward()
+
+
+def test_roberta_fwd_bwd():
+    from libai_amd.models import RobertaForPreTraining
+
+    torch.manual_seed(0)
+    m = RobertaForPreTraining(
+        vocab_size=128, hidden_size=32, hidden_layers=2, num_attention_heads=4,
+        intermediate_size=64, max_position_embeddings=34, num_tokentypes=0,
+        hidden_dropout_prob=0.0, attention_probs_dropout_prob=0.0,
+        add_pooling_layer=False,
+    )
+    ids = torch.randint(2, 128, (2, 16))
+    labels = torch.randint(0, 128, (2, 16))
+    loss_mask = torch.ones(2, 16, dtype=torch.long)
+    out = m(input_ids=ids, attention_mask=torch.ones(2, 16, dtype=torch.uint8),
+            lm_labels=labels, loss_mask=loss_mask)
+    out["lm_loss"].backward()
+
+
+def test_resmlp_fwd_bwd():
+    from libai_amd.models import ResMLP
+
+    m = ResMLP(img_size=32, patch_size=8, embed_dim=32, depth=2, num_classes=10)
+    out = m(images=torch.randn(2, 3, 32, 32), labels=torch.randint(0, 10, (2,)))
+    out["losses"].backward()
+
+
+def test_swin_fwd_bwd():
+    from libai_amd.models import SwinTransformer
+
+    m = SwinTransformer(img_size=32, patch_size=4, embed_dim=24, depths=(1, 1),
+                        num_heads=(2, 4), window_size=4, num_classes=10)
+    out = m(images=torch.randn(2, 3, 32, 32), labels=torch.randint(0, 10, (2,)))
+    out["losses"].backward()
+    assert torch.isfinite(out["losses"])
