@@ -89,6 +89,22 @@ def build(force=False, verbose=True):
         _run(cmd)
     objs.append(bind_obj)
 
+    # standalone pybind11 data-helpers extension (no torch dependency)
+    dh_src = os.path.join(CSRC, "data_helpers.cpp")
+    dh_so = os.path.join(PKG, "_data_helpers.so")
+    if os.path.exists(dh_src) and (force or not _newer(dh_so, dh_src)):
+        import pybind11
+
+        ext_suffix = sysconfig.get_config_var("EXT_SUFFIX") or ".so"
+        _run(
+            [
+                "g++", "-O3", "-std=c++17", "-fPIC", "-shared",
+                "-isystem", pybind11.get_include(),
+                "-isystem", py_include,
+                dh_src, "-o", dh_so,
+            ]
+        )
+
     torch_lib = cpp_ext.library_paths()[0]
     if force or not _newer(OUT_SO, *objs):
         _run(
