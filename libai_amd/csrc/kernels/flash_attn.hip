@@ -344,13 +344,18 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
     int causal) {
   constexpr int DT = D / 32;
   constexpr int KC = D / 16;
+  // D=64: 64-row staging tiles + persistent K/V fragments.  D=128: 32-row
+  // tiles and K/V fragments re-read from global per use (L2-resident) --
+  // the persistent variant spills 78 VGPRs.
+  constexpr int QTILE = (D == 64) ? 2 * QB : QB;
+  constexpr int NSUB = QTILE / QB;
 
-  __shared__ __align__(16) bf16_t q_row[2 * QB * D];
-  __shared__ __align__(16) bf16_t q_tr[2 * QB * D];
-  __shared__ __align__(16) bf16_t do_row[2 * QB * D];
-  __shared__ __align__(16) bf16_t do_tr[2 * QB * D];
-  __shared__ float lse_lds[2 * QB];
-  __shared__ float drow_lds[2 * QB];
+  __shared__ __align__(16) bf16_t q_row[QTILE * D];
+  __shared__ __align__(16) bf16_t q_tr[QTILE * D];
+  __shared__ __align__(16) bf16_t do_row[QTILE * D];
+  __shared__ __align__(16) bf16_t do_tr[QTILE * D];
+  __shared__ float lse_lds[QTILE];
+  __shared__ float drow_lds[QTILE];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -369,16 +374,25 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
   const bf16_t* vp = v + b * v_sb + h * v_sh;
   const bf16_t* dop = dout + b * do_sb + h * do_sh;
 
-  // persistent per-lane K and V rows (B-fragments: lane j = kv)
-  bf16x8_t kf[KC], vf[KC];
-  {
-    const int kvr = min(kvg, Sk - 1);
+  // per-lane K and V row fragments (B-operands: lane j = kv).  Persistent in
+  // registers for D=64; re-read from global (L2) per use for D=128.
+  const int kvr_ld = min(kvg, Sk - 1);
+  bf16x8_t kf[D == 64 ? KC : 1], vf[D == 64 ? KC : 1];
+  if constexpr (D == 64) {
 #pragma unroll
     for (int c = 0; c < KC; ++c) {
-      kf[c] = *(const bf16x8_t*)(kp + (int64_t)kvr * k_ss + c * 16 + hi * 8);
-      vf[c] = *(const bf16x8_t*)(vp + (int64_t)kvr * v_ss + c * 16 + hi * 8);
+      kf[c] = *(const bf16x8_t*)(kp + (int64_t)kvr_ld * k_ss + c * 16 + hi * 8);
+      vf[c] = *(const bf16x8_t*)(vp + (int64_t)kvr_ld * v_ss + c * 16 + hi * 8);
     }
   }
+  auto get_kf = [&](int c) {
+    if constexpr (D == 64) return kf[c];
+    else return *(const bf16x8_t*)(kp + (int64_t)kvr_ld * k_ss + c * 16 + hi * 8);
+  };
+  auto get_vf = [&](int c) {
+    if constexpr (D == 64) return vf[c];
+    else return *(const bf16x8_t*)(vp + (int64_t)kvr_ld * v_ss + c * 16 + hi * 8);
+  };
 
   f32x16_t dk_acc[DT], dv_acc[DT];
 #pragma unroll
@@ -388,25 +402,25 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
   }
 
   const float ks = (p_drop > 0.f) ? 1.0f / (1.0f - p_drop) : 1.0f;
-  const int qt_start = causal ? (kv_block / (2 * QB)) : 0;
-  const int n_qtiles = CDIV(Sq, 2 * QB);
+  const int qt_start = causal ? (kv_block / QTILE) : 0;
+  const int n_qtiles = CDIV(Sq, QTILE);
 
   for (int qt = qt_start; qt < n_qtiles; ++qt) {
-    const int q0 = qt * 2 * QB;  // 64 q rows staged per tile
+    const int q0 = qt * QTILE;
     __syncthreads();
-    stage_tile<2 * QB, D, true, true>(q_row, q_tr, qp, q0, Sq, q_ss, tid);
-    stage_tile<2 * QB, D, true, true>(do_row, do_tr, dop, q0, Sq, do_ss, tid);
-    if (tid < 2 * QB) {
+    stage_tile<QTILE, D, true, true>(q_row, q_tr, qp, q0, Sq, q_ss, tid);
+    stage_tile<QTILE, D, true, true>(do_row, do_tr, dop, q0, Sq, do_ss, tid);
+    if (tid < QTILE) {
       int qr = min(q0 + tid, Sq - 1);
       lse_lds[tid] = lse[(int64_t)bh * Sq + qr];
       drow_lds[tid] = drow[(int64_t)bh * Sq + qr];
     }
     __syncthreads();
 
-    if (causal && q0 + 2 * QB - 1 < kv_base) continue;  // entirely above diag
+    if (causal && q0 + QTILE - 1 < kv_base) continue;  // entirely above diag
 
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
+    for (int sub = 0; sub < NSUB; ++sub) {
       const int q0s = q0 + sub * QB;
       if (causal && q0s + QB - 1 < kv_base) continue;
       const int ro = sub * QB;  // row offset inside the staged images
@@ -417,9 +431,9 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
 #pragma unroll
       for (int c = 0; c < KC; ++c) {
         bf16x8_t qa = row_img_frag<D>(q_row, ro + l31, c, hi);
-        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, kf[c], s, 0, 0, 0);
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, get_kf(c), s, 0, 0, 0);
         bf16x8_t da = row_img_frag<D>(do_row, ro + l31, c, hi);
-        dpd = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, vf[c], dpd, 0, 0, 0);
+        dpd = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, get_vf(c), dpd, 0, 0, 0);
       }
 
       // per-reg: q = q0s + pattern(r); kv = lane's kvg
@@ -492,10 +506,12 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
     int causal) {
   constexpr int DT = D / 32;
   constexpr int KC = D / 16;
+  constexpr int KVTILE = (D == 64) ? 2 * QB : QB;
+  constexpr int NSUB = KVTILE / QB;
 
-  __shared__ __align__(16) bf16_t k_row[2 * QB * D];
-  __shared__ __align__(16) bf16_t k_tr[2 * QB * D];
-  __shared__ __align__(16) bf16_t v_row[2 * QB * D];
+  __shared__ __align__(16) bf16_t k_row[KVTILE * D];
+  __shared__ __align__(16) bf16_t k_tr[KVTILE * D];
+  __shared__ __align__(16) bf16_t v_row[KVTILE * D];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -514,16 +530,25 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
   const bf16_t* vp = v + b * v_sb + h * v_sh;
   const bf16_t* dop = dout + b * do_sb + h * do_sh;
 
-  // persistent per-lane Q and dO rows (B-fragments: lane j = q)
-  bf16x8_t qf[KC], dof[KC];
-  {
-    const int qr = min(qg, Sq - 1);
+  // per-lane Q and dO row fragments (B-operands: lane j = q); persistent for
+  // D=64, re-read from global (L2) per use for D=128 (register budget).
+  const int qr_ld = min(qg, Sq - 1);
+  bf16x8_t qf[D == 64 ? KC : 1], dof[D == 64 ? KC : 1];
+  if constexpr (D == 64) {
 #pragma unroll
     for (int c = 0; c < KC; ++c) {
-      qf[c] = *(const bf16x8_t*)(qp + (int64_t)qr * q_ss + c * 16 + hi * 8);
-      dof[c] = *(const bf16x8_t*)(dop + (int64_t)qr * do_ss + c * 16 + hi * 8);
+      qf[c] = *(const bf16x8_t*)(qp + (int64_t)qr_ld * q_ss + c * 16 + hi * 8);
+      dof[c] = *(const bf16x8_t*)(dop + (int64_t)qr_ld * do_ss + c * 16 + hi * 8);
     }
   }
+  auto get_qf = [&](int c) {
+    if constexpr (D == 64) return qf[c];
+    else return *(const bf16x8_t*)(qp + (int64_t)qr_ld * q_ss + c * 16 + hi * 8);
+  };
+  auto get_dof = [&](int c) {
+    if constexpr (D == 64) return dof[c];
+    else return *(const bf16x8_t*)(dop + (int64_t)qr_ld * do_ss + c * 16 + hi * 8);
+  };
   const float lse_lane = lse[(int64_t)bh * Sq + min(qg, Sq - 1)];
   const float drow_lane = drow[(int64_t)bh * Sq + min(qg, Sq - 1)];
 
@@ -533,19 +558,19 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
 
   const float ks = (p_drop > 0.f) ? 1.0f / (1.0f - p_drop) : 1.0f;
   const int kv_end = causal ? min(Sk, q_block + QBLK) : Sk;
-  const int n_tiles = CDIV(kv_end, 2 * QB);
+  const int n_tiles = CDIV(kv_end, KVTILE);
 
   for (int tile = 0; tile < n_tiles; ++tile) {
-    const int kv0 = tile * 2 * QB;  // 64 kv rows staged per tile
+    const int kv0 = tile * KVTILE;
     __syncthreads();
-    stage_tile<2 * QB, D, true, true>(k_row, k_tr, kp, kv0, Sk, k_ss, tid);
-    stage_tile<2 * QB, D, true, false>(v_row, nullptr, vp, kv0, Sk, v_ss, tid);
+    stage_tile<KVTILE, D, true, true>(k_row, k_tr, kp, kv0, Sk, k_ss, tid);
+    stage_tile<KVTILE, D, true, false>(v_row, nullptr, vp, kv0, Sk, v_ss, tid);
     __syncthreads();
 
     if (causal && kv0 > q_base + QB - 1) continue;
 
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
+    for (int sub = 0; sub < NSUB; ++sub) {
       const int kv0s = kv0 + sub * QB;
       if (kv0s >= kv_end || (causal && kv0s > q_base + QB - 1)) continue;
       const int ro = sub * QB;
@@ -556,9 +581,9 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
 #pragma unroll
       for (int c = 0; c < KC; ++c) {
         bf16x8_t ka = row_img_frag<D>(k_row, ro + l31, c, hi);
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qf[c], st, 0, 0, 0);
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, get_qf(c), st, 0, 0, 0);
         bf16x8_t va = row_img_frag<D>(v_row, ro + l31, c, hi);
-        dpdt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, dof[c], dpdt, 0, 0, 0);
+        dpdt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, get_dof(c), dpdt, 0, 0, 0);
       }
 
       float ds[16];
