@@ -84,7 +84,7 @@ def main():
     device = du.get_device()
     if device.type == "cuda":
         torch.cuda.set_device(device)
-    torch.manual_seed(du.same_seed_for_tp_group(1234))
+    torch.manual_seed(1234)  # identical init across DP ranks
 
     dp = dutil.data_parallel_size
     micro = args.micro_batch
@@ -130,6 +130,8 @@ def main():
         weight_decay=0.01,
         clip_grad=1.0,
     )
+
+    torch.manual_seed(du.same_seed_for_tp_group(1234))  # diverge dropout per dp rank
 
     # synthetic batches, pre-generated on host, moved in-step
     gen = torch.Generator().manual_seed(4321 + dutil.data_parallel_rank)

@@ -79,8 +79,11 @@ def default_setup(cfg, args=None):
     if args is not None:
         logger.info(f"Command line arguments: {args}")
 
+    # model init must be IDENTICAL across DP ranks (and TP ranks slice one
+    # full init): seed with the BASE seed here; DefaultTrainer re-seeds with
+    # the per-dp-rank stream after the model is built so dropout diverges.
     seed = try_get_key(cfg, "train.seed", default=1234)
-    torch.manual_seed(du.same_seed_for_tp_group(seed))
+    torch.manual_seed(seed)
     import random
 
     import numpy as np
@@ -105,6 +108,11 @@ class DefaultTrainer(TrainerBase):
         self.logger = logger
 
         dutil = du.get_dist_util()
+        if (
+            try_get_key(cfg, "train.global_batch_size", default=None) is None
+            or try_get_key(cfg, "train.num_accumulation_steps", default=None) is None
+        ):
+            _check_batch_size(cfg)
         micro, glob, acc = (
             cfg.train.train_micro_batch_size,
             cfg.train.global_batch_size,
@@ -145,6 +153,12 @@ class DefaultTrainer(TrainerBase):
         # flat-bucket optimizer freezes param storage)
         self.optimizer = self.build_optimizer(cfg, self.model)
         self.lr_scheduler = self.build_lr_scheduler(cfg, self.optimizer)
+
+        # diverge the RNG stream per DP rank now that init is done (dropout
+        # masks must differ across DP, stay identical within a TP group)
+        torch.manual_seed(
+            du.same_seed_for_tp_group(try_get_key(cfg, "train.seed", default=1234))
+        )
 
         self._trainer = EagerTrainer(
             self.model, self.train_loader, self.optimizer, acc,

@@ -161,6 +161,12 @@ class FusedAdamW(torch.optim.Optimizer):
                     (gi, _Bucket(plist, dtype, device, wd_on, tp_sharded, dp, dpr,
                                  zero))
                 )
+        # DP replicas must start bit-identical; broadcast once from dp rank 0
+        if dp > 1 and dist.is_initialized():
+            for _, b in self._buckets:
+                src_rank = dist.get_global_rank(dutil.data_parallel_group, 0)
+                dist.broadcast(b.flat_param, src=src_rank,
+                               group=dutil.data_parallel_group)
 
     @property
     def buckets(self):
