@@ -41,6 +41,9 @@ def parse_args():
     p.add_argument("--vocab", type=int, default=50304)
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--act-ckpt", action="store_true")
+    p.add_argument("--model", default="gpt2",
+                   choices=["gpt2", "bert-large", "llama7b", "llama1b", "vit-l16"],
+                   help="flagship benchmark model family")
     return p.parse_args()
 
 
@@ -60,6 +63,78 @@ def _enable_tuned_gemms():
         os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
         os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
         os.environ["PYTORCH_TUNABLEOP_FILENAME"] = os.path.join(d, "tunableop.csv")
+
+
+def build_bench_model(args):
+    """Construct the benchmark model for --model; gpt2 honors the shape flags
+    (the BASELINE contract); the others are the fixed BASELINE.json configs."""
+    if args.model == "gpt2":
+        from libai_amd.models import GPTForPreTraining
+
+        m = GPTForPreTraining(
+            hidden_layers=args.layers, vocab_size=args.vocab,
+            hidden_size=args.hidden, ffn_hidden_size=4 * args.hidden,
+            num_attention_heads=args.heads, max_seq_length=args.seq_len,
+            embedding_dropout_prob=0.1, attention_dropout_prob=0.1,
+            output_dropout_prob=0.1,
+        )
+        return m, f"gpt2-345m (nl{args.layers} h{args.hidden} nah{args.heads} seq{args.seq_len})"
+    if args.model == "bert-large":
+        from libai_amd.models import BertForPreTraining
+
+        args.seq_len = 512
+        args.vocab = 30592
+        args.layers = 24
+        m = BertForPreTraining(
+            vocab_size=args.vocab, hidden_size=1024, hidden_layers=24,
+            num_attention_heads=16, intermediate_size=4096,
+            max_position_embeddings=512,
+        )
+        return m, "bert-large (nl24 h1024 nah16 seq512)"
+    if args.model in ("llama7b", "llama1b"):
+        from libai_amd.models import LlamaForCausalLM
+
+        if args.model == "llama7b":
+            shape = dict(hidden_layers=32, hidden_size=4096, intermediate_size=11008,
+                         num_attention_heads=32)
+        else:
+            shape = dict(hidden_layers=16, hidden_size=2048, intermediate_size=5504,
+                         num_attention_heads=16)
+        args.seq_len = 2048
+        args.vocab = 32000
+        args.layers = shape["hidden_layers"]
+        m = LlamaForCausalLM(vocab_size=args.vocab,
+                             max_position_embeddings=args.seq_len, **shape)
+        return m, f"{args.model} (seq2048)"
+    if args.model == "vit-l16":
+        from libai_amd.models import VisionTransformer
+
+        args.seq_len = 197  # tokens per image (for tokens/s accounting)
+        args.layers = 24
+        m = VisionTransformer(img_size=224, patch_size=16, embed_dim=1024, depth=24,
+                              num_heads=16, num_classes=1000)
+        return m, "vit-l16 (224px patch16)"
+    raise ValueError(args.model)
+
+
+def make_batch(args, micro, gen, device):
+    if args.model == "vit-l16":
+        img = torch.randn(micro, 3, 224, 224, generator=gen)
+        lbl = torch.randint(0, 1000, (micro,), generator=gen)
+        return {"images": img, "labels": lbl}
+    if args.model == "bert-large":
+        ids = torch.randint(5, args.vocab, (micro, args.seq_len), generator=gen)
+        mask_pos = torch.rand(micro, args.seq_len, generator=gen) < 0.15
+        return {
+            "input_ids": ids,
+            "attention_mask": torch.ones(micro, args.seq_len, dtype=torch.uint8),
+            "ns_labels": torch.randint(0, 2, (micro,), generator=gen),
+            "lm_labels": ids.clone(),
+            "loss_mask": mask_pos.long(),
+        }
+    toks = torch.randint(0, args.vocab, (micro, args.seq_len + 1), generator=gen)
+    return {"input_ids": toks[:, :-1].contiguous(),
+            "labels": toks[:, 1:].contiguous()}
 
 
 def main():
@@ -96,21 +171,10 @@ def main():
     if device.type == "cpu" and args.dtype == "bf16":
         dtype = torch.float32  # CPU smoke runs in fp32
 
-    # model: GPT-2 345M-class random init
-    from libai_amd.models import GPTForPreTraining
+    # model: random init at the named benchmark shape
     from libai_amd.optim import FusedAdamW, get_default_optimizer_params
 
-    model = GPTForPreTraining(
-        hidden_layers=args.layers,
-        vocab_size=args.vocab,
-        hidden_size=args.hidden,
-        ffn_hidden_size=4 * args.hidden,
-        num_attention_heads=args.heads,
-        max_seq_length=args.seq_len,
-        embedding_dropout_prob=0.1,
-        attention_dropout_prob=0.1,
-        output_dropout_prob=0.1,
-    )
+    model, model_name = build_bench_model(args)
     if args.act_ckpt:
         model.set_activation_checkpoint(True)
     model = model.to(dtype)
@@ -137,17 +201,11 @@ def main():
     gen = torch.Generator().manual_seed(4321 + dutil.data_parallel_rank)
     pool = []
     for _ in range(4):
-        toks = torch.randint(0, args.vocab, (micro, args.seq_len + 1), generator=gen)
-        pool.append(
-            {
-                "input_ids": toks[:, :-1].contiguous().pin_memory()
-                if device.type == "cuda"
-                else toks[:, :-1].contiguous(),
-                "labels": toks[:, 1:].contiguous().pin_memory()
-                if device.type == "cuda"
-                else toks[:, 1:].contiguous(),
-            }
-        )
+        b = make_batch(args, micro, gen, device)
+        if device.type == "cuda":
+            b = {k: (v.pin_memory() if v.is_floating_point() or v.dtype == torch.long
+                     or v.dtype == torch.uint8 else v) for k, v in b.items()}
+        pool.append(b)
 
     import torch.distributed as dist
 
@@ -201,7 +259,7 @@ def main():
     tokens = args.steps * global_batch * args.seq_len
     tokens_per_s = tokens / elapsed
     ms_per_step = elapsed / args.steps * 1000
-    base = BASELINE_TOKENS_PER_S.get(n)
+    base = BASELINE_TOKENS_PER_S.get(n) if args.model == "gpt2" else None
     par = f"dp{dp}"
     if args.tp > 1:
         par += f"_tp{args.tp}"
@@ -212,7 +270,11 @@ def main():
         print(
             json.dumps(
                 {
-                    "metric": "tokens/sec (whole node), GPT-2 345M 3D-parallel at 1/2/4/8 MI355X",
+                    "metric": (
+                        "tokens/sec (whole node), GPT-2 345M 3D-parallel at 1/2/4/8 MI355X"
+                        if args.model == "gpt2"
+                        else f"tokens/sec (whole node), {model_name}"
+                    ),
                     "value": round(tokens_per_s, 1),
                     "unit": "tokens/s",
                     "n_gpus": n,
@@ -225,7 +287,7 @@ def main():
                     "dtype": args.dtype if device.type == "cuda" else "fp32",
                     "data": "synthetic",
                     "config": {
-                        "model": "gpt2-345m (nl24 h1024 nah16 seq1024)",
+                        "model": model_name,
                         "global_batch": global_batch,
                         "seq_len": args.seq_len,
                         "parallelism": par,
