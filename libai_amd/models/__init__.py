@@ -5,6 +5,7 @@ from .llama import LlamaForCausalLM, LlamaModel
 from .resmlp import ResMLP
 from .roberta_model import RobertaForCausalLM, RobertaForPreTraining, RobertaModel
 from .swin_transformer import SwinTransformer
+from .swin_transformer_v2 import SwinTransformerV2
 from .t5_model import T5ForPreTraining, T5Model
 from .vision_transformer import VisionTransformer
 
@@ -22,6 +23,7 @@ __all__ = [
     "LlamaForCausalLM",
     "VisionTransformer",
     "SwinTransformer",
+    "SwinTransformerV2",
     "ResMLP",
     "build_model",
 ]

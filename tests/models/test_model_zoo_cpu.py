@@ -157,3 +157,13 @@ def test_swin_fwd_bwd():
     out = m(images=torch.randn(2, 3, 32, 32), labels=torch.randint(0, 10, (2,)))
     out["losses"].backward()
     assert torch.isfinite(out["losses"])
+
+
+def test_swin_v2_fwd_bwd():
+    from libai_amd.models import SwinTransformerV2
+
+    m = SwinTransformerV2(img_size=32, patch_size=4, embed_dim=24, depths=(1, 1),
+                          num_heads=(2, 4), window_size=4, num_classes=10)
+    out = m(images=torch.randn(2, 3, 32, 32), labels=torch.randint(0, 10, (2,)))
+    out["losses"].backward()
+    assert torch.isfinite(out["losses"])
