@@ -9,6 +9,7 @@ from .common.models.gpt import cfg as gpt_cfg
 from .common.models.gpt import pretrain_model as model
 from .common.optim import optim
 from .common.train import train
+from libai_amd.scheduler import WarmupCosineLR
 
 # GPT-2 345M-class
 gpt_cfg.hidden_layers = 24
@@ -28,6 +29,13 @@ dataloader = dict(
         train_batch_size=4,
         num_workers=2,
     ),
+)
+
+train.scheduler = LazyCall(WarmupCosineLR)(
+    max_iter=1000,
+    warmup_iter=100,
+    warmup_factor=0.001,
+    alpha=0.1,
 )
 
 train.update(

@@ -171,6 +171,19 @@ class DefaultTrainer(TrainerBase):
             optimizer=self.optimizer,
             lr_scheduler=self.lr_scheduler,
         )
+        # reference auto-scales epochs -> iters (default.py:695-774)
+        train_epoch = try_get_key(cfg, "train.train_epoch", default=0) or 0
+        if train_epoch > 0 and self.train_loader is not None and hasattr(
+            self.train_loader, "dataset"
+        ):
+            iters_per_epoch = max(len(self.train_loader.dataset) // glob, 1)
+            cfg.train.train_iter = max(
+                int(train_epoch * iters_per_epoch), cfg.train.train_iter or 0
+            )
+            logger.info(
+                f"auto-scaled train_epoch={train_epoch} -> train_iter="
+                f"{cfg.train.train_iter} ({iters_per_epoch} iters/epoch)"
+            )
         self.max_iter = cfg.train.train_iter
         self.global_batch_size = glob
         self.start_iter = 0

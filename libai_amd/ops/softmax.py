@@ -63,9 +63,10 @@ def fused_scale_mask_softmax(scores, pad_mask=None, scale=1.0, p=0.0, causal=Tru
         return _ScaleMaskSoftmaxFn.apply(
             scores, pad_mask, scale, p if training else 0.0, causal
         )
-    if scores.is_cuda:
+    if scores.is_cuda and training and scores.requires_grad:
         raise RuntimeError(
-            f"fused_scale_mask_softmax HIP path requires SK % 8 == 0 and SK <= 8192, "
-            f"got SK={scores.shape[-1]}"
+            f"fused_scale_mask_softmax HIP path requires SK % 8 == 0 and SK <= 8192 "
+            f"for training, got SK={scores.shape[-1]}"
         )
+    # odd-width decode steps (KV-cache generation) use the composed path
     return _ref(scores, pad_mask, scale, p, causal, training)
