@@ -1,3 +1,4 @@
+from .bleu_evaluator import BleuEvaluator, corpus_bleu
 from .cls_evaluator import ClsEvaluator
 from .evaluator import (
     DatasetEvaluator,
@@ -6,6 +7,7 @@ from .evaluator import (
     inference_on_dataset,
 )
 from .ppl_evaluator import PPLEvaluator
+from .reg_evaluator import RegEvaluator
 
 __all__ = [
     "DatasetEvaluator",
@@ -14,4 +16,7 @@ __all__ = [
     "flatten_results_dict",
     "ClsEvaluator",
     "PPLEvaluator",
+    "BleuEvaluator",
+    "corpus_bleu",
+    "RegEvaluator",
 ]

@@ -82,3 +82,54 @@ def test_blending_indices():
     for d in (0, 1):
         sel = np.asarray(dsi)[np.asarray(di) == d]
         assert np.array_equal(sel, np.arange(len(sel)))
+
+
+def _sentence_corpus(tmp_path):
+    prefix = str(tmp_path / "sents")
+    b = MMapIndexedDatasetBuilder(data_file_path(prefix), np.uint16)
+    rng = np.random.RandomState(0)
+    for d in range(8):  # 8 docs x 5 sentences
+        for s in range(5):
+            b.add_item(rng.randint(200, 5000, size=rng.randint(5, 20)).astype(np.uint16))
+        b.end_document()
+    b.finalize(index_file_path(prefix))
+    return MMapIndexedDataset(prefix)
+
+
+def test_bert_dataset_masked_lm(tmp_path):
+    from libai_amd.data.datasets import BertDataset
+
+    ds = _sentence_corpus(tmp_path)
+    bd = BertDataset("t", ds, max_seq_length=64, vocab_size=6000, num_samples=10)
+    assert len(bd) > 0
+    inst = bd[0]
+    ids = inst.get("input_ids").tensor
+    labels = inst.get("lm_labels").tensor
+    loss_mask = inst.get("loss_mask").tensor
+    assert ids.shape == (64,)
+    assert ids[0] == 101  # [CLS]
+    masked_positions = (loss_mask == 1).nonzero().flatten()
+    assert len(masked_positions) >= 1
+    # labels hold the ORIGINAL token at masked positions, -1 elsewhere
+    assert (labels[loss_mask == 0] == -1).all()
+    assert (labels[loss_mask == 1] >= 0).all()
+    # deterministic per index
+    inst2 = bd[0]
+    assert torch.equal(ids, inst2.get("input_ids").tensor)
+
+
+def test_t5_dataset_span_corruption(tmp_path):
+    from libai_amd.data.datasets import T5Dataset
+
+    ds = _sentence_corpus(tmp_path)
+    td = T5Dataset("t", ds, max_seq_length=64, max_seq_length_dec=32,
+                   vocab_size=6000, num_samples=8)
+    assert len(td) > 0
+    inst = td[0]
+    enc = inst.get("encoder_input_ids").tensor
+    dec_in = inst.get("decoder_input_ids").tensor
+    tgt = inst.get("lm_labels").tensor
+    assert enc.shape == (64,) and dec_in.shape == (32,) and tgt.shape == (32,)
+    # sentinels (counting down from vocab-1) appear in encoder and decoder
+    assert (enc >= 5900).any()
+    assert (dec_in >= 5900).any()
