@@ -6,6 +6,7 @@ from .build import (
     build_train_valid_test_loaders,
     trivial_batch_collator,
 )
+from .mixup import Mixup
 from .samplers import CyclicSampler, SingleRoundSampler
 from .structures import DistTensorData, Instance
 
@@ -13,6 +14,7 @@ __all__ = [
     "DistTensorData",
     "Instance",
     "CyclicSampler",
+    "Mixup",
     "SingleRoundSampler",
     "trivial_batch_collator",
     "build_nlp_train_loader",
