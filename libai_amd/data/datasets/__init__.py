@@ -1,4 +1,5 @@
 from .bert_dataset import BertDataset, create_masked_lm_predictions
+from .cv_datasets import CIFAR10Dataset, ImageFolderDataset, MNISTDataset
 from .gpt_dataset import GPT2Dataset
 from .synthetic import SyntheticBertDataset, SyntheticGPTDataset, SyntheticImageDataset
 from .t5_dataset import T5Dataset
@@ -8,6 +9,9 @@ __all__ = [
     "BertDataset",
     "T5Dataset",
     "create_masked_lm_predictions",
+    "CIFAR10Dataset",
+    "MNISTDataset",
+    "ImageFolderDataset",
     "SyntheticGPTDataset",
     "SyntheticBertDataset",
     "SyntheticImageDataset",

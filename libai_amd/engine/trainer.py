@@ -16,7 +16,7 @@ import torch
 from ..utils import distributed as du
 from ..utils.events import EventStorage
 
-__all__ = ["HookBase", "TrainerBase", "EagerTrainer"]
+__all__ = ["HookBase", "TrainerBase", "EagerTrainer", "GraphTrainer"]
 
 
 class HookBase:
@@ -203,3 +203,18 @@ class EagerTrainer(TrainerBase):
         self.optimizer.step()
         self.optimizer.zero_grad()
         self.write_metrics(loss_dict, data_time)
+
+
+class GraphTrainer(EagerTrainer):
+    """Compatibility name for the reference's compiled-graph trainer
+    (reference: libai/engine/trainer.py:305-349).
+
+    The reference's GraphTrainer exists because OneFlow needs an nn.Graph
+    compilation pass to enable AMP/ZeRO/1F1B.  Here those are explicit engine
+    features of the eager path (bf16 parameters, FusedAdamW ZeRO stages, the
+    1F1B scheduler), and the measured step is GPU-bound (kernel time == wall
+    time at the bench shapes), so a captured-graph replay would save only the
+    host launch overhead the hardware already hides.  hipGraph capture of the
+    steady-state step is a planned optimization gated on moving the dropout
+    seed draws device-side (a captured CPU seed would freeze the masks).
+    """
