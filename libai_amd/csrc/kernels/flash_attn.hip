@@ -181,12 +181,20 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
   const bf16_t* kp = k + b * k_sb + h * k_sh;
   const bf16_t* vp = v + b * v_sb + h * v_sh;
 
+  // Q fragments pre-scaled by scale*log2e: scores come out of the MFMA
+  // already in the exp2 domain, saving a mul per score element and the
+  // exp->exp2 conversion mul (softmax runs on v_exp_f32 directly).
+  const float qscale = scale * 1.4426950408889634f;
   bf16x8_t qfrag[KC];
   {
     const int qrow = min(q_base + l31, Sq - 1);
 #pragma unroll
-    for (int c = 0; c < KC; ++c)
-      qfrag[c] = *(const bf16x8_t*)(qp + (int64_t)qrow * q_ss + c * 16 + hi * 8);
+    for (int c = 0; c < KC; ++c) {
+      bf16x8_t raw = *(const bf16x8_t*)(qp + (int64_t)qrow * q_ss + c * 16 + hi * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) raw[j] = (bf16_t)((float)raw[j] * qscale);
+      qfrag[c] = raw;
+    }
   }
 
   f32x16_t oacc[DT];
@@ -218,13 +226,14 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
       s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb, qfrag[c], s1, 0, 0, 0);
     }
 
+    // scores are already in the exp2 domain (Q pre-scale); m/l run in it too
     float sv[2][16];
     float pmax = -3.0e38f;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int kva = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
       int kvb = kva + 32;
-      float a = s0[r] * scale, bb = s1[r] * scale;
+      float a = s0[r], bb = s1[r];
       if (kva >= Sk || (causal && kva > qg)) a = -3.0e38f;
       if (kvb >= Sk || (causal && kvb > qg)) bb = -3.0e38f;
       sv[0][r] = a;
@@ -234,14 +243,14 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
     pmax = fmaxf(pmax, __shfl_xor(pmax, 32));
 
     const float m_new = fmaxf(m_run, pmax);
-    const float alpha = (m_run <= -3.0e38f) ? 0.f : __expf(m_run - m_new);
+    const float alpha = (m_run <= -3.0e38f) ? 0.f : __builtin_amdgcn_exp2f(m_run - m_new);
     m_run = m_new;
     float lsum = 0.f;
 #pragma unroll
     for (int t = 0; t < 2; ++t)
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        float e = (sv[t][r] <= -3.0e38f) ? 0.f : __expf(sv[t][r] - m_new);
+        float e = (sv[t][r] <= -3.0e38f) ? 0.f : __builtin_amdgcn_exp2f(sv[t][r] - m_new);
         sv[t][r] = e;
         lsum += e;
       }
@@ -291,8 +300,9 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
         for (int j = 0; j < 4; ++j) pk[j] = f2bf(oacc[dt][r4 * 4 + j] * inv_l);
         *(u16x4*)(op + dt * 32 + 8 * r4 + 4 * hi) = pk;
       }
-    if (hi == 0 && lse != nullptr)
-      lse[((int64_t)bh * Sq) + qg] = m_run + __logf(l_run > 0.f ? l_run : 1.f);
+    if (hi == 0 && lse != nullptr)  // convert the exp2-domain max back to ln
+      lse[((int64_t)bh * Sq) + qg] =
+          m_run * 0.6931471805599453f + __logf(l_run > 0.f ? l_run : 1.f);
   }
 }
 
