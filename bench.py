@@ -194,6 +194,7 @@ def main():
         weight_decay=0.01,
         clip_grad=1.0,
     )
+    overlap = pipeline is None and optimizer.register_overlap_hooks()
 
     torch.manual_seed(du.same_seed_for_tp_group(1234))  # diverge dropout per dp rank
 
@@ -233,6 +234,8 @@ def main():
         else:
             for j in range(args.acc):
                 data = to_dev(pool[(step_i * args.acc + j) % len(pool)])
+                if overlap and j == args.acc - 1:
+                    optimizer.begin_overlap_step()
                 losses = model(**data)
                 total = sum(v for v in losses.values()) / args.acc
                 total.backward()

@@ -149,6 +149,9 @@ class EagerTrainer(TrainerBase):
         self.optimizer = optimizer
         self.grad_acc_steps = grad_acc_steps
         self.pipeline_scheduler = pipeline_scheduler
+        self._overlap = False
+        if pipeline_scheduler is None and hasattr(optimizer, "register_overlap_hooks"):
+            self._overlap = bool(optimizer.register_overlap_hooks())
 
     def get_batch(self, data):
         from ..data.structures import Instance
@@ -186,10 +189,12 @@ class EagerTrainer(TrainerBase):
         else:
             data_time = 0.0
             loss_dict = None
-            for _ in range(self.grad_acc_steps):
+            for micro in range(self.grad_acc_steps):
                 t0 = time.perf_counter()
                 data = self.get_batch(next(self._data_loader_iter))
                 data_time += time.perf_counter() - t0
+                if self._overlap and micro == self.grad_acc_steps - 1:
+                    self.optimizer.begin_overlap_step()
                 losses = self.model(**data)
                 losses = {k: v for k, v in losses.items() if v.requires_grad or v.is_floating_point()}
                 total = sum(losses.values()) / self.grad_acc_steps

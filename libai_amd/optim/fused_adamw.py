@@ -182,12 +182,17 @@ class FusedAdamW(torch.optim.Optimizer):
 
     def grad_sync(self):
         """Average gradients over the DP group; with ZeRO-2, reduce-scatter so
-        only the local slice is received."""
+        only the local slice is received.  If comm/compute overlap is active
+        (register_overlap_hooks + begin_overlap_step), this just drains the
+        in-flight chunk all-reduces and handles stragglers."""
         dutil = du.get_dist_util()
         dp = dutil.data_parallel_size
         if dp == 1 or not dist.is_initialized():
             return
         group = dutil.data_parallel_group
+        if getattr(self, "_overlap_active", False):
+            self._finish_overlap(group, dp)
+            return
         for _, b in self.buckets:
             b.flat_grad.div_(dp)
             if self._zero_eff >= 2:
@@ -195,6 +200,87 @@ class FusedAdamW(torch.optim.Optimizer):
                 dist.reduce_scatter_tensor(out, b.flat_grad, group=group)
             else:
                 dist.all_reduce(b.flat_grad, group=group)
+
+    # -- backward/comm overlap (bucketed async all-reduce, C5+C11) ----------
+    #
+    # Flat buckets are split into ~32M-element comm chunks following model
+    # order; a post-accumulate-grad hook fires the chunk's async all-reduce as
+    # soon as its last parameter's gradient lands, overlapping the remaining
+    # backward.  Enabled per-step by the trainer (the LAST micro-batch only,
+    # so gradient accumulation sees full sums).  ZeRO-2 keeps the simpler
+    # post-backward reduce-scatter.
+
+    OVERLAP_CHUNK = 32 * 1024 * 1024  # elements
+
+    def register_overlap_hooks(self):
+        dutil = du.get_dist_util()
+        if (dutil.data_parallel_size == 1 or not dist.is_initialized()
+                or self.zero_stage >= 2):
+            return False
+        self._overlap_active = False
+        self._chunks = []  # (bucket, start, end, param_ids)
+        self._param_chunk = {}
+        for _, b in self.buckets:
+            off = 0
+            cur_params, cur_start = [], 0
+            for p in b.params:
+                cur_params.append(id(p))
+                off += p.numel()
+                if off - cur_start >= self.OVERLAP_CHUNK:
+                    self._chunks.append([b, cur_start, off, set(cur_params)])
+                    cur_start, cur_params = off, []
+            if cur_params:
+                self._chunks.append([b, cur_start, off, set(cur_params)])
+        for ci, (bkt, s, e, pids) in enumerate(self._chunks):
+            for p in bkt.params:
+                if id(p) in pids:
+                    self._param_chunk[id(p)] = ci
+        self._pending = [set(c[3]) for c in self._chunks]
+        self._handles = {}
+
+        def make_hook():
+            def hook(p):
+                if not self._overlap_active:
+                    return
+                ci = self._param_chunk.get(id(p))
+                if ci is None:
+                    return
+                pend = self._pending[ci]
+                pend.discard(id(p))
+                if not pend and ci not in self._handles:
+                    dutil2 = du.get_dist_util()
+                    bkt, s, e, _ = self._chunks[ci]
+                    view = bkt.flat_grad[s:e]
+                    view.div_(dutil2.data_parallel_size)
+                    self._handles[ci] = dist.all_reduce(
+                        view, group=dutil2.data_parallel_group, async_op=True
+                    )
+            return hook
+
+        for _, b in self.buckets:
+            for p in b.params:
+                p.register_post_accumulate_grad_hook(make_hook())
+        return True
+
+    def begin_overlap_step(self):
+        """Arm the hooks for the FINAL micro-batch's backward."""
+        if not hasattr(self, "_chunks"):
+            return
+        self._pending = [set(c[3]) for c in self._chunks]
+        self._handles = {}
+        self._overlap_active = True
+
+    def _finish_overlap(self, group, dp):
+        for ci, (bkt, s, e, _) in enumerate(self._chunks):
+            if ci not in self._handles:  # straggler (param without grad flow)
+                view = bkt.flat_grad[s:e]
+                view.div_(dp)
+                self._handles[ci] = dist.all_reduce(view, group=group,
+                                                    async_op=True)
+        for h in self._handles.values():
+            h.wait()
+        self._overlap_active = False
+        self._handles = {}
 
     def _gather_params(self):
         dutil = du.get_dist_util()
