@@ -213,9 +213,12 @@ class FusedAdamW(torch.optim.Optimizer):
     OVERLAP_CHUNK = 32 * 1024 * 1024  # elements
 
     def register_overlap_hooks(self):
+        import os
+
         dutil = du.get_dist_util()
         if (dutil.data_parallel_size == 1 or not dist.is_initialized()
-                or self.zero_stage >= 2):
+                or self.zero_stage >= 2
+                or os.environ.get("LIBAI_NO_OVERLAP", "0") == "1"):
             return False
         self._overlap_active = False
         self._chunks = []  # (bucket, start, end, param_ids)
