@@ -219,6 +219,15 @@ class DefaultTrainer(TrainerBase):
         if try_get_key(cfg, "train.activation_checkpoint.enabled", default=False):
             if hasattr(model, "set_activation_checkpoint"):
                 model.set_activation_checkpoint(True)
+        if try_get_key(cfg, "train.lora.enabled", default=False):
+            from ..lora import apply_lora
+
+            apply_lora(
+                model,
+                r=try_get_key(cfg, "train.lora.r", default=8),
+                alpha=try_get_key(cfg, "train.lora.alpha", default=16),
+                dropout=try_get_key(cfg, "train.lora.dropout", default=0.0),
+            )
         logger = logging.getLogger(__name__)
         n_params = sum(p.numel() for p in model.parameters())
         logger.info(f"Model built: {n_params / 1e6:.1f}M local parameters")
