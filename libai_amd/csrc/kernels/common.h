@@ -142,18 +142,28 @@ __device__ __forceinline__ float u32_to_uniform(uint32_t x) {
   return (x >> 8) * (1.0f / 16777216.0f);
 }
 
-// Cheap per-element counter RNG (splitmix64 finalizer): used by attention
-// dropout where fwd and bwd kernels index elements in DIFFERENT lane layouts,
-// so a per-element (not per-4-group) generator keeps regeneration cheap on
-// both sides.
+// Cheap per-element counter RNG: used by attention dropout where fwd and bwd
+// kernels index elements in DIFFERENT lane layouts, so a per-element (not
+// per-4-group) generator keeps regeneration cheap on both sides.
+// 32-BIT ops only: 64-bit integer multiplies are emulated on gfx950 and a
+// splitmix64 here measured ~25 VALU/element (PMC: 104 issued instructions per
+// MFMA in flash_fwd).  This lowCbias mixer is ~9 VALU.
 __device__ __forceinline__ uint32_t rnd_hash(uint64_t seed, uint64_t idx) {
-  uint64_t x = seed + idx * 0x9E3779B97F4A7C15ull;
-  x ^= x >> 30;
-  x *= 0xBF58476D1CE4E5B9ull;
-  x ^= x >> 27;
-  x *= 0x94D049BB133111EBull;
-  x ^= x >> 31;
-  return (uint32_t)x;
+  uint32_t h = (uint32_t)seed ^ ((uint32_t)idx * 0x9E3779B9u) ^
+               ((uint32_t)(idx >> 32) * 0x85EBCA6Bu) ^ (uint32_t)(seed >> 32);
+  h ^= h >> 16;
+  h *= 0x7FEB352Du;
+  h ^= h >> 15;
+  h *= 0x846CA68Bu;
+  h ^= h >> 16;
+  return h;
+}
+
+// keep-decision without the float conversion: compare the hash against a
+// precomputed uint32 threshold = p * 2^32.
+__device__ __forceinline__ uint32_t drop_threshold_u32(float p) {
+  double t = (double)p * 4294967296.0;
+  return (t >= 4294967295.0) ? 0xFFFFFFFFu : (uint32_t)t;
 }
 
 // ---------------------------------------------------------------------------

@@ -40,13 +40,16 @@ typedef float f32x16_t __attribute__((ext_vector_type(16)));
 #define QBLK (QB * NW)
 #define KVB 64   // kv rows per LDS tile (forward)
 
+// native bf16 converts (v_cvt_pk_bf16_f32-class) instead of the bit-manip
+// RNE helper: the repack runs per score element and the manual rounding was
+// ~6 VALU per value.
 __device__ __forceinline__ uint32_t pack_bf16x2(float lo, float hi) {
   union {
     uint32_t u;
-    uint16_t h[2];
+    bf16_t h[2];
   } r;
-  r.h[0] = f2bf(lo);
-  r.h[1] = f2bf(hi);
+  r.h[0] = (bf16_t)lo;
+  r.h[1] = (bf16_t)hi;
   return r.u;
 }
 
@@ -141,12 +144,12 @@ __device__ __forceinline__ void stage_tile(bf16_t* row_lds, bf16_t* tr_lds,
   }
 }
 
-// dropout keep factor for score element (bh, q, kv)
+// dropout keep factor for score element (bh, q, kv); thr = p * 2^32
 __device__ __forceinline__ float drop_keep(uint64_t seed, uint64_t bh, int64_t Sq,
-                                           int64_t Sk, int q, int kv, float p,
+                                           int64_t Sk, int q, int kv, uint32_t thr,
                                            float ks) {
   uint64_t idx = (bh * Sq + q) * Sk + kv;
-  return (u32_to_uniform(rnd_hash(seed, idx)) > p) ? ks : 0.f;
+  return (rnd_hash(seed, idx) > thr) ? ks : 0.f;
 }
 
 // ===========================================================================
@@ -262,12 +265,13 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
       for (int r = 0; r < 16; ++r) oacc[t][r] *= alpha;
 
     if (p_drop > 0.f) {
+      const uint32_t thr = drop_threshold_u32(p_drop);
 #pragma unroll
       for (int t = 0; t < 2; ++t)
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           int kv = kv0 + t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          sv[t][r] *= drop_keep(seed, bh, Sq, Sk, qg, kv, p_drop, ks);
+          sv[t][r] *= drop_keep(seed, bh, Sq, Sk, qg, kv, thr, ks);
         }
     }
 
@@ -295,10 +299,10 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
       for (int r4 = 0; r4 < 4; ++r4) {
-        u16x4 pk;
+        bf16x4_t pk;
 #pragma unroll
-        for (int j = 0; j < 4; ++j) pk[j] = f2bf(oacc[dt][r4 * 4 + j] * inv_l);
-        *(u16x4*)(op + dt * 32 + 8 * r4 + 4 * hi) = pk;
+        for (int j = 0; j < 4; ++j) pk[j] = (bf16_t)(oacc[dt][r4 * 4 + j] * inv_l);
+        *(bf16x4_t*)(op + dt * 32 + 8 * r4 + 4 * hi) = pk;
       }
     if (hi == 0 && lse != nullptr)  // convert the exp2-domain max back to ln
       lse[((int64_t)bh * Sq) + qg] =
@@ -455,7 +459,8 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
         float p = 0.f;
         if (valid) p = __expf(s[r] * scale - lse_lds[qrow - q0]);
         float keep = (p_drop > 0.f && valid)
-                         ? drop_keep(seed, bh, Sq, Sk, qrow, kvg, p_drop, ks)
+                         ? drop_keep(seed, bh, Sq, Sk, qrow, kvg,
+                                     drop_threshold_u32(p_drop), ks)
                          : (valid ? 1.f : 0.f);
         pd[r] = p * keep;
         ds[r] = scale * p * (dpd[r] * keep - drow_lds[qrow - q0]);
@@ -489,14 +494,14 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
       for (int r4 = 0; r4 < 4; ++r4) {
-        u16x4 a, c;
+        bf16x4_t a, c;
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
-          a[j] = f2bf(dk_acc[dt][r4 * 4 + j]);
-          c[j] = f2bf(dv_acc[dt][r4 * 4 + j]);
+          a[j] = (bf16_t)dk_acc[dt][r4 * 4 + j];
+          c[j] = (bf16_t)dv_acc[dt][r4 * 4 + j];
         }
-        *(u16x4*)(dkp + dt * 32 + 8 * r4 + 4 * hi) = a;
-        *(u16x4*)(dvp + dt * 32 + 8 * r4 + 4 * hi) = c;
+        *(bf16x4_t*)(dkp + dt * 32 + 8 * r4 + 4 * hi) = a;
+        *(bf16x4_t*)(dvp + dt * 32 + 8 * r4 + 4 * hi) = c;
       }
   }
 }
@@ -603,7 +608,8 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
         const bool valid = kv < Sk && qg < Sq && (!causal || kv <= qg);
         float p = valid ? __expf(st[r] * scale - lse_lane) : 0.f;
         float keep = (p_drop > 0.f && valid)
-                         ? drop_keep(seed, bh, Sq, Sk, qg, kv, p_drop, ks)
+                         ? drop_keep(seed, bh, Sq, Sk, qg, kv,
+                                     drop_threshold_u32(p_drop), ks)
                          : (valid ? 1.f : 0.f);
         ds[r] = valid ? scale * p * (dpdt[r] * keep - drow_lane) : 0.f;
       }
@@ -629,10 +635,10 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
       for (int r4 = 0; r4 < 4; ++r4) {
-        u16x4 a;
+        bf16x4_t a;
 #pragma unroll
-        for (int j = 0; j < 4; ++j) a[j] = f2bf(dq_acc[dt][r4 * 4 + j]);
-        *(u16x4*)(dqp + dt * 32 + 8 * r4 + 4 * hi) = a;
+        for (int j = 0; j < 4; ++j) a[j] = (bf16_t)dq_acc[dt][r4 * 4 + j];
+        *(bf16x4_t*)(dqp + dt * 32 + 8 * r4 + 4 * hi) = a;
       }
   }
 }
@@ -643,9 +649,10 @@ __global__ void attn_dropout_apply_kernel(typename E::T* __restrict__ x, int64_t
                                           int64_t Sq, int64_t Sk, float p,
                                           float keep_scale, uint64_t seed) {
   const int64_t total = BH * Sq * Sk;
+  const uint32_t thr = drop_threshold_u32(p);
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
-    float kp = (u32_to_uniform(rnd_hash(seed, (uint64_t)i)) > p) ? keep_scale : 0.f;
+    float kp = (rnd_hash(seed, (uint64_t)i) > thr) ? keep_scale : 0.f;
     x[i] = E::from_f(E::to_f(x[i]) * kp);
   }
 }
