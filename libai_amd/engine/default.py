@@ -287,6 +287,8 @@ class DefaultTrainer(TrainerBase):
             ),
         ]
         eval_period = try_get_key(cfg, "train.evaluation.eval_period", default=0)
+        if du.get_dist_util().pipeline_parallel_size > 1:
+            eval_period = 0  # eval loop runs the full forward; stages are pruned
         if eval_period and try_get_key(cfg, "train.evaluation.enabled", default=True):
             def _eval():
                 return self.test(self.cfg, model=self.model)
