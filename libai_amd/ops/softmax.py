@@ -64,9 +64,17 @@ def fused_scale_mask_softmax(scores, pad_mask=None, scale=1.0, p=0.0, causal=Tru
             scores, pad_mask, scale, p if training else 0.0, causal
         )
     if scores.is_cuda and training and scores.requires_grad:
-        raise RuntimeError(
-            f"fused_scale_mask_softmax HIP path requires SK % 8 == 0 and SK <= 8192 "
-            f"for training, got SK={scores.shape[-1]}"
+        # non-conforming score widths (e.g. ViT's 8x8 patches + cls = 65
+        # tokens) run the composed torch path; this is a SHAPE fallback, not
+        # an extension fallback — log it once so it cannot hide silently.
+        from ..utils.logger import log_first_n
+        import logging
+
+        log_first_n(
+            logging.WARNING,
+            f"fused softmax: SK={scores.shape[-1]} is not a multiple of 8 "
+            f"(or > 8192); using the composed torch path for this shape",
+            n=1,
         )
-    # odd-width decode steps (KV-cache generation) use the composed path
+    # odd widths (ViT token counts, KV-cache decode steps) -> composed path
     return _ref(scores, pad_mask, scale, p, causal, training)
