@@ -346,7 +346,7 @@ __global__ void rowdot_kernel(const bf16_t* __restrict__ dout,
 // backward over kv tiles: dK, dV   (block = 128 kv rows, wave owns 32)
 // ===========================================================================
 template <int D>
-__global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
+__global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
@@ -390,22 +390,15 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
 
   // per-lane K and V row fragments (B-operands: lane j = kv).  Persistent in
   // registers for D=64; re-read from global (L2) per use for D=128.
+  // K/V row fragments are re-read from global per use (L2-resident: one K row
+  // per lane, reused across every q tile).  Keeping them persistent cost 32
+  // VGPRs and held the kernel at 2 waves/SIMD.
   const int kvr_ld = min(kvg, Sk - 1);
-  bf16x8_t kf[D == 64 ? KC : 1], vf[D == 64 ? KC : 1];
-  if constexpr (D == 64) {
-#pragma unroll
-    for (int c = 0; c < KC; ++c) {
-      kf[c] = *(const bf16x8_t*)(kp + (int64_t)kvr_ld * k_ss + c * 16 + hi * 8);
-      vf[c] = *(const bf16x8_t*)(vp + (int64_t)kvr_ld * v_ss + c * 16 + hi * 8);
-    }
-  }
   auto get_kf = [&](int c) {
-    if constexpr (D == 64) return kf[c];
-    else return *(const bf16x8_t*)(kp + (int64_t)kvr_ld * k_ss + c * 16 + hi * 8);
+    return *(const bf16x8_t*)(kp + (int64_t)kvr_ld * k_ss + c * 16 + hi * 8);
   };
   auto get_vf = [&](int c) {
-    if constexpr (D == 64) return vf[c];
-    else return *(const bf16x8_t*)(vp + (int64_t)kvr_ld * v_ss + c * 16 + hi * 8);
+    return *(const bf16x8_t*)(vp + (int64_t)kvr_ld * v_ss + c * 16 + hi * 8);
   };
 
   f32x16_t dk_acc[DT], dv_acc[DT];
@@ -450,28 +443,30 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_kv_kernel(
         dpd = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, get_vf(c), dpd, 0, 0, 0);
       }
 
-      // per-reg: q = q0s + pattern(r); kv = lane's kvg
-      float pd[16], ds[16];
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int qrow = q0s + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        const bool valid = qrow < Sq && kvg < Sk && (!causal || qrow >= kvg);
-        float p = 0.f;
-        if (valid) p = __expf(s[r] * scale - lse_lds[qrow - q0]);
-        float keep = (p_drop > 0.f && valid)
-                         ? drop_keep(seed, bh, Sq, Sk, qrow, kvg,
-                                     drop_threshold_u32(p_drop), ks)
-                         : (valid ? 1.f : 0.f);
-        pd[r] = p * keep;
-        ds[r] = scale * p * (dpd[r] * keep - drow_lds[qrow - q0]);
-        if (!valid) ds[r] = 0.f;
-      }
-
-      // dV^T[d][kv] += dO^T x Pd ;  dK^T[d][kv] += Q^T x dS
+      // per-chunk: compute Pd/dS for 8 regs, repack, feed the MFMAs — the
+      // short lifetimes keep the kernel at 3 waves/SIMD (full pd[16]/ds[16]
+      // arrays held it at 2).
 #pragma unroll
       for (int c16 = 0; c16 < 2; ++c16) {
-        bf16x8_t pdf = repack_chunk(&pd[c16 * 8]);
-        bf16x8_t dsf = repack_chunk(&ds[c16 * 8]);
+        float pd8[8], ds8[8];
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+          const int r = c16 * 8 + k;
+          const int qrow = q0s + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          const bool valid = qrow < Sq && kvg < Sk && (!causal || qrow >= kvg);
+          float p = 0.f;
+          if (valid) p = __expf(s[r] * scale - lse_lds[qrow - q0]);
+          float keep = (p_drop > 0.f && valid)
+                           ? drop_keep(seed, bh, Sq, Sk, qrow, kvg,
+                                       drop_threshold_u32(p_drop), ks)
+                           : (valid ? 1.f : 0.f);
+          pd8[k] = p * keep;
+          ds8[k] = valid
+                       ? scale * p * (dpd[r] * keep - drow_lds[qrow - q0])
+                       : 0.f;
+        }
+        bf16x8_t pdf = repack_chunk(pd8);
+        bf16x8_t dsf = repack_chunk(ds8);
         const int qc = ro + c16 * 16;
 #pragma unroll
         for (int dt = 0; dt < DT; ++dt) {
