@@ -144,12 +144,32 @@ __device__ __forceinline__ void stage_tile(bf16_t* row_lds, bf16_t* tr_lds,
   }
 }
 
-// dropout keep factor for score element (bh, q, kv); thr = p * 2^32
+// Attention dropout: ONE 32-bit hash keyed on (bh, q, kv>>2) yields four
+// byte-granular keep decisions (kv&3 selects the byte; p quantized to 1/256,
+// well below dropout's statistical resolution).  The forward and dQ backward
+// iterate kv in aligned 4-runs at fixed q, so they amortize the hash 4x; the
+// dK/dV backward (q-major layout) pays one hash per element.
+__device__ __forceinline__ uint32_t drop_hash4(uint64_t seed, uint64_t bh,
+                                               int64_t Sq, int64_t Sk4, int q,
+                                               int kv4) {
+  return rnd_hash(seed, (bh * Sq + q) * Sk4 + kv4);
+}
+
+__device__ __forceinline__ float drop_keep_byte(uint32_t h4, int kv_lo,
+                                                uint32_t thr8, float ks) {
+  return (((h4 >> (kv_lo * 8)) & 0xFFu) >= thr8) ? ks : 0.f;
+}
+
+__device__ __forceinline__ uint32_t drop_threshold_u8(float p) {
+  float t = p * 256.0f + 0.5f;
+  return t >= 255.f ? 255u : (uint32_t)t;
+}
+
 __device__ __forceinline__ float drop_keep(uint64_t seed, uint64_t bh, int64_t Sq,
-                                           int64_t Sk, int q, int kv, uint32_t thr,
+                                           int64_t Sk, int q, int kv, uint32_t thr8,
                                            float ks) {
-  uint64_t idx = (bh * Sq + q) * Sk + kv;
-  return (rnd_hash(seed, idx) > thr) ? ks : 0.f;
+  uint32_t h = drop_hash4(seed, bh, Sq, Sk >> 2, q, kv >> 2);
+  return drop_keep_byte(h, kv & 3, thr8, ks);
 }
 
 // ===========================================================================
@@ -265,13 +285,17 @@ __global__ __launch_bounds__(256, 2) void flash_fwd_kernel(
       for (int r = 0; r < 16; ++r) oacc[t][r] *= alpha;
 
     if (p_drop > 0.f) {
-      const uint32_t thr = drop_threshold_u32(p_drop);
+      const uint32_t thr8 = drop_threshold_u8(p_drop);
 #pragma unroll
       for (int t = 0; t < 2; ++t)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          int kv = kv0 + t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          sv[t][r] *= drop_keep(seed, bh, Sq, Sk, qg, kv, thr, ks);
+        for (int r4 = 0; r4 < 4; ++r4) {
+          // regs r4*4..r4*4+3 cover kv run kvb..kvb+3 (4-aligned)
+          const int kvb = kv0 + t * 32 + 8 * r4 + 4 * hi;
+          const uint32_t h = drop_hash4(seed, bh, Sq, Sk >> 2, qg, kvb >> 2);
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            sv[t][r4 * 4 + j] *= drop_keep_byte(h, j, thr8, ks);
         }
     }
 
@@ -458,7 +482,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
           if (valid) p = __expf(s[r] * scale - lse_lds[qrow - q0]);
           float keep = (p_drop > 0.f && valid)
                            ? drop_keep(seed, bh, Sq, Sk, qrow, kvg,
-                                       drop_threshold_u32(p_drop), ks)
+                                       drop_threshold_u8(p_drop), ks)
                            : (valid ? 1.f : 0.f);
           pd8[k] = p * keep;
           ds8[k] = valid
@@ -597,16 +621,23 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
       }
 
       float ds[16];
+      const uint32_t thr8 = drop_threshold_u8(p_drop);
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int kv = kv0s + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        const bool valid = kv < Sk && qg < Sq && (!causal || kv <= qg);
-        float p = valid ? __expf(st[r] * scale - lse_lane) : 0.f;
-        float keep = (p_drop > 0.f && valid)
-                         ? drop_keep(seed, bh, Sq, Sk, qg, kv,
-                                     drop_threshold_u32(p_drop), ks)
-                         : (valid ? 1.f : 0.f);
-        ds[r] = valid ? scale * p * (dpdt[r] * keep - drow_lane) : 0.f;
+      for (int r4 = 0; r4 < 4; ++r4) {
+        const int kvb = kv0s + 8 * r4 + 4 * hi;  // 4-aligned kv run
+        const uint32_t h = (p_drop > 0.f)
+                               ? drop_hash4(seed, bh, Sq, Sk >> 2, qg, kvb >> 2)
+                               : 0;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int r = r4 * 4 + j;
+          const int kv = kvb + j;
+          const bool valid = kv < Sk && qg < Sq && (!causal || kv <= qg);
+          float p = valid ? __expf(st[r] * scale - lse_lane) : 0.f;
+          float keep = (p_drop > 0.f && valid) ? drop_keep_byte(h, j, thr8, ks)
+                                               : (valid ? 1.f : 0.f);
+          ds[r] = valid ? scale * p * (dpdt[r] * keep - drow_lane) : 0.f;
+        }
       }
 
       // dQ^T[d][q] += K^T x dS^T
@@ -644,10 +675,13 @@ __global__ void attn_dropout_apply_kernel(typename E::T* __restrict__ x, int64_t
                                           int64_t Sq, int64_t Sk, float p,
                                           float keep_scale, uint64_t seed) {
   const int64_t total = BH * Sq * Sk;
-  const uint32_t thr = drop_threshold_u32(p);
+  const uint32_t thr8 = drop_threshold_u8(p);
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < total;
        i += (int64_t)gridDim.x * blockDim.x) {
-    float kp = (rnd_hash(seed, (uint64_t)i) > thr) ? keep_scale : 0.f;
+    const int64_t row = i / Sk;
+    const int kv = (int)(i % Sk);
+    uint32_t h = rnd_hash(seed, row * (Sk >> 2) + (kv >> 2));
+    float kp = drop_keep_byte(h, kv & 3, thr8, keep_scale);
     x[i] = E::from_f(E::to_f(x[i]) * kp);
   }
 }
