@@ -14,7 +14,6 @@ from torch.utils.checkpoint import checkpoint as act_checkpoint
 from ..config import configurable
 from ..layers import Linear1D, LMLogits, ParallelCrossEntropyLoss, RMSLayerNorm, VocabEmbedding
 from ..ops.attention import flash_attention, flash_attention_available
-from ..ops.fused_bias import bias_dropout_add
 from ..ops.rope import apply_rotary_pos_emb
 from ..ops.softmax import fused_scale_mask_softmax
 from ..ops.swiglu import swiglu
