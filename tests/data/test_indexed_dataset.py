@@ -133,3 +133,19 @@ def test_t5_dataset_span_corruption(tmp_path):
     # sentinels (counting down from vocab-1) appear in encoder and decoder
     assert (enc >= 5900).any()
     assert (dec_in >= 5900).any()
+
+
+def test_roberta_dataset_no_nsp(tmp_path):
+    from libai_amd.data.datasets import RobertaDataset
+
+    ds = _sentence_corpus(tmp_path)
+    rd = RobertaDataset("t", ds, max_seq_length=64, vocab_size=6000,
+                        num_samples=10)
+    assert len(rd) > 0
+    inst = rd[0]
+    assert "ns_labels" not in inst.get_fields()
+    assert (inst.get("tokentype_ids").tensor == 0).all()
+    loss_mask = inst.get("loss_mask").tensor
+    labels = inst.get("lm_labels").tensor
+    assert (labels[loss_mask == 0] == -1).all()
+    assert (labels[loss_mask == 1] >= 0).all()
