@@ -148,15 +148,20 @@ __device__ __forceinline__ float u32_to_uniform(uint32_t x) {
 // 32-BIT ops only: 64-bit integer multiplies are emulated on gfx950 and a
 // splitmix64 here measured ~25 VALU/element (PMC: 104 issued instructions per
 // MFMA in flash_fwd).  This lowCbias mixer is ~9 VALU.
-__device__ __forceinline__ uint32_t rnd_hash(uint64_t seed, uint64_t idx) {
-  uint32_t h = (uint32_t)seed ^ ((uint32_t)idx * 0x9E3779B9u) ^
-               ((uint32_t)(idx >> 32) * 0x85EBCA6Bu) ^ (uint32_t)(seed >> 32);
+__device__ __forceinline__ uint32_t rnd_hash2(uint64_t seed, uint32_t lo,
+                                               uint32_t hi) {
+  uint32_t h = (uint32_t)seed ^ (lo * 0x9E3779B9u) ^ (hi * 0x85EBCA6Bu) ^
+               (uint32_t)(seed >> 32);
   h ^= h >> 16;
   h *= 0x7FEB352Du;
   h ^= h >> 15;
   h *= 0x846CA68Bu;
   h ^= h >> 16;
   return h;
+}
+
+__device__ __forceinline__ uint32_t rnd_hash(uint64_t seed, uint64_t idx) {
+  return rnd_hash2(seed, (uint32_t)idx, (uint32_t)(idx >> 32));
 }
 
 // keep-decision without the float conversion: compare the hash against a

@@ -149,10 +149,15 @@ __device__ __forceinline__ void stage_tile(bf16_t* row_lds, bf16_t* tr_lds,
 // well below dropout's statistical resolution).  The forward and dQ backward
 // iterate kv in aligned 4-runs at fixed q, so they amortize the hash 4x; the
 // dK/dV backward (q-major layout) pays one hash per element.
+// 32-bit keying: (q*Sk4 + kv4) is collision-free below 2^32 (seq up to ~128k)
+// and avoids gfx950's emulated 64-bit multiplies in the per-element bwd path;
+// the head index feeds the mixer's second word.
 __device__ __forceinline__ uint32_t drop_hash4(uint64_t seed, uint64_t bh,
                                                int64_t Sq, int64_t Sk4, int q,
                                                int kv4) {
-  return rnd_hash(seed, (bh * Sq + q) * Sk4 + kv4);
+  (void)Sq;
+  return rnd_hash2(seed, (uint32_t)q * (uint32_t)Sk4 + (uint32_t)kv4,
+                   (uint32_t)bh);
 }
 
 __device__ __forceinline__ float drop_keep_byte(uint32_t h4, int kv_lo,
@@ -479,21 +484,42 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
 #pragma unroll
       for (int c16 = 0; c16 < 2; ++c16) {
         float pd8[8], ds8[8];
+        // dropout hashes, quad-distributed: each element needs the draw for
+        // (its qrow, lane's kv4); the 4 lanes of a quad share kv4 and the 4
+        // qrows of a run are consecutive, so lane (l&3) computes the run's
+        // (qbase + l&3) hash and DPP quad-perm broadcasts each one back.
+        // (A straight per-element hash makes the compiler hoist 16 hash
+        // bases per subtile -> 25+ VGPR spills into the hot loop.)
+        const uint32_t thr8 = drop_threshold_u8(p_drop);
+        const int kv4q = (kv_base + (l31 & ~3)) >> 2;
+        const int byte_sh = (l31 & 3) * 8;
 #pragma unroll
-        for (int k = 0; k < 8; ++k) {
-          const int r = c16 * 8 + k;
-          const int qrow = q0s + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          const bool valid = qrow < Sq && kvg < Sk && (!causal || qrow >= kvg);
-          float p = 0.f;
-          if (valid) p = __expf(s[r] * scale - lse_lds[qrow - q0]);
-          float keep = (p_drop > 0.f && valid)
-                           ? drop_keep(seed, bh, Sq, Sk, qrow, kvg,
-                                       drop_threshold_u8(p_drop), ks)
-                           : (valid ? 1.f : 0.f);
-          pd8[k] = p * keep;
-          ds8[k] = valid
-                       ? scale * p * (dpd[r] * keep - drow_lds[qrow - q0])
-                       : 0.f;
+        for (int r4 = 0; r4 < 2; ++r4) {
+          const int qbase = q0s + c16 * 16 + 8 * r4 + 4 * hi;
+          uint32_t hq = 0;
+          if (p_drop > 0.f)
+            hq = drop_hash4(seed, bh, Sq, Sk >> 2, qbase + (l31 & 3), kv4q);
+          uint32_t hqj[4];
+          hqj[0] = (uint32_t)__builtin_amdgcn_mov_dpp((int)hq, 0x00, 0xF, 0xF, true);
+          hqj[1] = (uint32_t)__builtin_amdgcn_mov_dpp((int)hq, 0x55, 0xF, 0xF, true);
+          hqj[2] = (uint32_t)__builtin_amdgcn_mov_dpp((int)hq, 0xAA, 0xF, 0xF, true);
+          hqj[3] = (uint32_t)__builtin_amdgcn_mov_dpp((int)hq, 0xFF, 0xF, 0xF, true);
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            const int k = r4 * 4 + j;
+            const int r = c16 * 8 + k;
+            const int qrow = qbase + j;
+            const bool valid = qrow < Sq && kvg < Sk && (!causal || qrow >= kvg);
+            float p = 0.f;
+            if (valid) p = __expf(s[r] * scale - lse_lds[qrow - q0]);
+            float keep = valid ? 1.f : 0.f;
+            if (p_drop > 0.f && valid)
+              keep = (((hqj[j] >> byte_sh) & 0xFFu) >= thr8) ? ks : 0.f;
+            pd8[k] = p * keep;
+            ds8[k] = valid
+                         ? scale * p * (dpd[r] * keep - drow_lds[qrow - q0])
+                         : 0.f;
+          }
         }
         bf16x8_t pdf = repack_chunk(pd8);
         bf16x8_t dsf = repack_chunk(ds8);
@@ -686,7 +712,9 @@ __global__ void attn_dropout_apply_kernel(typename E::T* __restrict__ x, int64_t
        i += (int64_t)gridDim.x * blockDim.x) {
     const int64_t row = i / Sk;
     const int kv = (int)(i % Sk);
-    uint32_t h = rnd_hash(seed, row * (Sk >> 2) + (kv >> 2));
+    uint32_t h = rnd_hash2(seed, (uint32_t)(row % Sq) * (uint32_t)(Sk >> 2) +
+                                     (uint32_t)(kv >> 2),
+                           (uint32_t)(row / Sq));
     float kp = drop_keep_byte(h, kv & 3, thr8, keep_scale);
     x[i] = E::from_f(E::to_f(x[i]) * kp);
   }
