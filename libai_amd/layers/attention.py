@@ -112,7 +112,10 @@ class MultiheadAttention(nn.Module):
             and past_key_value is None
             and not use_cache
         ):
-            from ..ops.attention import flash_attention, flash_attention_available
+            from ..ops.attention import (
+                flash_attention_available,
+                flash_attention_qkv,
+            )
 
             b, s, _ = hidden_states.shape
             if flash_attention_available(
@@ -121,8 +124,8 @@ class MultiheadAttention(nn.Module):
             ):
                 qkv = self.query_key_value(hidden_states)
                 qkv5 = qkv.view(b, s, self.num_heads_local, 3, self.head_size)
-                o = flash_attention(
-                    qkv5[..., 0, :], qkv5[..., 1, :], qkv5[..., 2, :],
+                o = flash_attention_qkv(
+                    qkv5,
                     scale=self.norm_factor * (self.coeff if self.coeff else 1.0),
                     p_drop=self.attention_dropout_prob,
                     causal=self.attn_mask_type == AttnMaskType.causal,

@@ -138,3 +138,22 @@ def test_attention_layer_uses_flash_and_matches_unfused():
         out_unfused.float().abs().max() + 1e-6
     )
     assert rel < 3e-2, f"flash vs unfused layer mismatch {rel}"
+
+
+def test_flash_qkv_packed_matches_view_path():
+    """Packed-qkv autograd path == separate-views path (incl. packed dqkv)."""
+    from libai_amd.ops.attention import flash_attention, flash_attention_qkv
+
+    torch.manual_seed(7)
+    b, s, h, d = 2, 256, 4, 64
+    qkv = torch.randn(b, s, h, 3, d, device="cuda", dtype=torch.bfloat16)
+    q1 = qkv.clone().requires_grad_(True)
+    q2 = qkv.clone().requires_grad_(True)
+    o1 = flash_attention_qkv(q1, scale=d ** -0.5, p_drop=0.0, causal=True)
+    o2 = flash_attention(q2[..., 0, :], q2[..., 1, :], q2[..., 2, :],
+                         scale=d ** -0.5, p_drop=0.0, causal=True)
+    assert torch.equal(o1, o2)
+    g = torch.randn_like(o1)
+    o1.backward(g)
+    o2.backward(g)
+    assert torch.equal(q1.grad, q2.grad)
