@@ -164,6 +164,13 @@ __device__ __forceinline__ uint32_t rnd_hash(uint64_t seed, uint64_t idx) {
   return rnd_hash2(seed, (uint32_t)idx, (uint32_t)(idx >> 32));
 }
 
+// byte-granular keep threshold: p quantized to 1/256 (one hash draw yields
+// 4 keep decisions via its bytes)
+__device__ __forceinline__ uint32_t drop_threshold_u8(float p) {
+  float t = p * 256.0f + 0.5f;
+  return t >= 255.f ? 255u : (uint32_t)t;
+}
+
 // keep-decision without the float conversion: compare the hash against a
 // precomputed uint32 threshold = p * 2^32.
 __device__ __forceinline__ uint32_t drop_threshold_u32(float p) {

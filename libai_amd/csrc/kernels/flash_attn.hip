@@ -165,11 +165,6 @@ __device__ __forceinline__ float drop_keep_byte(uint32_t h4, int kv_lo,
   return (((h4 >> (kv_lo * 8)) & 0xFFu) >= thr8) ? ks : 0.f;
 }
 
-__device__ __forceinline__ uint32_t drop_threshold_u8(float p) {
-  float t = p * 256.0f + 0.5f;
-  return t >= 255.f ? 255u : (uint32_t)t;
-}
-
 __device__ __forceinline__ float drop_keep(uint64_t seed, uint64_t bh, int64_t Sq,
                                            int64_t Sk, int q, int kv, uint32_t thr8,
                                            float ks) {

@@ -11,6 +11,9 @@ namespace {
 // ---------------------------------------------------------------------------
 // y = gelu(x + b)   (erf gelu, matches torch.nn.functional.gelu default)
 // ---------------------------------------------------------------------------
+// 2D launch: thread owns ONE vector-column (bias loaded once, no per-element
+// modulo -- a flat grid-stride variant spent ~30 of its 406 loop instructions
+// on the emulated 64-bit `i % wvec`), rows strided by gridDim.y.
 template <class E, bool GRAD>
 __global__ void bias_gelu_kernel(const typename E::T* __restrict__ x,
                                  const typename E::T* __restrict__ b,
@@ -18,13 +21,15 @@ __global__ void bias_gelu_kernel(const typename E::T* __restrict__ x,
                                  typename E::T* __restrict__ out, int64_t n, int W) {
   using VecT = typename E::VecT;
   constexpr int V = E::VEC;
-  const int64_t nvec = n / V;
   const int wvec = W / V;
-  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nvec;
-       i += (int64_t)gridDim.x * blockDim.x) {
+  const int64_t R = n / W;
+  const int cv = blockIdx.x * blockDim.x + threadIdx.x;
+  if (cv >= wvec) return;
+  VecT vb;
+  if (b) vb = ((const VecT*)b)[cv];
+  for (int64_t r = blockIdx.y; r < R; r += gridDim.y) {
+    const int64_t i = r * wvec + cv;
     VecT vx = ((const VecT*)x)[i];
-    VecT vb;
-    if (b) vb = ((const VecT*)b)[i % wvec];
     VecT o;
     if (GRAD) {
       VecT vdy = ((const VecT*)dy)[i];
@@ -48,6 +53,10 @@ __global__ void bias_gelu_kernel(const typename E::T* __restrict__ x,
 // y = residual + dropout(x + b);  backward dx = dy * mask / (1-p)
 // mask is philox(seed, elem_idx) — identical in fwd and bwd.
 // ---------------------------------------------------------------------------
+// Dropout mask: one rnd_hash draw yields 4 byte-granular keep decisions
+// (p quantized to 1/256 like the attention dropout; a philox4 here cost ~60
+// VALU per 4 elements vs ~10 for the mixer).  fwd and bwd regenerate the
+// identical mask from (seed, element index).
 template <class E, bool GRAD>
 __global__ void bias_dropout_res_kernel(const typename E::T* __restrict__ x,
                                         const typename E::T* __restrict__ b,
@@ -56,23 +65,26 @@ __global__ void bias_dropout_res_kernel(const typename E::T* __restrict__ x,
                                         int W, float p, uint64_t seed) {
   using VecT = typename E::VecT;
   constexpr int V = E::VEC;
-  const int64_t nvec = n / V;
   const int wvec = W / V;
+  const int64_t R = n / W;
   const float scale = 1.0f / (1.0f - p);
-  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nvec;
-       i += (int64_t)gridDim.x * blockDim.x) {
+  const uint32_t thr8 = drop_threshold_u8(p);
+  const int cv = blockIdx.x * blockDim.x + threadIdx.x;
+  if (cv >= wvec) return;
+  VecT vb;
+  if (b) vb = ((const VecT*)b)[cv];
+  for (int64_t r = blockIdx.y; r < R; r += gridDim.y) {
+    const int64_t i = r * wvec + cv;
     VecT vx = ((const VecT*)x)[i];
-    VecT vb;
-    if (b) vb = ((const VecT*)b)[i % wvec];
     float keep[V];
     if (p > 0.f) {
-      uint32_t r[4];
 #pragma unroll
       for (int q = 0; q < V / 4; ++q) {
-        philox4(seed, (uint64_t)i * (V / 4) + q, r);
+        const uint64_t idx = (uint64_t)i * (V / 4) + q;
+        const uint32_t h = rnd_hash(seed, idx);
 #pragma unroll
         for (int j = 0; j < 4; ++j)
-          keep[q * 4 + j] = (u32_to_uniform(r[j]) > p) ? scale : 0.f;
+          keep[q * 4 + j] = (((h >> (j * 8)) & 0xFFu) >= thr8) ? scale : 0.f;
       }
     } else {
 #pragma unroll
@@ -138,37 +150,55 @@ inline int64_t ew_grid(int64_t nvec) {
   return g < 2048 ? g : 2048;  // grid-stride past 2048 blocks (guide G11)
 }
 
+// 2D launch shape for the fixed-column kernels: x covers the vector-columns,
+// y strides rows, total blocks ~4096 (>> 256 CUs, bounded queue cost)
+inline dim3 col_grid(int64_t n, int W, int V, int* block_out) {
+  const int wvec = W / V;
+  const int block = wvec < 256 ? wvec : 256;
+  const int gx = (int)CDIV(wvec, block);
+  int64_t rows = n / W;
+  int64_t gy = 4096 / gx;
+  if (gy > rows) gy = rows;
+  if (gy < 1) gy = 1;
+  *block_out = block;
+  return dim3(gx, (uint32_t)gy);
+}
+
 }  // namespace
 
 #define BIAS_LAUNCHERS(SUFF, ETYPE)                                                      \
   extern "C" void bias_gelu_fwd_##SUFF(const void* x, const void* b, void* y, int64_t n, \
                                        int W, hipStream_t stream) {                      \
-    bias_gelu_kernel<ETYPE, false>                                                       \
-        <<<dim3(ew_grid(n / ETYPE::VEC)), dim3(256), 0, stream>>>(                       \
-            (const ETYPE::T*)x, (const ETYPE::T*)b, nullptr, (ETYPE::T*)y, n, W);        \
+    int blk;                                                                             \
+    dim3 g = col_grid(n, W, ETYPE::VEC, &blk);                                           \
+    bias_gelu_kernel<ETYPE, false><<<g, dim3(blk), 0, stream>>>(                         \
+        (const ETYPE::T*)x, (const ETYPE::T*)b, nullptr, (ETYPE::T*)y, n, W);            \
   }                                                                                      \
   extern "C" void bias_gelu_bwd_##SUFF(const void* x, const void* b, const void* dy,     \
                                        void* dx, int64_t n, int W, hipStream_t stream) { \
-    bias_gelu_kernel<ETYPE, true>                                                        \
-        <<<dim3(ew_grid(n / ETYPE::VEC)), dim3(256), 0, stream>>>(                       \
-            (const ETYPE::T*)x, (const ETYPE::T*)b, (const ETYPE::T*)dy, (ETYPE::T*)dx,  \
-            n, W);                                                                       \
+    int blk;                                                                             \
+    dim3 g = col_grid(n, W, ETYPE::VEC, &blk);                                           \
+    bias_gelu_kernel<ETYPE, true><<<g, dim3(blk), 0, stream>>>(                          \
+        (const ETYPE::T*)x, (const ETYPE::T*)b, (const ETYPE::T*)dy, (ETYPE::T*)dx, n,   \
+        W);                                                                              \
   }                                                                                      \
   extern "C" void bias_dropout_res_fwd_##SUFF(const void* x, const void* b,              \
                                               const void* res, void* y, int64_t n,       \
                                               int W, float p, uint64_t seed,             \
                                               hipStream_t stream) {                      \
-    bias_dropout_res_kernel<ETYPE, false>                                                \
-        <<<dim3(ew_grid(n / ETYPE::VEC)), dim3(256), 0, stream>>>(                       \
-            (const ETYPE::T*)x, (const ETYPE::T*)b, (const ETYPE::T*)res, (ETYPE::T*)y,  \
-            n, W, p, seed);                                                              \
+    int blk;                                                                             \
+    dim3 g = col_grid(n, W, ETYPE::VEC, &blk);                                           \
+    bias_dropout_res_kernel<ETYPE, false><<<g, dim3(blk), 0, stream>>>(                  \
+        (const ETYPE::T*)x, (const ETYPE::T*)b, (const ETYPE::T*)res, (ETYPE::T*)y, n,   \
+        W, p, seed);                                                                     \
   }                                                                                      \
   extern "C" void bias_dropout_res_bwd_##SUFF(const void* dy, void* dx, int64_t n,       \
                                               int W, float p, uint64_t seed,             \
                                               hipStream_t stream) {                      \
-    bias_dropout_res_kernel<ETYPE, true>                                                 \
-        <<<dim3(ew_grid(n / ETYPE::VEC)), dim3(256), 0, stream>>>(                       \
-            (const ETYPE::T*)dy, nullptr, nullptr, (ETYPE::T*)dx, n, W, p, seed);        \
+    int blk;                                                                             \
+    dim3 g = col_grid(n, W, ETYPE::VEC, &blk);                                           \
+    bias_dropout_res_kernel<ETYPE, true><<<g, dim3(blk), 0, stream>>>(                   \
+        (const ETYPE::T*)dy, nullptr, nullptr, (ETYPE::T*)dx, n, W, p, seed);            \
   }                                                                                      \
   extern "C" void colsum_##SUFF(const void* in, float* partial, void* out, int64_t R,    \
                                 int W, int P, hipStream_t stream) {                      \
