@@ -157,7 +157,9 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ln_bwd(
   auto dx = torch::empty_like(x);
   auto dgamma = torch::empty_like(gamma);
   auto dbeta = need_dbeta ? torch::empty_like(gamma) : torch::Tensor();
-  const int P = (int)std::min<int64_t>(256, std::max<int64_t>(1, R / 4));
+  const int64_t gxw = std::max<int64_t>(1, (H / 8 + 255) / 256);
+  const int P = (int)std::min<int64_t>(std::max<int64_t>(1, 2048 / gxw),
+                                       std::max<int64_t>(1, R / 4));
   auto partial = torch::empty({(need_dbeta ? 2L : 1L) * P, (int64_t)H},
                               x.options().dtype(torch::kFloat32));
   float* pgamma = partial.data_ptr<float>();
@@ -171,6 +173,10 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ln_bwd(
      dgamma.data_ptr(), need_dbeta ? dbeta.data_ptr() : nullptr, R, H, P, rms,
      cur_stream());
   check_launch("ln_bwd");
+  // fold the [P, H] fp32 partials with torch's reduction (the hand-rolled
+  // fold was grid-starved at large P)
+  dgamma.copy_(partial.narrow(0, 0, P).sum(0));
+  if (need_dbeta) dbeta.copy_(partial.narrow(0, P, P).sum(0));
   return {dx, dgamma, dbeta};
 }
 
@@ -204,13 +210,16 @@ torch::Tensor bias_gelu_bwd(torch::Tensor x, c10::optional<torch::Tensor> b,
 torch::Tensor colsum(torch::Tensor x, int64_t W) {
   CHECK_IN(x);
   const int64_t R = x.numel() / W;
-  const int P = (int)std::min<int64_t>(256, std::max<int64_t>(1, R / 4));
+  const int64_t gxw = std::max<int64_t>(1, (W / 8 + 255) / 256);
+  const int P = (int)std::min<int64_t>(std::max<int64_t>(1, 2048 / gxw),
+                                       std::max<int64_t>(1, R / 4));
   auto partial = torch::empty({P, W}, x.options().dtype(torch::kFloat32));
   auto out = torch::empty({W}, x.options());
   auto fn = is_bf16(x) ? colsum_bf16 : colsum_f32;
   fn(x.data_ptr(), partial.data_ptr<float>(), out.data_ptr(), R, (int)W, P,
      cur_stream());
   check_launch("colsum");
+  out.copy_(partial.sum(0));
   return out;
 }
 

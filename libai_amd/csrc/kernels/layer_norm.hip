@@ -76,6 +76,141 @@ __global__ void norm_fwd_kernel(const typename E::T* __restrict__ x,
   }
 }
 
+// Wave-per-row variants (H <= 2048, H % VEC == 0): one 64-lane wave owns a
+// row, reduces with shfl_xor (no LDS, no barriers), and keeps the row's x
+// vectors in registers between the stats pass and the normalize pass.  The
+// block-per-row kernels above remain the fallback for wide/ragged rows.
+template <class E, bool RMS, int NV>
+__global__ void norm_fwd_wave_kernel(const typename E::T* __restrict__ x,
+                                     const typename E::T* __restrict__ gamma,
+                                     const typename E::T* __restrict__ beta,
+                                     typename E::T* __restrict__ y,
+                                     float* __restrict__ mean_out,
+                                     float* __restrict__ rstd_out, int H, float eps,
+                                     int64_t R) {
+  using T = typename E::T;
+  using VecT = typename E::VecT;
+  constexpr int V = E::VEC;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int64_t row = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave;
+  if (row >= R) return;
+  const T* xr = x + row * (int64_t)H;
+  T* yr = y + row * (int64_t)H;
+  const int nvec = H / V;
+
+  VecT vx[NV];
+  float sum = 0.f, sumsq = 0.f;
+#pragma unroll
+  for (int n = 0; n < NV; ++n) {
+    const int i = lane + n * 64;
+    if (i < nvec) {
+      vx[n] = ((const VecT*)xr)[i];
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float f = E::to_f(vx[n][j]);
+        sum += f;
+        sumsq += f * f;
+      }
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    sum += __shfl_xor(sum, off, 64);
+    sumsq += __shfl_xor(sumsq, off, 64);
+  }
+  const float mean = RMS ? 0.f : sum / H;
+  const float rstd = rsqrtf(sumsq / H - mean * mean + eps);
+  if (lane == 0) {
+    if (!RMS && mean_out) mean_out[row] = mean;
+    if (rstd_out) rstd_out[row] = rstd;
+  }
+#pragma unroll
+  for (int n = 0; n < NV; ++n) {
+    const int i = lane + n * 64;
+    if (i < nvec) {
+      VecT g = ((const VecT*)gamma)[i];
+      VecT o;
+      if (!RMS && beta != nullptr) {
+        VecT b = ((const VecT*)beta)[i];
+#pragma unroll
+        for (int j = 0; j < V; ++j)
+          o[j] = E::from_f((E::to_f(vx[n][j]) - mean) * rstd * E::to_f(g[j]) +
+                           E::to_f(b[j]));
+      } else {
+#pragma unroll
+        for (int j = 0; j < V; ++j)
+          o[j] = E::from_f((E::to_f(vx[n][j]) - mean) * rstd * E::to_f(g[j]));
+      }
+      ((VecT*)yr)[i] = o;
+    }
+  }
+}
+
+template <class E, bool RMS, int NV>
+__global__ void norm_bwd_dx_wave_kernel(const typename E::T* __restrict__ dy,
+                                        const typename E::T* __restrict__ x,
+                                        const typename E::T* __restrict__ gamma,
+                                        const float* __restrict__ mean_in,
+                                        const float* __restrict__ rstd_in,
+                                        typename E::T* __restrict__ dx, int H,
+                                        int64_t R) {
+  using T = typename E::T;
+  using VecT = typename E::VecT;
+  constexpr int V = E::VEC;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int64_t row = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave;
+  if (row >= R) return;
+  const T* dyr = dy + row * (int64_t)H;
+  const T* xr = x + row * (int64_t)H;
+  T* dxr = dx + row * (int64_t)H;
+  const int nvec = H / V;
+  const float mean = RMS ? 0.f : mean_in[row];
+  const float rstd = rstd_in[row];
+
+  VecT vdy[NV], vx[NV], vg[NV];
+  float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+  for (int n = 0; n < NV; ++n) {
+    const int i = lane + n * 64;
+    if (i < nvec) {
+      vdy[n] = ((const VecT*)dyr)[i];
+      vx[n] = ((const VecT*)xr)[i];
+      vg[n] = ((const VecT*)gamma)[i];
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float dyg = E::to_f(vdy[n][j]) * E::to_f(vg[n][j]);
+        const float xhat = (E::to_f(vx[n][j]) - mean) * rstd;
+        s1 += dyg;
+        s2 += dyg * xhat;
+      }
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s1 += __shfl_xor(s1, off, 64);
+    s2 += __shfl_xor(s2, off, 64);
+  }
+  s1 /= H;
+  s2 /= H;
+#pragma unroll
+  for (int n = 0; n < NV; ++n) {
+    const int i = lane + n * 64;
+    if (i < nvec) {
+      VecT o;
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float dyg = E::to_f(vdy[n][j]) * E::to_f(vg[n][j]);
+        const float xhat = (E::to_f(vx[n][j]) - mean) * rstd;
+        o[j] = E::from_f(RMS ? rstd * (dyg - xhat * s2)
+                             : rstd * (dyg - s1 - xhat * s2));
+      }
+      ((VecT*)dxr)[i] = o;
+    }
+  }
+}
+
 // dx for LayerNorm:  dx = rstd * (dyg - mean(dyg) - xhat * mean(dyg*xhat))
 // dx for RMSNorm:    dx = rstd * dyg - x * rstd^3 / H * sum(dyg * x)
 template <class E, bool RMS>
@@ -146,6 +281,29 @@ __global__ void norm_bwd_dx_kernel(const typename E::T* __restrict__ dy,
 
 // Stage 1 of dgamma/dbeta: each thread owns one column, strides row-groups.
 // grid = (CDIV(H, 256), P); partials layout [P][H] fp32 (dgamma) + [P][H] (dbeta).
+// scalar fallback for H not divisible by the vector width
+template <class E, bool RMS>
+__global__ void norm_bwd_wgrad_partial_scalar_kernel(
+    const typename E::T* __restrict__ dy, const typename E::T* __restrict__ x,
+    const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
+    float* __restrict__ pgamma, float* __restrict__ pbeta, int64_t R, int H) {
+  const int j = blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= H) return;
+  const int P = gridDim.y;
+  float dg = 0.f, db = 0.f;
+  for (int64_t r = blockIdx.y; r < R; r += P) {
+    const float mean = RMS ? 0.f : mean_in[r];
+    const float rstd = rstd_in[r];
+    const float dyv = E::to_f(dy[r * H + j]);
+    dg += dyv * (E::to_f(x[r * H + j]) - mean) * rstd;
+    db += dyv;
+  }
+  pgamma[(int64_t)blockIdx.y * H + j] = dg;
+  if (pbeta) pbeta[(int64_t)blockIdx.y * H + j] = db;
+}
+
+// vectorized: thread owns E::VEC columns (16B loads of dy and x per row --
+// the scalar 2B-per-thread variant ran at ~2 TB/s)
 template <class E, bool RMS>
 __global__ void norm_bwd_wgrad_partial_kernel(const typename E::T* __restrict__ dy,
                                               const typename E::T* __restrict__ x,
@@ -154,52 +312,81 @@ __global__ void norm_bwd_wgrad_partial_kernel(const typename E::T* __restrict__ 
                                               float* __restrict__ pgamma,
                                               float* __restrict__ pbeta, int64_t R,
                                               int H) {
-  using T = typename E::T;
-  const int j = blockIdx.x * blockDim.x + threadIdx.x;
-  if (j >= H) return;
+  using VecT = typename E::VecT;
+  constexpr int V = E::VEC;
+  const int wv = H / V;
+  const int jv = blockIdx.x * blockDim.x + threadIdx.x;
+  if (jv >= wv) return;
   const int P = gridDim.y;
-  float dg = 0.f, db = 0.f;
+  float dg[V], db[V];
+#pragma unroll
+  for (int j = 0; j < V; ++j) dg[j] = db[j] = 0.f;
   for (int64_t r = blockIdx.y; r < R; r += P) {
-    float mean = RMS ? 0.f : mean_in[r];
-    float rstd = rstd_in[r];
-    float dyv = E::to_f(dy[r * H + j]);
-    float xhat = (E::to_f(x[r * H + j]) - mean) * rstd;
-    dg += dyv * xhat;
-    db += dyv;
+    const float mean = RMS ? 0.f : mean_in[r];
+    const float rstd = rstd_in[r];
+    VecT vd = ((const VecT*)(dy + r * H))[jv];
+    VecT vx = ((const VecT*)(x + r * H))[jv];
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      const float dyv = E::to_f(vd[j]);
+      dg[j] += dyv * (E::to_f(vx[j]) - mean) * rstd;
+      db[j] += dyv;
+    }
   }
-  pgamma[(int64_t)blockIdx.y * H + j] = dg;
-  if (pbeta) pbeta[(int64_t)blockIdx.y * H + j] = db;
+#pragma unroll
+  for (int j = 0; j < V; ++j) {
+    pgamma[(int64_t)blockIdx.y * H + jv * V + j] = dg[j];
+    if (pbeta) pbeta[(int64_t)blockIdx.y * H + jv * V + j] = db[j];
+  }
 }
 
-// fold over P row-group partials: 64 cols x 8 p-lanes per block (TLP over the
-// strided reads; a [H]-thread fold is latency-bound at H~1k).
-template <class E>
-__global__ void wgrad_fold_kernel(const float* __restrict__ pgamma,
-                                  const float* __restrict__ pbeta,
-                                  typename E::T* __restrict__ dgamma,
-                                  typename E::T* __restrict__ dbeta, int P, int H) {
-  __shared__ float lds[2][8][64];
-  const int j = blockIdx.x * 64 + (int)(threadIdx.x % 64);
-  const int pl = threadIdx.x / 64;
-  float dg = 0.f, db = 0.f;
-  if (j < H) {
-    for (int p = pl; p < P; p += 8) {
-      dg += pgamma[(int64_t)p * H + j];
-      if (pbeta) db += pbeta[(int64_t)p * H + j];
-    }
+template <class E, bool RMS>
+void launch_norm_fwd(const void* x, const void* gamma, const void* beta, void* y,
+                     float* mean, float* rstd, int64_t R, int H, float eps,
+                     hipStream_t stream, int th) {
+  const int nvl = CDIV(H / E::VEC, 64);
+  if (H % E::VEC == 0 && nvl <= 4) {
+    dim3 g((uint32_t)CDIV(R, 4));
+#define FWD_WAVE(NV)                                                          \
+  norm_fwd_wave_kernel<E, RMS, NV><<<g, dim3(256), 0, stream>>>(              \
+      (const typename E::T*)x, (const typename E::T*)gamma,                   \
+      (const typename E::T*)beta, (typename E::T*)y, mean, rstd, H, eps, R)
+    if (nvl <= 1)
+      FWD_WAVE(1);
+    else if (nvl == 2)
+      FWD_WAVE(2);
+    else
+      FWD_WAVE(4);
+#undef FWD_WAVE
+  } else {
+    norm_fwd_kernel<E, RMS><<<dim3((uint32_t)R), dim3(th), 0, stream>>>(
+        (const typename E::T*)x, (const typename E::T*)gamma,
+        (const typename E::T*)beta, (typename E::T*)y, mean, rstd, H, eps);
   }
-  lds[0][pl][threadIdx.x % 64] = dg;
-  lds[1][pl][threadIdx.x % 64] = db;
-  __syncthreads();
-  if (pl == 0 && j < H) {
-    float sdg = 0.f, sdb = 0.f;
-#pragma unroll
-    for (int q = 0; q < 8; ++q) {
-      sdg += lds[0][q][threadIdx.x % 64];
-      sdb += lds[1][q][threadIdx.x % 64];
-    }
-    dgamma[j] = E::from_f(sdg);
-    if (dbeta) dbeta[j] = E::from_f(sdb);
+}
+
+template <class E, bool RMS>
+void launch_norm_bwd_dx(const void* dy, const void* x, const void* gamma,
+                        const float* mean, const float* rstd, void* dx, int64_t R,
+                        int H, hipStream_t stream, int th) {
+  const int nvl = CDIV(H / E::VEC, 64);
+  if (H % E::VEC == 0 && nvl <= 4) {
+    dim3 g((uint32_t)CDIV(R, 4));
+#define DX_WAVE(NV)                                                           \
+  norm_bwd_dx_wave_kernel<E, RMS, NV><<<g, dim3(256), 0, stream>>>(           \
+      (const typename E::T*)dy, (const typename E::T*)x,                      \
+      (const typename E::T*)gamma, mean, rstd, (typename E::T*)dx, H, R)
+    if (nvl <= 1)
+      DX_WAVE(1);
+    else if (nvl == 2)
+      DX_WAVE(2);
+    else
+      DX_WAVE(4);
+#undef DX_WAVE
+  } else {
+    norm_bwd_dx_kernel<E, RMS><<<dim3((uint32_t)R), dim3(th), 0, stream>>>(
+        (const typename E::T*)dy, (const typename E::T*)x,
+        (const typename E::T*)gamma, mean, rstd, (typename E::T*)dx, H);
   }
 }
 
@@ -220,40 +407,49 @@ inline int row_block_threads(int H, int vec) {
                                 float eps, bool rms, hipStream_t stream) {              \
     int th = row_block_threads(H, ETYPE::VEC);                                          \
     if (rms)                                                                            \
-      norm_fwd_kernel<ETYPE, true><<<dim3((uint32_t)R), dim3(th), 0, stream>>>(         \
-          (const ETYPE::T*)x, (const ETYPE::T*)gamma, (const ETYPE::T*)beta,            \
-          (ETYPE::T*)y, mean, rstd, H, eps);                                            \
+      launch_norm_fwd<ETYPE, true>(x, gamma, beta, y, mean, rstd, R, H, eps, stream,    \
+                                   th);                                                 \
     else                                                                                \
-      norm_fwd_kernel<ETYPE, false><<<dim3((uint32_t)R), dim3(th), 0, stream>>>(        \
-          (const ETYPE::T*)x, (const ETYPE::T*)gamma, (const ETYPE::T*)beta,            \
-          (ETYPE::T*)y, mean, rstd, H, eps);                                            \
+      launch_norm_fwd<ETYPE, false>(x, gamma, beta, y, mean, rstd, R, H, eps, stream,   \
+                                    th);                                                \
   }                                                                                     \
   extern "C" void ln_bwd_dx_##SUFF(const void* dy, const void* x, const void* gamma,    \
                                    const float* mean, const float* rstd, void* dx,      \
                                    int64_t R, int H, bool rms, hipStream_t stream) {    \
     int th = row_block_threads(H, ETYPE::VEC);                                          \
     if (rms)                                                                            \
-      norm_bwd_dx_kernel<ETYPE, true><<<dim3((uint32_t)R), dim3(th), 0, stream>>>(      \
-          (const ETYPE::T*)dy, (const ETYPE::T*)x, (const ETYPE::T*)gamma, mean, rstd,  \
-          (ETYPE::T*)dx, H);                                                            \
+      launch_norm_bwd_dx<ETYPE, true>(dy, x, gamma, mean, rstd, dx, R, H, stream, th);  \
     else                                                                                \
-      norm_bwd_dx_kernel<ETYPE, false><<<dim3((uint32_t)R), dim3(th), 0, stream>>>(     \
-          (const ETYPE::T*)dy, (const ETYPE::T*)x, (const ETYPE::T*)gamma, mean, rstd,  \
-          (ETYPE::T*)dx, H);                                                            \
+      launch_norm_bwd_dx<ETYPE, false>(dy, x, gamma, mean, rstd, dx, R, H, stream,      \
+                                       th);                                             \
   }                                                                                     \
   extern "C" void ln_bwd_wgrad_##SUFF(const void* dy, const void* x, const float* mean, \
                                       const float* rstd, float* pgamma, float* pbeta,   \
                                       void* dgamma, void* dbeta, int64_t R, int H,      \
                                       int P, bool rms, hipStream_t stream) {            \
-    dim3 grid1(CDIV(H, 256), P);                                                        \
-    if (rms)                                                                            \
-      norm_bwd_wgrad_partial_kernel<ETYPE, true><<<grid1, dim3(256), 0, stream>>>(      \
-          (const ETYPE::T*)dy, (const ETYPE::T*)x, mean, rstd, pgamma, pbeta, R, H);    \
-    else                                                                                \
-      norm_bwd_wgrad_partial_kernel<ETYPE, false><<<grid1, dim3(256), 0, stream>>>(     \
-          (const ETYPE::T*)dy, (const ETYPE::T*)x, mean, rstd, pgamma, pbeta, R, H);    \
-    wgrad_fold_kernel<ETYPE><<<dim3(CDIV(H, 64)), dim3(512), 0, stream>>>(              \
-        pgamma, pbeta, (ETYPE::T*)dgamma, (ETYPE::T*)dbeta, P, H);                      \
+    if (H % ETYPE::VEC == 0) {                                                          \
+      dim3 grid1(CDIV(H / ETYPE::VEC, 256), P);                                         \
+      if (rms)                                                                          \
+        norm_bwd_wgrad_partial_kernel<ETYPE, true><<<grid1, dim3(256), 0, stream>>>(    \
+            (const ETYPE::T*)dy, (const ETYPE::T*)x, mean, rstd, pgamma, pbeta, R, H);  \
+      else                                                                              \
+        norm_bwd_wgrad_partial_kernel<ETYPE, false><<<grid1, dim3(256), 0, stream>>>(   \
+            (const ETYPE::T*)dy, (const ETYPE::T*)x, mean, rstd, pgamma, pbeta, R, H);  \
+    } else {                                                                            \
+      dim3 grid1(CDIV(H, 256), P);                                                      \
+      if (rms)                                                                          \
+        norm_bwd_wgrad_partial_scalar_kernel<ETYPE, true>                               \
+            <<<grid1, dim3(256), 0, stream>>>((const ETYPE::T*)dy,                      \
+                                              (const ETYPE::T*)x, mean, rstd, pgamma,   \
+                                              pbeta, R, H);                             \
+      else                                                                              \
+        norm_bwd_wgrad_partial_scalar_kernel<ETYPE, false>                              \
+            <<<grid1, dim3(256), 0, stream>>>((const ETYPE::T*)dy,                      \
+                                              (const ETYPE::T*)x, mean, rstd, pgamma,   \
+                                              pbeta, R, H);                             \
+    }                                                                                   \
+    (void)dgamma;                                                                       \
+    (void)dbeta;                                                                        \
   }
 
 NORM_LAUNCHERS(bf16, BF16Elem)
