@@ -164,7 +164,7 @@ def main():
     dp = dutil.data_parallel_size
     micro = args.micro_batch
     if micro is None:
-        micro = 32 if device.type == "cuda" else 2  # best measured 345M shape
+        micro = 48 if device.type == "cuda" else 2  # best measured 345M shape
     global_batch = micro * dp * args.acc
 
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32

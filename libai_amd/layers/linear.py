@@ -111,12 +111,11 @@ class Linear1D(nn.Module):
     def forward(self, x):
         if self.parallel == "col":
             x = copy_to_tensor_parallel_region(x)
-            out = F.linear(x, self.weight)
             if self.skip_bias_add:
-                return out, self.bias
-            if self.bias is not None:
-                out = out + self.bias
-            return out
+                return F.linear(x, self.weight), self.bias
+            # bias fused into the GEMM epilogue (a separate add costs a full
+            # HBM round-trip over the 3h-wide qkv activations)
+            return F.linear(x, self.weight, self.bias)
         if self.parallel == "row":
             out = F.linear(x, self.weight)
             out = reduce_from_tensor_parallel_region(out)
@@ -125,12 +124,9 @@ class Linear1D(nn.Module):
             if self.bias is not None:
                 out = out + self.bias
             return out
-        out = F.linear(x, self.weight)
         if self.skip_bias_add:
-            return out, self.bias
-        if self.bias is not None:
-            out = out + self.bias
-        return out
+            return F.linear(x, self.weight), self.bias
+        return F.linear(x, self.weight, self.bias)
 
     def extra_repr(self):
         return (
