@@ -164,7 +164,10 @@ def main():
     dp = dutil.data_parallel_size
     micro = args.micro_batch
     if micro is None:
-        micro = 48 if device.type == "cuda" else 2  # best measured 345M shape
+        # best-measured per-model micro batch on one MI355X (288 GB)
+        per_model = {"gpt2": 48, "bert-large": 32, "llama7b": 4, "llama1b": 8,
+                     "vit-l16": 64}
+        micro = per_model[args.model] if device.type == "cuda" else 2
     global_batch = micro * dp * args.acc
 
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
