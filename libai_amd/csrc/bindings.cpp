@@ -158,7 +158,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ln_bwd(
   auto dgamma = torch::empty_like(gamma);
   auto dbeta = need_dbeta ? torch::empty_like(gamma) : torch::Tensor();
   const int64_t gxw = std::max<int64_t>(1, (H / 8 + 255) / 256);
-  const int P = (int)std::min<int64_t>(std::max<int64_t>(1, 2048 / gxw),
+  const int P = (int)std::min<int64_t>(std::max<int64_t>(1, 1024 / gxw),
                                        std::max<int64_t>(1, R / 4));
   auto partial = torch::empty({(need_dbeta ? 2L : 1L) * P, (int64_t)H},
                               x.options().dtype(torch::kFloat32));
@@ -211,7 +211,7 @@ torch::Tensor colsum(torch::Tensor x, int64_t W) {
   CHECK_IN(x);
   const int64_t R = x.numel() / W;
   const int64_t gxw = std::max<int64_t>(1, (W / 8 + 255) / 256);
-  const int P = (int)std::min<int64_t>(std::max<int64_t>(1, 2048 / gxw),
+  const int P = (int)std::min<int64_t>(std::max<int64_t>(1, 1024 / gxw),
                                        std::max<int64_t>(1, R / 4));
   auto partial = torch::empty({P, W}, x.options().dtype(torch::kFloat32));
   auto out = torch::empty({W}, x.options());

@@ -40,20 +40,31 @@ def kernels(pat, steps):
 def pmc(pat, filt):
     con, sfx = open_db(pat)
     strings = dict(con.execute(f"SELECT id, string FROM rocpd_string_{sfx}"))
-    # counter tables
+    # schema introspection (rocprofv3 writes rocpd_pmc_event + rocpd_info_pmc)
+    def cols(t):
+        return [r[1] for r in con.execute(f"PRAGMA table_info({t})")]
+    ev = f"rocpd_pmc_event_{sfx}"
+    info = f"rocpd_info_pmc_{sfx}"
     try:
+        evc, infc = cols(ev), cols(info)
+        print("# pmc_event cols:", evc, file=sys.stderr)
+        print("# info_pmc cols:", infc, file=sys.stderr)
+        # best-effort join: event has (pmc_id, value, <dispatch link>)
+        dispatch_col = next(c for c in evc if "dispatch" in c or c == "event_id")
+        name_col = "name" if "name" in infc else infc[1]
         rows = con.execute(
-            f"SELECT ks.display_name, di.name, SUM(c.value), COUNT(*) "
-            f"FROM rocpd_counter_{sfx} c "
-            f"JOIN rocpd_kernel_dispatch_{sfx} k ON c.dispatch_id=k.id "
+            f"SELECT ks.display_name, di.{name_col}, SUM(e.value), COUNT(*) "
+            f"FROM {ev} e "
+            f"JOIN rocpd_kernel_dispatch_{sfx} k ON e.{dispatch_col}=k.id "
             f"JOIN rocpd_info_kernel_symbol_{sfx} ks ON k.kernel_id=ks.id "
-            f"JOIN rocpd_info_counter_{sfx} di ON c.counter_id=di.id "
+            f"JOIN {info} di ON e.pmc_id=di.id "
             f"GROUP BY 1, 2").fetchall()
-    except sqlite3.OperationalError as e:
+    except (sqlite3.OperationalError, StopIteration) as e:
         print("schema probe:", e)
         for r in con.execute("SELECT name FROM sqlite_master WHERE type='table'"):
             if 'counter' in r[0] or 'pmc' in r[0]:
-                print("  table:", r[0])
+                t = r[0]
+                print("  table:", t, cols(t))
         return
     agg = {}
     for n, cn, v, cnt in rows:
