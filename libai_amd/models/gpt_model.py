@@ -215,7 +215,16 @@ class GPTForPreTraining(nn.Module):
     def from_config(cls, cfg):
         return {"cfg": cfg}
 
-    def forward(self, input_ids, labels=None):
+    def forward(self, input_ids, labels=None, past_key_values=None,
+                use_cache=False):
+        if past_key_values is not None or use_cache:
+            # incremental-decode path for the generation pipelines
+            out = self.GPT_model(input_ids, past_key_values=past_key_values,
+                                 use_cache=use_cache)
+            if use_cache:
+                logits, presents = out
+                return {"prediction_scores": logits, "past_key_values": presents}
+            return {"prediction_scores": out}
         logits = self.GPT_model(input_ids)
         if labels is not None:
             return self.loss_func(logits, labels)
