@@ -152,9 +152,22 @@ class Generator:
     def _model(self):
         return self._gen_model if self._gen_model is not None else self
 
+    _enc_input_ids = None  # set by generate() for encoder-decoder models
+
     @torch.no_grad()
     def _step_logits(self, input_ids, past):
         model = self._model
+        if self._enc_input_ids is not None:
+            # encoder-decoder (T5-style): recompute the decoder over the
+            # generated prefix each step (reference generation_utils.py's
+            # enc-dec branch; cache-free v1 at pipeline scale)
+            out = model(encoder_input_ids=self._enc_input_ids,
+                        decoder_input_ids=input_ids)
+            logits = out["prediction_scores"] if isinstance(out, dict) else out
+            logits = logits[:, -1, :].float()
+            if du.get_dist_util().tensor_parallel_size > 1:
+                logits = gather_from_tensor_parallel_region(logits)
+            return logits, None
         feed = input_ids if past is None else input_ids[:, -1:]
         out = model(input_ids=feed, past_key_values=past, use_cache=True)
         if isinstance(out, dict):
@@ -303,8 +316,21 @@ class Generator:
     def generate(self, input_ids, max_length=64, min_length=0, do_sample=False,
                  num_beams=1, temperature=1.0, top_k=0, top_p=1.0,
                  repetition_penalty=1.0, no_repeat_ngram_size=0, eos_token_id=None,
-                 pad_token_id=0, length_penalty=1.0, **kwargs):
-        """Dispatcher (reference: generation_utils.py:787+)."""
+                 pad_token_id=0, length_penalty=1.0, encoder_input_ids=None,
+                 decoder_start_token_id=None, **kwargs):
+        """Dispatcher (reference: generation_utils.py:787+).
+
+        For encoder-decoder models pass ``encoder_input_ids``; ``input_ids``
+        then seeds the decoder (defaults to a [batch, 1] tensor of
+        ``decoder_start_token_id``).
+        """
+        self._enc_input_ids = encoder_input_ids
+        if encoder_input_ids is not None and input_ids is None:
+            start = decoder_start_token_id if decoder_start_token_id is not None \
+                else pad_token_id
+            input_ids = torch.full((encoder_input_ids.shape[0], 1), start,
+                                   dtype=torch.long,
+                                   device=encoder_input_ids.device)
         processors = LogitsProcessorList()
         if repetition_penalty != 1.0:
             processors.append(RepetitionPenaltyLogitsProcessor(repetition_penalty))
