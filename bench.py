@@ -165,7 +165,7 @@ def main():
     micro = args.micro_batch
     if micro is None:
         # best-measured per-model micro batch on one MI355X (288 GB)
-        per_model = {"gpt2": 48, "bert-large": 32, "llama7b": 4, "llama1b": 8,
+        per_model = {"gpt2": 48, "bert-large": 96, "llama7b": 12, "llama1b": 8,
                      "vit-l16": 64}
         micro = per_model[args.model] if device.type == "cuda" else 2
     global_batch = micro * dp * args.acc
