@@ -1,0 +1,73 @@
+"""EventStorage/HistoryBuffer metrics store + Mixup augmentation.
+
+Reference behavior: libai/utils/events.py:265-450, history_buffer.py,
+mixup in data/build.py + engine/default.py:509-515.
+"""
+
+import json
+import os
+
+import torch
+
+from libai_amd.data.mixup import Mixup
+from libai_amd.utils.events import (
+    EventStorage,
+    HistoryBuffer,
+    JSONWriter,
+    get_event_storage,
+)
+
+
+def test_history_buffer_median_latest():
+    hb = HistoryBuffer(max_length=8)
+    for i in range(20):
+        hb.update(float(i), i)
+    assert hb.latest() == 19.0
+    # only the last 8 kept: median over window 8 = median(12..19)
+    assert 14.0 <= hb.median(8) <= 16.0
+
+
+def test_event_storage_scalars_and_scope():
+    with EventStorage(start_iter=5) as storage:
+        assert get_event_storage() is storage
+        storage.put_scalar("loss", 2.0)
+        with storage.name_scope("eval"):
+            storage.put_scalar("acc", 0.5, smoothing_hint=False)
+        storage.step()
+        storage.put_scalar("loss", 1.0)
+        storage.step()
+        storage.put_scalar("loss", 4.0)
+        latest = storage.latest()
+        assert latest["loss"] == (4.0, 7)
+        assert latest["eval/acc"] == (0.5, 5)
+        sm = storage.latest_with_smoothing_hint(window_size=3)
+        assert sm["loss"][0] == 2.0  # median of [2, 1, 4]
+        assert sm["eval/acc"][0] == 0.5  # unsmoothed
+
+
+def test_json_writer(tmp_path):
+    path = os.path.join(tmp_path, "metrics.json")
+    with EventStorage() as storage:
+        w = JSONWriter(path)
+        storage.put_scalar("loss", 3.0)
+        w.write()
+        storage.step()
+        storage.put_scalar("loss", 2.0)
+        w.write()
+        w.close()
+    lines = [json.loads(l) for l in open(path)]
+    assert len(lines) == 2 and lines[1]["loss"] == 2.0
+
+
+def test_mixup_soft_labels():
+    torch.manual_seed(0)
+    mix = Mixup(mixup_alpha=0.8, cutmix_alpha=1.0, prob=1.0,
+                label_smoothing=0.1, num_classes=10)
+    imgs = torch.randn(8, 3, 16, 16)
+    labels = torch.randint(0, 10, (8,))
+    out_imgs, out_labels = mix(imgs.clone(), labels)
+    assert out_imgs.shape == imgs.shape
+    assert out_labels.shape == (8, 10)
+    # soft labels sum to 1 and reflect smoothing
+    assert torch.allclose(out_labels.sum(1), torch.ones(8), atol=1e-5)
+    assert (out_labels > 0).all()
