@@ -5,13 +5,14 @@ fused_bias_add_dropout call sites).  The output bias/dropout/residual is
 fused by the caller via ``residual=``, like MultiheadAttention.
 """
 
+import torch
 from torch import nn
 
 from ..ops.fused_bias import bias_dropout_add, bias_gelu
 from .activation import build_activation
 from .linear import Linear1D
 
-__all__ = ["MLP"]
+__all__ = ["MLP", "GatedMLP"]
 
 
 class MLP(nn.Module):
@@ -54,5 +55,47 @@ class MLP(nn.Module):
         out, bias = self.dense_4h_to_h(inter)
         return bias_dropout_add(
             out, bias=bias, residual=residual, p=self.output_dropout_prob,
+            training=self.training,
+        )
+
+
+class GatedMLP(nn.Module):
+    """MT5/GLM-style gated MLP: out = W2 (act(x W1) * (x W3)), bias-free.
+
+    Reference: projects/MT5 fused_fast_gelu_mul (mlp_layer.py:123) and the
+    Llama SwiGLU variant.  The fused [gate|up] column
+    projection keeps one GEMM on the 2*ffn width; activation "silu" uses
+    the fused SwiGLU HIP kernel, "gelu" composes gelu(gate)*up.
+    """
+
+    def __init__(self, hidden_size, ffn_hidden_size, output_dropout_prob=0.0,
+                 activation="gelu", init_method=nn.init.xavier_normal_,
+                 output_layer_init_method=None, *, layer_idx=0):
+        super().__init__()
+        self.output_dropout_prob = output_dropout_prob
+        self.activation = activation
+        output_layer_init_method = output_layer_init_method or init_method
+        self.gate_up_proj = Linear1D(
+            hidden_size, 2 * ffn_hidden_size, bias=False, parallel="col",
+            init_method=init_method, layer_idx=layer_idx,
+        )
+        self.down_proj = Linear1D(
+            ffn_hidden_size, hidden_size, bias=False, parallel="row",
+            init_method=output_layer_init_method, skip_bias_add=True,
+            layer_idx=layer_idx,
+        )
+
+    def forward(self, hidden_states, residual=None):
+        gu = self.gate_up_proj(hidden_states)
+        if self.activation == "silu":
+            from ..ops.swiglu import swiglu
+
+            inter = swiglu(gu)
+        else:
+            gate, up = gu.chunk(2, dim=-1)
+            inter = torch.nn.functional.gelu(gate, approximate="tanh") * up
+        out, _ = self.down_proj(inter)
+        return bias_dropout_add(
+            out, bias=None, residual=residual, p=self.output_dropout_prob,
             training=self.training,
         )

@@ -35,6 +35,8 @@ class TransformerLayer(nn.Module):
         apply_query_key_layer_scaling=False,
         apply_residual_post_layernorm=False,
         attn_mask_type=AttnMaskType.padding,
+        mlp_type="dense",
+        activation="gelu",
         *,
         layer_idx=0,
     ):
@@ -73,15 +75,28 @@ class TransformerLayer(nn.Module):
             self.post_cross_attention_layernorm = LayerNorm(
                 hidden_size, eps=layernorm_epsilon, layer_idx=layer_idx
             )
-        self.mlp = MLP(
-            hidden_size, ffn_hidden_size,
-            output_dropout_prob=output_dropout_prob,
-            init_method=init_method,
-            output_layer_init_method=output_layer_init_method,
-            bias_gelu_fusion=bias_gelu_fusion,
-            bias_dropout_fusion=bias_dropout_fusion,
-            layer_idx=layer_idx,
-        )
+        if mlp_type == "gated":
+            from .mlp import GatedMLP
+
+            self.mlp = GatedMLP(
+                hidden_size, ffn_hidden_size,
+                output_dropout_prob=output_dropout_prob,
+                activation=activation,
+                init_method=init_method,
+                output_layer_init_method=output_layer_init_method,
+                layer_idx=layer_idx,
+            )
+        else:
+            self.mlp = MLP(
+                hidden_size, ffn_hidden_size,
+                output_dropout_prob=output_dropout_prob,
+                init_method=init_method,
+                output_layer_init_method=output_layer_init_method,
+                bias_gelu_fusion=bias_gelu_fusion,
+                bias_dropout_fusion=bias_dropout_fusion,
+                activation=activation,
+                layer_idx=layer_idx,
+            )
 
     def forward(
         self,

@@ -79,6 +79,8 @@ class T5Model(nn.Module):
         apply_query_key_layer_scaling=False,
         apply_residual_post_layernorm=False,
         amp_enabled=False,
+        mlp_type="dense",
+        activation="gelu",
     ):
         super().__init__()
         init_method = init_method_normal(initializer_range)
@@ -107,6 +109,8 @@ class T5Model(nn.Module):
                 attn_mask_type=(
                     AttnMaskType.causal if is_decoder else AttnMaskType.padding
                 ),
+                mlp_type=mlp_type,
+                activation=activation,
                 layer_idx=i,
             )
 
