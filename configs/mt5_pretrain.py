@@ -1,10 +1,11 @@
 """MT5-style pretraining: T5 with the gated-gelu MLP (reference:
 projects/MT5 -- fused_fast_gelu_mul gated MLP, shared enc/dec embedding)."""
 
-from .common.models.t5 import cfg as t5_cfg
 from .t5_pretrain import dataloader, model, optim, train  # noqa: F401
 
-t5_cfg.mlp_type = "gated"
-t5_cfg.activation = "gelu"
+# mutate the model's own cfg node (each config import gets its own module
+# instance, so the common-file `cfg` alias would be a different object)
+model.cfg.mlp_type = "gated"
+model.cfg.activation = "gelu"
 
 train.update(output_dir="./output/mt5_pretrain")

@@ -21,6 +21,7 @@ import ast
 import builtins
 import copy
 import importlib
+import importlib.util
 import os
 import uuid
 from collections import abc

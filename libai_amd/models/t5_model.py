@@ -148,6 +148,8 @@ class T5Model(nn.Module):
             "apply_query_key_layer_scaling": cfg.get("apply_query_key_layer_scaling", False),
             "apply_residual_post_layernorm": cfg.get("apply_residual_post_layernorm", False),
             "amp_enabled": cfg.get("amp_enabled", False),
+            "mlp_type": cfg.get("mlp_type", "dense"),
+            "activation": cfg.get("activation", "gelu"),
         }
 
     def _run(self, layer, *args, **kw):
