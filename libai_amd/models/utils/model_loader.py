@@ -265,3 +265,83 @@ class LlamaLoaderHuggerFace(ModelLoaderHuggerFace):
             if "g" in d and "u" in d:
                 out[base + "mlp.gate_up_proj.weight"] = torch.cat([d["g"], d["u"]], 0)
         return out
+
+
+class ViTLoaderHuggerFace(ModelLoaderHuggerFace):
+    """HF ViT (google/vit-*) -> libai_amd VisionTransformer.
+
+    Reference: model_loader/vit_loader.py.  HF stores separate q/k/v
+    projections; our attention uses the fused per-head-interleaved qkv.
+    """
+
+    def _convert_state_dict(self, hf):
+        cfg = self.cfg
+        nh = getattr(cfg, "num_heads", None) if cfg is not None else None
+        out = {}
+        qkv = {}
+        for k, v in hf.items():
+            k = k[len("vit."):] if k.startswith("vit.") else k
+            if k == "embeddings.cls_token":
+                out["embedding.cls_token"] = v
+            elif k == "embeddings.position_embeddings":
+                out["embedding.pos_embed"] = v
+            elif k == "embeddings.patch_embeddings.projection.weight":
+                out["embedding.patch_embed.proj.weight"] = v
+            elif k == "embeddings.patch_embeddings.projection.bias":
+                out["embedding.patch_embed.proj.bias"] = v
+            elif k == "layernorm.weight":
+                out["norm.weight"] = v
+            elif k == "layernorm.bias":
+                out["norm.bias"] = v
+            elif k == "classifier.weight":
+                out["head.weight"] = v
+            elif k == "classifier.bias":
+                out["head.bias"] = v
+            elif k.startswith("encoder.layer."):
+                parts = k.split(".")
+                i, rest = parts[2], ".".join(parts[3:])
+                base = f"blocks.{i}."
+                m = {
+                    "layernorm_before.weight": "input_layernorm.weight",
+                    "layernorm_before.bias": "input_layernorm.bias",
+                    "layernorm_after.weight": "post_attention_layernorm.weight",
+                    "layernorm_after.bias": "post_attention_layernorm.bias",
+                    "attention.output.dense.weight": "self_attention.dense.weight",
+                    "attention.output.dense.bias": "self_attention.dense.bias",
+                    "intermediate.dense.weight": "mlp.dense_h_to_4h.weight",
+                    "intermediate.dense.bias": "mlp.dense_h_to_4h.bias",
+                    "output.dense.weight": "mlp.dense_4h_to_h.weight",
+                    "output.dense.bias": "mlp.dense_4h_to_h.bias",
+                }
+                if rest in m:
+                    out[base + m[rest]] = v
+                elif rest.startswith("attention.attention."):
+                    which = rest.split(".")[2]  # query/key/value
+                    kind = rest.split(".")[3]
+                    qkv.setdefault((i, kind), {})[which] = v
+        for (i, kind), d in qkv.items():
+            if len(d) == 3:
+                stacked = torch.cat([d["query"], d["key"], d["value"]], dim=0)
+                out[f"blocks.{i}.self_attention.query_key_value.{kind}"] = \
+                    GPT2LoaderHuggerFace._interleave_qkv(stacked, nh)
+        return out
+
+
+class RobertaLoaderHuggerFace(BertLoaderHuggerFace):
+    """HF RoBERTa -> libai_amd RobertaModel (reference: roberta_loader.py).
+
+    RoBERTa is BERT-shaped with a `roberta.` prefix and no NSP head; reuse
+    the BERT mapping after re-prefixing the keys onto the `roberta.` module.
+    """
+
+    def _convert_state_dict(self, hf):
+        renamed = {}
+        for k, v in hf.items():
+            if k.startswith("roberta."):
+                k = "bert." + k[len("roberta."):]
+            renamed[k] = v
+        bert_out = super()._convert_state_dict(renamed)
+        return {
+            ("roberta." + k[len("bert."):]) if k.startswith("bert.") else k: v
+            for k, v in bert_out.items()
+        }

@@ -1,0 +1,90 @@
+"""HF-format loader round-trips (reference: models/utils/model_loader/*).
+
+Builds an HF-shaped state dict from a randomly-initialized libai_amd model
+via the INVERSE key mapping, loads it through the loader, and requires the
+restored model to reproduce the source model's forward exactly.
+"""
+
+import torch
+
+from libai_amd.models import VisionTransformer
+from libai_amd.models.utils.model_loader import (
+    GPT2LoaderHuggerFace,
+    ViTLoaderHuggerFace,
+)
+from libai_amd.utils import distributed as du
+
+du.setup_dist_util({})
+
+
+def _deinterleave_qkv(w, num_heads):
+    """Inverse of GPT2LoaderHuggerFace._interleave_qkv."""
+    three_h = w.shape[0]
+    h = three_h // 3
+    hs = h // num_heads
+    qs, ks, vs = [], [], []
+    for head in range(num_heads):
+        base = head * 3 * hs
+        qs.append(w[base: base + hs])
+        ks.append(w[base + hs: base + 2 * hs])
+        vs.append(w[base + 2 * hs: base + 3 * hs])
+    return torch.cat(qs), torch.cat(ks), torch.cat(vs)
+
+
+def test_vit_hf_loader_roundtrip():
+    torch.manual_seed(0)
+    nh = 4
+    src = VisionTransformer(img_size=32, patch_size=8, embed_dim=64, depth=2,
+                            num_heads=nh, num_classes=10).eval()
+    sd = src.state_dict()
+
+    hf = {}
+    hf["vit.embeddings.cls_token"] = sd["embedding.cls_token"]
+    hf["vit.embeddings.position_embeddings"] = sd["embedding.pos_embed"]
+    hf["vit.embeddings.patch_embeddings.projection.weight"] = \
+        sd["embedding.patch_embed.proj.weight"]
+    hf["vit.embeddings.patch_embeddings.projection.bias"] = \
+        sd["embedding.patch_embed.proj.bias"]
+    hf["vit.layernorm.weight"] = sd["norm.weight"]
+    hf["vit.layernorm.bias"] = sd["norm.bias"]
+    hf["classifier.weight"] = sd["head.weight"]
+    hf["classifier.bias"] = sd["head.bias"]
+    inv = {
+        "input_layernorm.weight": "layernorm_before.weight",
+        "input_layernorm.bias": "layernorm_before.bias",
+        "post_attention_layernorm.weight": "layernorm_after.weight",
+        "post_attention_layernorm.bias": "layernorm_after.bias",
+        "self_attention.dense.weight": "attention.output.dense.weight",
+        "self_attention.dense.bias": "attention.output.dense.bias",
+        "mlp.dense_h_to_4h.weight": "intermediate.dense.weight",
+        "mlp.dense_h_to_4h.bias": "intermediate.dense.bias",
+        "mlp.dense_4h_to_h.weight": "output.dense.weight",
+        "mlp.dense_4h_to_h.bias": "output.dense.bias",
+    }
+    for i in range(2):
+        for ours, hfk in inv.items():
+            hf[f"vit.encoder.layer.{i}.{hfk}"] = sd[f"blocks.{i}.{ours}"]
+        for kind in ("weight", "bias"):
+            q, k, v = _deinterleave_qkv(
+                sd[f"blocks.{i}.self_attention.query_key_value.{kind}"], nh
+            )
+            hf[f"vit.encoder.layer.{i}.attention.attention.query.{kind}"] = q
+            hf[f"vit.encoder.layer.{i}.attention.attention.key.{kind}"] = k
+            hf[f"vit.encoder.layer.{i}.attention.attention.value.{kind}"] = v
+
+    class _Cfg:
+        num_heads = nh
+
+    dst = VisionTransformer(img_size=32, patch_size=8, embed_dim=64, depth=2,
+                            num_heads=nh, num_classes=10).eval()
+    loader = ViTLoaderHuggerFace(dst, _Cfg())
+    converted = loader._convert_state_dict(hf)
+    missing, unexpected = dst.load_state_dict(converted, strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+
+    x = torch.randn(2, 3, 32, 32)
+    with torch.no_grad():
+        a = src(images=x)["prediction_scores"]
+        b = dst(images=x)["prediction_scores"]
+    assert torch.equal(a, b)
