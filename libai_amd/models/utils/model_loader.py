@@ -345,3 +345,75 @@ class RobertaLoaderHuggerFace(BertLoaderHuggerFace):
             ("roberta." + k[len("bert."):]) if k.startswith("bert.") else k: v
             for k, v in bert_out.items()
         }
+
+
+class SwinLoaderHuggerFace(ModelLoaderHuggerFace):
+    """HF Swin (microsoft/swin-*) -> libai_amd SwinTransformer.
+
+    Reference: model_loader/swin_loader.py.  HF stores per-block modules
+    under encoder.layers.{i}.blocks.{j}; ours are layers.{i}.0.{j} with the
+    patch-merging downsample at layers.{i}.1.  HF's separate q/k/v stack
+    into our fused qkv (q|k|v over the whole width; window attention is
+    not head-interleaved).
+    """
+
+    def _convert_state_dict(self, hf):
+        out = {}
+        qkv = {}
+        blk = {
+            "layernorm_before.weight": "norm1.weight",
+            "layernorm_before.bias": "norm1.bias",
+            "layernorm_after.weight": "norm2.weight",
+            "layernorm_after.bias": "norm2.bias",
+            "attention.self.relative_position_bias_table":
+                "attn.relative_position_bias_table",
+            "attention.output.dense.weight": "attn.proj.weight",
+            "attention.output.dense.bias": "attn.proj.bias",
+            "intermediate.dense.weight": "mlp.0.weight",
+            "intermediate.dense.bias": "mlp.0.bias",
+            "output.dense.weight": "mlp.3.weight",
+            "output.dense.bias": "mlp.3.bias",
+        }
+        for k, v in hf.items():
+            if k.startswith("swin."):
+                k = k[len("swin."):]
+            if k == "embeddings.patch_embeddings.projection.weight":
+                out["patch_embed.weight"] = v
+            elif k == "embeddings.patch_embeddings.projection.bias":
+                out["patch_embed.bias"] = v
+            elif k == "embeddings.norm.weight":
+                out["patch_norm.weight"] = v
+            elif k == "embeddings.norm.bias":
+                out["patch_norm.bias"] = v
+            elif k == "layernorm.weight":
+                out["norm.weight"] = v
+            elif k == "layernorm.bias":
+                out["norm.bias"] = v
+            elif k == "classifier.weight":
+                out["head.weight"] = v
+            elif k == "classifier.bias":
+                out["head.bias"] = v
+            elif k.startswith("encoder.layers."):
+                parts = k.split(".")
+                i = parts[2]
+                if parts[3] == "blocks":
+                    j, rest = parts[4], ".".join(parts[5:])
+                    base = f"layers.{i}.0.{j}."
+                    if rest in blk:
+                        out[base + blk[rest]] = v
+                    elif rest.startswith("attention.self."):
+                        which = rest.split(".")[2]  # query/key/value
+                        kind = rest.split(".")[3]
+                        qkv.setdefault((i, j, kind), {})[which] = v
+                elif parts[3] == "downsample":
+                    rest = ".".join(parts[4:])
+                    m = {"reduction.weight": "reduction.weight",
+                         "norm.weight": "norm.weight", "norm.bias": "norm.bias"}
+                    if rest in m:
+                        out[f"layers.{i}.1.{m[rest]}"] = v
+        for (i, j, kind), d in qkv.items():
+            if len(d) == 3:
+                out[f"layers.{i}.0.{j}.attn.qkv.{kind}"] = torch.cat(
+                    [d["query"], d["key"], d["value"]], dim=0
+                )
+        return out
