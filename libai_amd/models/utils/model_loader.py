@@ -417,3 +417,36 @@ class SwinLoaderHuggerFace(ModelLoaderHuggerFace):
                     [d["query"], d["key"], d["value"]], dim=0
                 )
         return out
+
+
+class SwinV2LoaderHuggerFace(SwinLoaderHuggerFace):
+    """HF SwinV2 (microsoft/swinv2-*) -> libai_amd SwinTransformerV2.
+
+    Reference: model_loader/swinv2_loader.py.  On top of the Swin mapping,
+    V2 adds the per-head logit scale and the continuous-position-bias MLP
+    (HF: continuous_position_bias_mlp / logit_scale).
+    """
+
+    def _convert_state_dict(self, hf):
+        out = super()._convert_state_dict(hf)
+        for k, v in hf.items():
+            if k.startswith("swin."):
+                k = k[len("swin."):]
+            elif k.startswith("swinv2."):
+                k = k[len("swinv2."):]
+            if not k.startswith("encoder.layers."):
+                continue
+            parts = k.split(".")
+            i = parts[2]
+            if parts[3] != "blocks":
+                continue
+            j, rest = parts[4], ".".join(parts[5:])
+            base = f"layers.{i}.0.{j}.attn."
+            if rest == "attention.self.logit_scale":
+                out[base + "logit_scale"] = v
+            elif rest.startswith(
+                "attention.self.continuous_position_bias_mlp."
+            ):
+                idx_rest = rest[len("attention.self.continuous_position_bias_mlp."):]
+                out[base + "cpb_mlp." + idx_rest] = v
+        return out
