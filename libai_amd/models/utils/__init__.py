@@ -7,6 +7,10 @@ from .model_loader import (
     LlamaLoaderHuggerFace,
     ModelLoaderHuggerFace,
     ModelLoaderLiBai,
+    RobertaLoaderHuggerFace,
+    SwinLoaderHuggerFace,
+    SwinV2LoaderHuggerFace,
+    ViTLoaderHuggerFace,
 )
 
 __all__ += [
@@ -14,5 +18,9 @@ __all__ += [
     "ModelLoaderHuggerFace",
     "GPT2LoaderHuggerFace",
     "BertLoaderHuggerFace",
+    "RobertaLoaderHuggerFace",
     "LlamaLoaderHuggerFace",
+    "ViTLoaderHuggerFace",
+    "SwinLoaderHuggerFace",
+    "SwinV2LoaderHuggerFace",
 ]
