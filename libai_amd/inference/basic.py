@@ -151,6 +151,14 @@ class TextClassificationPipeline(BasePipeline):
 
 
 class ImageClassificationPipeline(BasePipeline):
+    """Top-k image classification; pass ``class_names`` (list indexed by
+    class id, e.g. the ImageNet-1k label table that the reference ships in
+    inference/utils/imagenet_class.py) to get readable labels."""
+
+    def __init__(self, *args, class_names=None, **kwargs):
+        super().__init__(*args, **kwargs)
+        self.class_names = class_names
+
     def preprocess(self, inputs, **kwargs):
         if torch.is_tensor(inputs):
             images = inputs if inputs.dim() == 4 else inputs.unsqueeze(0)
@@ -166,6 +174,10 @@ class ImageClassificationPipeline(BasePipeline):
         logits = model_outputs["prediction_scores"].float()
         probs = torch.softmax(logits, dim=-1)
         scores, idx = probs.topk(min(topk, probs.shape[-1]), dim=-1)
-        return [
-            {"classes": i.tolist(), "scores": s.tolist()} for i, s in zip(idx, scores)
-        ]
+        results = []
+        for i, s in zip(idx, scores):
+            r = {"classes": i.tolist(), "scores": s.tolist()}
+            if self.class_names:
+                r["labels"] = [self.class_names[c] for c in r["classes"]]
+            results.append(r)
+        return results
