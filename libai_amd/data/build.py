@@ -14,6 +14,7 @@ __all__ = [
     "build_image_test_loader",
     "trivial_batch_collator",
     "build_train_valid_test_loaders",
+    "build_nlp_train_val_test_loader",
 ]
 
 
@@ -118,3 +119,54 @@ def build_train_valid_test_loaders(train_dataset, valid_dataset, test_dataset,
         else None
     )
     return train, valid, test
+
+
+def build_nlp_train_val_test_loader(data_prefix, splits="949,50,1",
+                                    max_seq_length=1024,
+                                    train_val_test_num_samples=(None, None, None),
+                                    train_batch_size=4, test_batch_size=4,
+                                    num_workers=4, consumed_samples=0, seed=1234,
+                                    dataset_cls=None, collate_fn=None):
+    """Split ONE indexed corpus into train/val/test by document ranges
+    (reference: libai/data/build.py:28-150).
+
+    ``splits`` is the Megatron-style comma ratio string ("949,50,1"); the
+    document index space of ``data_prefix``'s .idx/.bin pair is partitioned
+    proportionally and each partition backs its own GPT2Dataset.
+    """
+    import numpy as np
+
+    from .datasets.gpt_dataset import GPT2Dataset
+    from .indexed_dataset import MMapIndexedDataset
+
+    dataset_cls = dataset_cls or GPT2Dataset
+    indexed = MMapIndexedDataset(data_prefix)
+    n_docs = len(indexed.doc_idx) - 1
+    ratios = [float(x) for x in str(splits).split(",")]
+    while len(ratios) < 3:
+        ratios.append(0.0)
+    total = sum(ratios[:3])
+    bounds = [0]
+    acc = 0.0
+    for r in ratios[:3]:
+        acc += r
+        bounds.append(int(round(n_docs * acc / total)))
+    bounds[-1] = n_docs
+
+    names = ("train", "valid", "test")
+    datasets = []
+    for i, name in enumerate(names):
+        lo, hi = bounds[i], bounds[i + 1]
+        if hi <= lo:
+            datasets.append(None)
+            continue
+        datasets.append(dataset_cls(
+            name, indexed, documents=np.arange(lo, hi, dtype=np.int32),
+            num_samples=train_val_test_num_samples[i],
+            max_seq_length=max_seq_length, seed=seed,
+        ))
+    return build_train_valid_test_loaders(
+        datasets[0], datasets[1], datasets[2], train_batch_size,
+        test_batch_size, num_workers=num_workers,
+        consumed_samples=consumed_samples, seed=seed, collate_fn=collate_fn,
+    )

@@ -149,3 +149,30 @@ def test_roberta_dataset_no_nsp(tmp_path):
     labels = inst.get("lm_labels").tensor
     assert (labels[loss_mask == 0] == -1).all()
     assert (labels[loss_mask == 1] >= 0).all()
+
+
+def test_train_val_test_split_loader(tmp_path):
+    """One corpus -> doc-range train/val/test datasets (reference build.py:28-150)."""
+    from libai_amd.data.build import build_nlp_train_val_test_loader
+
+    prefix = str(tmp_path / "corpus")
+    b = MMapIndexedDatasetBuilder(data_file_path(prefix), np.uint16)
+    rng = np.random.RandomState(0)
+    for d in range(40):
+        b.add_item(rng.randint(10, 5000, size=200).astype(np.uint16))
+        b.end_document()
+    b.finalize(index_file_path(prefix))
+
+    train, valid, test = build_nlp_train_val_test_loader(
+        prefix, splits="80,15,5", max_seq_length=64, train_batch_size=2,
+        test_batch_size=2, num_workers=0,
+    )
+    batch = next(iter(train))
+    ids = batch.get("input_ids").tensor
+    assert ids.shape == (2, 64)
+    # val/test datasets see disjoint, smaller doc ranges
+    assert valid is not None and test is not None
+    n_tr = len(train.batch_sampler.dataset.documents) if hasattr(
+        train.batch_sampler, "dataset") else None
+    vb = next(iter(valid))
+    assert vb.get("input_ids").tensor.shape[1] == 64
