@@ -1,8 +1,10 @@
+from .blendable import BlendableDataset
 from .build import (
     build_image_test_loader,
     build_image_train_loader,
     build_nlp_test_loader,
     build_nlp_train_loader,
+    build_nlp_train_val_test_loader,
     build_train_valid_test_loaders,
     trivial_batch_collator,
 )
@@ -13,6 +15,8 @@ from .structures import DistTensorData, Instance
 __all__ = [
     "DistTensorData",
     "Instance",
+    "BlendableDataset",
+    "build_nlp_train_val_test_loader",
     "CyclicSampler",
     "Mixup",
     "SingleRoundSampler",

@@ -48,9 +48,20 @@ def _train_loader(dataset, micro_batch_size, shuffle, consumed_samples, seed,
 
 def build_nlp_train_loader(dataset, train_batch_size, test_batch_size=None,
                            sampler=None, num_workers=4, consumed_samples=0, seed=0,
-                           collate_fn=None, dataset_mixer=None, shuffle=True, **kwargs):
+                           collate_fn=None, dataset_mixer=None, weights=None,
+                           shuffle=True, **kwargs):
     if isinstance(dataset, (list, tuple)):
-        dataset = dataset[0] if len(dataset) == 1 else ConcatDataset(dataset)
+        if len(dataset) == 1:
+            dataset = dataset[0]
+        elif weights is not None or dataset_mixer is not None:
+            # weighted multi-corpus blending (reference: blendable dataset
+            # over build_blending_indices)
+            from .blendable import BlendableDataset
+
+            mixer = dataset_mixer or BlendableDataset
+            dataset = mixer(dataset, weights or [1.0] * len(dataset))
+        else:
+            dataset = ConcatDataset(dataset)
     loader = _train_loader(
         dataset, train_batch_size, shuffle, consumed_samples, seed, num_workers,
         collate_fn,

@@ -176,3 +176,26 @@ def test_train_val_test_split_loader(tmp_path):
         train.batch_sampler, "dataset") else None
     vb = next(iter(valid))
     assert vb.get("input_ids").tensor.shape[1] == 64
+
+
+def test_blendable_dataset_weights():
+    from libai_amd.data import BlendableDataset
+
+    class _Fixed(torch.utils.data.Dataset):
+        def __init__(self, tag, n=100):
+            self.tag, self.n = tag, n
+
+        def __len__(self):
+            return self.n
+
+        def __getitem__(self, i):
+            return (self.tag, i)
+
+    bd = BlendableDataset([_Fixed("a"), _Fixed("b")], weights=[0.7, 0.3],
+                          size=1000)
+    tags = [bd[i][0] for i in range(1000)]
+    frac_a = tags.count("a") / 1000
+    assert abs(frac_a - 0.7) < 0.02
+    # per-dataset sample indices advance sequentially (modulo wrap)
+    seen_a = [bd[i][1] for i in range(50) if bd[i][0] == "a"]
+    assert seen_a == sorted(seen_a)
