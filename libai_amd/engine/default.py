@@ -163,6 +163,7 @@ class DefaultTrainer(TrainerBase):
         self._trainer = EagerTrainer(
             self.model, self.train_loader, self.optimizer, acc,
             pipeline_scheduler=self.pipeline_scheduler,
+            comm_bucket_mb=try_get_key(cfg, "train.comm.bucket_mb", default=None),
         )
 
         self.checkpointer = Checkpointer(

@@ -140,7 +140,7 @@ class EagerTrainer(TrainerBase):
     """
 
     def __init__(self, model, data_loader, optimizer, grad_acc_steps=1,
-                 pipeline_scheduler=None):
+                 pipeline_scheduler=None, comm_bucket_mb=None):
         super().__init__()
         model.train()
         self.model = model
@@ -151,7 +151,8 @@ class EagerTrainer(TrainerBase):
         self.pipeline_scheduler = pipeline_scheduler
         self._overlap = False
         if pipeline_scheduler is None and hasattr(optimizer, "register_overlap_hooks"):
-            self._overlap = bool(optimizer.register_overlap_hooks())
+            self._overlap = bool(optimizer.register_overlap_hooks(
+                bucket_mb=comm_bucket_mb))
 
     def get_batch(self, data):
         from ..data.structures import Instance

@@ -210,9 +210,14 @@ class FusedAdamW(torch.optim.Optimizer):
     # so gradient accumulation sees full sums).  ZeRO-2 keeps the simpler
     # post-backward reduce-scatter.
 
-    OVERLAP_CHUNK = 32 * 1024 * 1024  # elements
+    OVERLAP_CHUNK = 32 * 1024 * 1024  # elements (= 64 MB bf16 per collective)
 
-    def register_overlap_hooks(self):
+    def register_overlap_hooks(self, bucket_mb=None):
+        """``bucket_mb`` overrides the per-collective fusion size -- the
+        explicit analog of the reference's NCCL fusion-threshold knob
+        (engine/default.py:194-199 / C11): bigger buckets amortize the
+        xGMI ring latency, smaller ones start overlapping earlier in the
+        backward.  Env LIBAI_BUCKET_MB / LIBAI_NO_OVERLAP also apply."""
         import os
 
         dutil = du.get_dist_util()
@@ -220,6 +225,11 @@ class FusedAdamW(torch.optim.Optimizer):
                 or self.zero_stage >= 2
                 or os.environ.get("LIBAI_NO_OVERLAP", "0") == "1"):
             return False
+        mb = bucket_mb or os.environ.get("LIBAI_BUCKET_MB")
+        if mb:
+            # elements assuming 2-byte grads (bf16); fp32 buckets just fuse
+            # half as many bytes per collective
+            self.OVERLAP_CHUNK = max(1, int(float(mb) * 1024 * 1024 // 2))
         self._overlap_active = False
         self._chunks = []  # (bucket, start, end, param_ids)
         self._param_chunk = {}
