@@ -67,6 +67,8 @@ def extended_attn_mask(attention_mask):
     product of visibilities (reference: bert_model.py:40-60)."""
     if attention_mask is None:
         return None
+    if attention_mask.dim() == 3:  # pre-built [b, sq, sk] (1=MASKED)
+        return attention_mask.to(torch.uint8)
     m = attention_mask.to(torch.uint8)
     visible = m.unsqueeze(1) * m.unsqueeze(2)  # [b, s, s]
     return (1 - visible).to(torch.uint8)

@@ -105,4 +105,33 @@ class RobertaForPreTraining(nn.Module):
         self.roberta.set_activation_checkpoint(enabled)
 
 
-RobertaForCausalLM = RobertaForPreTraining  # reference exposes a CLM variant alias
+class RobertaForCausalLM(RobertaForPreTraining):
+    """RoBERTa as a left-to-right LM (reference: roberta_model.py's CLM
+    variant): the padding mask is combined with a causal mask and the loss
+    is next-token CE over the shifted sequence."""
+
+    def forward(self, input_ids, attention_mask=None, tokentype_ids=None,
+                labels=None, **kwargs):
+        b, s = input_ids.shape
+        causal = torch.tril(
+            torch.ones(s, s, dtype=torch.bool, device=input_ids.device)
+        )
+        if attention_mask is not None:
+            vis = attention_mask.to(torch.bool)[:, None, :] & causal[None]
+        else:
+            vis = causal[None].expand(b, s, s)
+        # RobertaModel's extended_attn_mask passes a prebuilt [b, s, s]
+        # (1 = MASKED) through unchanged
+        masked = (~vis).to(torch.uint8)
+        seq_out, _ = self.roberta(input_ids, masked, tokentype_ids)
+        h = self.lm_head(seq_out)
+        logits = self.lm_logits(h, self.roberta.word_embeddings_weight)
+        if labels is not None:
+            shift_logits = logits[:, :-1].contiguous()
+            shift_labels = labels[:, 1:].contiguous()
+            mask = torch.ones_like(shift_labels)
+            if attention_mask is not None:
+                mask = attention_mask[:, 1:].to(mask.dtype)
+            out = self.loss_func(shift_logits, shift_labels, mask)
+            return {"lm_loss": out["lm_loss"]}
+        return {"prediction_scores": logits}

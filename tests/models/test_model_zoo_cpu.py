@@ -167,3 +167,26 @@ def test_swin_v2_fwd_bwd():
     out = m(images=torch.randn(2, 3, 32, 32), labels=torch.randint(0, 10, (2,)))
     out["losses"].backward()
     assert torch.isfinite(out["losses"])
+
+
+def test_roberta_causal_lm():
+    """True CLM head: causal masking verified (future edit leaves prefix logits)."""
+    import torch
+
+    from libai_amd.models import RobertaForCausalLM
+
+    torch.manual_seed(0)
+    m = RobertaForCausalLM(vocab_size=128, hidden_size=64, hidden_layers=2,
+                           num_attention_heads=4, intermediate_size=128,
+                           max_position_embeddings=66)
+    ids = torch.randint(2, 128, (2, 16))
+    out = m(input_ids=ids, labels=ids)
+    out["lm_loss"].backward()
+    assert torch.isfinite(out["lm_loss"])
+    m.eval()
+    with torch.no_grad():
+        a = m(input_ids=ids)["prediction_scores"]
+        ids2 = ids.clone()
+        ids2[:, -1] = 3
+        b = m(input_ids=ids2)["prediction_scores"]
+    assert torch.allclose(a[:, :-1], b[:, :-1], atol=1e-5)
