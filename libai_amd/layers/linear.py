@@ -29,7 +29,33 @@ from ..utils import distributed as du
 __all__ = ["Linear1D", "Linear"]
 
 
-def init_tp_shard_(param, full_shape, init_method, shard_dim):
+def tp_slice(full, tp, tpr, dim, fused_chunks=1):
+    """This rank's TP shard of a full tensor.
+
+    ``fused_chunks > 1`` handles fused projections like [gate | up]: the
+    canonical full layout is the concatenation of the chunks, but each rank
+    must hold a PAIRED slice ([gate_r | up_r]) -- a plain contiguous chunk
+    would give rank 0 all of gate and rank 1 all of up, silently changing
+    the function the gated MLP computes under TP.
+    """
+    if fused_chunks <= 1:
+        return full.chunk(tp, dim=dim)[tpr]
+    parts = full.chunk(fused_chunks, dim=dim)
+    return torch.cat([h.chunk(tp, dim=dim)[tpr] for h in parts], dim=dim)
+
+
+def tp_merge(shards, dim, fused_chunks=1):
+    """Inverse of tp_slice: per-rank shards -> canonical full tensor."""
+    if fused_chunks <= 1:
+        return torch.cat(shards, dim=dim)
+    per = [s.chunk(fused_chunks, dim=dim) for s in shards]
+    return torch.cat(
+        [torch.cat([p[h] for p in per], dim=dim) for h in range(fused_chunks)],
+        dim=dim,
+    )
+
+
+def init_tp_shard_(param, full_shape, init_method, shard_dim, fused_chunks=1):
     """Initialize a TP-sharded parameter as a SLICE of the full-tensor init.
 
     All TP ranks draw the same full tensor (they run init under the same seed)
@@ -44,7 +70,7 @@ def init_tp_shard_(param, full_shape, init_method, shard_dim):
         with torch.no_grad():
             param.copy_(full)
         return
-    shard = full.chunk(tp, dim=shard_dim)[tpr]
+    shard = tp_slice(full, tp, tpr, shard_dim, fused_chunks)
     with torch.no_grad():
         param.copy_(shard)
 
@@ -58,6 +84,7 @@ class Linear1D(nn.Module):
         parallel="data",
         init_method=nn.init.xavier_normal_,
         skip_bias_add=False,
+        fused_chunks=1,
         *,
         layer_idx=0,
         dtype=None,
@@ -81,11 +108,14 @@ class Linear1D(nn.Module):
             )
             self.weight.tensor_parallel = True
             self.weight.tp_shard_dim = 0
-            init_tp_shard_(self.weight, (out_features, in_features), init_method, 0)
+            self.weight.tp_fused_chunks = fused_chunks
+            init_tp_shard_(self.weight, (out_features, in_features), init_method, 0,
+                           fused_chunks)
             if bias:
                 self.bias = nn.Parameter(torch.zeros(out_features // tp, dtype=dtype))
                 self.bias.tensor_parallel = True
                 self.bias.tp_shard_dim = 0
+                self.bias.tp_fused_chunks = fused_chunks
             else:
                 self.register_parameter("bias", None)
         elif parallel == "row":

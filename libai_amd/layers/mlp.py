@@ -77,7 +77,7 @@ class GatedMLP(nn.Module):
         output_layer_init_method = output_layer_init_method or init_method
         self.gate_up_proj = Linear1D(
             hidden_size, 2 * ffn_hidden_size, bias=False, parallel="col",
-            init_method=init_method, layer_idx=layer_idx,
+            init_method=init_method, fused_chunks=2, layer_idx=layer_idx,
         )
         self.down_proj = Linear1D(
             ffn_hidden_size, hidden_size, bias=False, parallel="row",

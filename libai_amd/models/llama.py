@@ -97,7 +97,7 @@ class LlamaMLP(nn.Module):
         super().__init__()
         self.gate_up_proj = Linear1D(hidden_size, 2 * intermediate_size, bias=False,
                                      parallel="col", init_method=init_method,
-                                     layer_idx=layer_idx)
+                                     fused_chunks=2, layer_idx=layer_idx)
         self.down_proj = Linear1D(intermediate_size, hidden_size, bias=False,
                                   parallel="row", init_method=output_init_method,
                                   skip_bias_add=True, layer_idx=layer_idx)

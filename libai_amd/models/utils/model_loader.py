@@ -32,7 +32,10 @@ def _shard_and_load(model, full_state, strict=False):
         t = full_state[name]
         p = params.get(name)
         if p is not None and getattr(p, "tensor_parallel", False) and tp > 1:
-            t = t.chunk(tp, dim=getattr(p, "tp_shard_dim", 0))[tpr]
+            from ...layers.linear import tp_slice
+
+            t = tp_slice(t, tp, tpr, getattr(p, "tp_shard_dim", 0),
+                         getattr(p, "tp_fused_chunks", 1))
         if tuple(t.shape) != tuple(cur.shape):
             logger.warning(
                 f"loader: {name} shape {tuple(t.shape)} != model {tuple(cur.shape)}"

@@ -34,10 +34,13 @@ def _consolidate_model_state(model):
     for name, t in model.state_dict().items():
         p = params.get(name)
         if p is not None and getattr(p, "tensor_parallel", False) and tp > 1:
+            from ..layers.linear import tp_merge
+
             dim = getattr(p, "tp_shard_dim", 0)
             shards = [torch.empty_like(t) for _ in range(tp)]
             dist.all_gather(shards, t.contiguous(), group=dutil.tensor_parallel_group)
-            out[name] = torch.cat(shards, dim=dim).cpu()
+            out[name] = tp_merge(shards, dim,
+                                 getattr(p, "tp_fused_chunks", 1)).cpu()
         else:
             out[name] = t.detach().cpu()
     return out
@@ -55,8 +58,10 @@ def _shard_for_load(model, full_state):
         full = full_state[name]
         p = params.get(name)
         if p is not None and getattr(p, "tensor_parallel", False) and tp > 1:
+            from ..layers.linear import tp_slice
+
             dim = getattr(p, "tp_shard_dim", 0)
-            full = full.chunk(tp, dim=dim)[tpr]
+            full = tp_slice(full, tp, tpr, dim, getattr(p, "tp_fused_chunks", 1))
         if tuple(full.shape) != tuple(cur.shape):
             logging.getLogger(__name__).warning(
                 f"checkpoint key {name}: shape {tuple(full.shape)} != model "

@@ -156,3 +156,45 @@ def _tp2_transformer_worker(rank, world):
 
 def test_tp2_transformer_layer_equivalence():
     assert all(run_dist(_tp2_transformer_worker, 2))
+
+
+def _tp2_gated_mlp_worker(rank, world):
+    import torch
+    import torch.distributed as dist
+
+    from libai_amd.layers.linear import tp_merge
+    from libai_amd.layers.mlp import GatedMLP
+    from libai_amd.utils import distributed as du
+
+    du.setup_dist_util({"tensor_parallel_size": 2})
+    dutil = du.get_dist_util()
+    torch.manual_seed(42)
+    sharded = GatedMLP(16, 32, activation="silu")
+
+    # reconstruct the CANONICAL full weights from the shards and compute the
+    # gated MLP in plain torch: the [gate|up] projection must shard as PAIRED
+    # slices ([gate_r | up_r]) for the local swiglu halves to mean the same
+    # thing the canonical [gate | up] layout does
+    def gather_full(p, fused):
+        shards = [torch.empty_like(p.data) for _ in range(2)]
+        dist.all_gather(shards, p.data.contiguous(),
+                        group=dutil.tensor_parallel_group)
+        return tp_merge(shards, p.tp_shard_dim, fused)
+
+    gu_full = gather_full(sharded.gate_up_proj.weight,
+                          getattr(sharded.gate_up_proj.weight,
+                                  "tp_fused_chunks", 1))
+    w2_full = gather_full(sharded.down_proj.weight, 1)
+
+    torch.manual_seed(7)
+    x = torch.randn(4, 16)
+    out_s = sharded(x)
+    g, u = gu_full.chunk(2, dim=0)
+    ref = (torch.nn.functional.silu(x @ g.t()) * (x @ u.t())) @ w2_full.t()
+    assert torch.allclose(out_s, ref, atol=1e-5), (
+        f"gated MLP sharded != canonical: {(out_s - ref).abs().max()}"
+    )
+
+
+def test_tp2_gated_mlp_equivalence():
+    run_dist(_tp2_gated_mlp_worker, world_size=2)
