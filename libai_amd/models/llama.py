@@ -31,7 +31,11 @@ class LlamaAttention(nn.Module):
         self.num_heads = num_heads
         self.head_dim = hidden_size // num_heads
         dutil = du.get_dist_util()
-        self.num_heads_local = num_heads // dutil.tensor_parallel_size
+        tp = dutil.tensor_parallel_size
+        # the fused qkv weight is per-head interleaved; whole heads must land
+        # on each TP rank for the contiguous column shard to stay coherent
+        assert num_heads % tp == 0, (num_heads, tp)
+        self.num_heads_local = num_heads // tp
         self.max_pos = max_position_embeddings
         self.rope_theta = rope_theta
         self.layer_idx = layer_idx
