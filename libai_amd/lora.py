@@ -48,6 +48,9 @@ class LoRALinear(nn.Module):
         if base.parallel == "col":
             self.lora_B.tensor_parallel = True
             self.lora_B.tp_shard_dim = 0
+            # inherit fused-projection pairing (e.g. [gate|up]) so checkpoint
+            # consolidation/reshard keeps the canonical layout
+            self.lora_B.tp_fused_chunks = getattr(base.weight, "tp_fused_chunks", 1)
         elif base.parallel == "row":
             self.lora_A.tensor_parallel = True
             self.lora_A.tp_shard_dim = 1
