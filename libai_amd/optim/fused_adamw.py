@@ -390,6 +390,42 @@ class FusedAdamW(torch.optim.Optimizer):
             outs.append(full)
         return outs
 
+    @staticmethod
+    def _tp_gather_param_state(p, flat):
+        """TP-sharded param state -> canonical full tensor (flattened).
+
+        Makes the optimizer checkpoint TP-topology-independent like the
+        model weights (reference: OneFlow global-tensor save)."""
+        if not (getattr(p, "tensor_parallel", False) and dist.is_initialized()):
+            return flat.clone(), list(p.shape)
+        dutil = du.get_dist_util()
+        tp = dutil.tensor_parallel_size
+        if tp == 1:
+            return flat.clone(), list(p.shape)
+        from ..layers.linear import tp_merge
+
+        local = flat.view(p.shape)
+        shards = [torch.empty_like(local) for _ in range(tp)]
+        dist.all_gather(shards, local.contiguous(),
+                        group=dutil.tensor_parallel_group)
+        full = tp_merge(shards, getattr(p, "tp_shard_dim", 0),
+                        getattr(p, "tp_fused_chunks", 1))
+        return full.reshape(-1), list(full.shape)
+
+    @staticmethod
+    def _tp_slice_param_state(p, full_flat, full_shape):
+        if not (getattr(p, "tensor_parallel", False) and dist.is_initialized()):
+            return full_flat
+        dutil = du.get_dist_util()
+        tp, tpr = dutil.tensor_parallel_size, dutil.tensor_parallel_rank
+        if tp == 1 or list(full_shape) == list(p.shape):
+            return full_flat
+        from ..layers.linear import tp_slice
+
+        full = full_flat.view(full_shape)
+        return tp_slice(full, tp, tpr, getattr(p, "tp_shard_dim", 0),
+                        getattr(p, "tp_fused_chunks", 1)).reshape(-1)
+
     def state_dict(self):
         per_param = []
         for gi, b in self.buckets:
@@ -397,12 +433,15 @@ class FusedAdamW(torch.optim.Optimizer):
             off = 0
             for p in b.params:
                 n = p.numel()
+                mast, full_shape = self._tp_gather_param_state(p, master[off:off + n])
+                ea, _ = self._tp_gather_param_state(p, m[off:off + n])
+                es, _ = self._tp_gather_param_state(p, v[off:off + n])
                 per_param.append(
                     {
-                        "master": master[off : off + n].clone(),
-                        "exp_avg": m[off : off + n].clone(),
-                        "exp_avg_sq": v[off : off + n].clone(),
-                        "shape": list(p.shape),
+                        "master": mast,
+                        "exp_avg": ea,
+                        "exp_avg_sq": es,
+                        "shape": full_shape,
                     }
                 )
                 off += n
@@ -428,9 +467,13 @@ class FusedAdamW(torch.optim.Optimizer):
             for p in b.params:
                 n = p.numel()
                 entry = state_dict["per_param"][idx]
-                master[off : off + n].copy_(entry["master"])
-                m[off : off + n].copy_(entry["exp_avg"])
-                v[off : off + n].copy_(entry["exp_avg_sq"])
+                shp = entry.get("shape", list(p.shape))
+                master[off : off + n].copy_(
+                    self._tp_slice_param_state(p, entry["master"], shp))
+                m[off : off + n].copy_(
+                    self._tp_slice_param_state(p, entry["exp_avg"], shp))
+                v[off : off + n].copy_(
+                    self._tp_slice_param_state(p, entry["exp_avg_sq"], shp))
                 idx += 1
                 off += n
             sl = slice(b.shard_off, b.shard_off + b.shard) if self._zero_eff > 0 \
