@@ -152,6 +152,8 @@ class DefaultTrainer(TrainerBase):
         # optimizer + scheduler (after model is on its final device: the
         # flat-bucket optimizer freezes param storage)
         self.optimizer = self.build_optimizer(cfg, self.model)
+        if hasattr(self.optimizer, "set_param_names"):
+            self.optimizer.set_param_names(self.model.named_parameters())
         self.lr_scheduler = self.build_lr_scheduler(cfg, self.optimizer)
 
         # diverge the RNG stream per DP rank now that init is done (dropout

@@ -133,6 +133,7 @@ class FusedAdamW(torch.optim.Optimizer):
 
     def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
                  weight_decay=0.01, clip_grad=0.0, zero_stage=0):
+        self._param_names = {}
         defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
         super().__init__(params, defaults)
         self.clip_grad = clip_grad
@@ -390,6 +391,13 @@ class FusedAdamW(torch.optim.Optimizer):
             outs.append(full)
         return outs
 
+    def set_param_names(self, named_parameters):
+        """Register param names so optimizer checkpoints match by NAME:
+        resuming on a different pipeline split (different local param list)
+        then restores state for every overlapping parameter instead of
+        relying on positional order."""
+        self._param_names = {id(p): n for n, p in named_parameters}
+
     @staticmethod
     def _tp_gather_param_state(p, flat):
         """TP-sharded param state -> canonical full tensor (flattened).
@@ -442,6 +450,7 @@ class FusedAdamW(torch.optim.Optimizer):
                         "exp_avg": ea,
                         "exp_avg_sq": es,
                         "shape": full_shape,
+                        "name": self._param_names.get(id(p)),
                     }
                 )
                 off += n
@@ -457,6 +466,9 @@ class FusedAdamW(torch.optim.Optimizer):
         self._step = state_dict["step"]
         for g, saved in zip(self.param_groups, state_dict["param_groups"]):
             g.update(saved)
+        by_name = {
+            e["name"]: e for e in state_dict["per_param"] if e.get("name")
+        }
         idx = 0
         for gi, b in self.buckets:
             device = b.flat_param.device
@@ -466,7 +478,15 @@ class FusedAdamW(torch.optim.Optimizer):
             off = 0
             for p in b.params:
                 n = p.numel()
-                entry = state_dict["per_param"][idx]
+                name = self._param_names.get(id(p))
+                if name is not None and name in by_name:
+                    entry = by_name[name]
+                elif idx < len(state_dict["per_param"]):
+                    entry = state_dict["per_param"][idx]
+                else:
+                    idx += 1
+                    off += n
+                    continue
                 shp = entry.get("shape", list(p.shape))
                 master[off : off + n].copy_(
                     self._tp_slice_param_state(p, entry["master"], shp))
