@@ -107,3 +107,22 @@ def test_lora_gpu_step():
         out["lm_loss"].backward()
         opt.step()
     assert torch.isfinite(out["lm_loss"])
+
+
+def test_t5_gated_gpu_bf16():
+    from libai_amd.models import T5ForPreTraining
+
+    torch.manual_seed(0)
+    m = T5ForPreTraining(
+        vocab_size=1024, hidden_size=256, hidden_layers=2, num_attention_heads=4,
+        intermediate_size=512, max_position_embeddings=128, mlp_type="gated",
+        activation="silu",
+    ).to(torch.bfloat16).cuda()
+    b = 2
+    _step(m, dict(
+        encoder_input_ids=torch.randint(0, 1024, (b, 64), device="cuda"),
+        decoder_input_ids=torch.randint(0, 1024, (b, 32), device="cuda"),
+        encoder_attn_mask=torch.ones(b, 64, dtype=torch.uint8, device="cuda"),
+        lm_labels=torch.randint(0, 1024, (b, 32), device="cuda"),
+        loss_mask=torch.ones(b, 32, dtype=torch.long, device="cuda"),
+    ))
