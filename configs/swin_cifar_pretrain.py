@@ -6,6 +6,7 @@ Reference recipe: configs/swin_cifar100.py (SwinTransformer, mixup, AdamW).
 from libai_amd.config import LazyCall
 from libai_amd.data import build_image_train_loader
 from libai_amd.data.datasets import SyntheticImageDataset
+from libai_amd.data.mixup import Mixup
 from libai_amd.models import SwinTransformer
 
 from .common.optim import optim
@@ -28,6 +29,8 @@ dataloader = dict(
                                                 num_classes=100),
         train_batch_size=64,
         num_workers=2,
+        mixup_func=LazyCall(Mixup)(mixup_alpha=0.8, cutmix_alpha=1.0,
+                                   label_smoothing=0.1, num_classes=100),
     ),
 )
 

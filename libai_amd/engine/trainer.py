@@ -160,10 +160,17 @@ class EagerTrainer(TrainerBase):
         if isinstance(data, Instance):
             data = data.to_dict()
         device = du.get_device()
-        return {
+        out = {
             k: (v.to(device, non_blocking=True) if torch.is_tensor(v) else v)
             for k, v in data.items()
         }
+        # CV mixup/cutmix, applied at batch time like the reference
+        # (engine/default.py:509-515); produces soft labels which the default
+        # nn.CrossEntropyLoss consumes directly
+        mixup = getattr(self.data_loader, "mixup_func", None)
+        if mixup is not None and "images" in out and "labels" in out:
+            out["images"], out["labels"] = mixup(out["images"], out["labels"])
+        return out
 
     def _sync_dp_grads(self):
         if hasattr(self.optimizer, "grad_sync"):

@@ -71,3 +71,40 @@ def test_mixup_soft_labels():
     # soft labels sum to 1 and reflect smoothing
     assert torch.allclose(out_labels.sum(1), torch.ones(8), atol=1e-5)
     assert (out_labels > 0).all()
+
+
+def test_trainer_applies_mixup():
+    """EagerTrainer.get_batch applies the loader's mixup_func (soft labels)."""
+    from torch import nn
+
+    from libai_amd.engine.trainer import EagerTrainer
+    from libai_amd.optim import FusedAdamW
+    from libai_amd.utils import distributed as du
+
+    du.setup_dist_util({})
+
+    class _CV(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.head = nn.Linear(3 * 8 * 8, 10)
+            self.loss = nn.CrossEntropyLoss()
+
+        def forward(self, images, labels=None):
+            logits = self.head(images.flatten(1))
+            return {"losses": self.loss(logits, labels)}
+
+    class _Loader:
+        mixup_func = Mixup(num_classes=10, prob=1.0)
+
+        def __iter__(self):
+            while True:
+                yield {"images": torch.randn(4, 3, 8, 8),
+                       "labels": torch.randint(0, 10, (4,))}
+
+    model = _CV()
+    opt = FusedAdamW(model.parameters(), lr=1e-2)
+    tr = EagerTrainer(model, _Loader(), opt, grad_acc_steps=1)
+    batch = tr.get_batch({"images": torch.randn(4, 3, 8, 8),
+                          "labels": torch.randint(0, 10, (4,))})
+    assert batch["labels"].shape == (4, 10)  # soft labels
+    tr.train(0, 2)  # steps run with soft-label CE
