@@ -26,20 +26,26 @@ class Mixup:
         return lam * y + (1 - lam) * y[perm]
 
     def __call__(self, images, labels):
-        if np.random.rand() > self.prob:
+        # All randomness comes from torch's generator: the trainer seeds
+        # torch identically within each TP group (same_seed_for_tp_group),
+        # so TP ranks apply the SAME mix to the replicated batch.  np.random
+        # is seeded per-global-rank and would desync TP activations.
+        if float(torch.rand(())) > self.prob:
             return images, self._one_hot(labels, 1.0, torch.arange(len(labels)))
         perm = torch.randperm(images.size(0), device=images.device)
-        use_cutmix = np.random.rand() < self.switch_prob and self.cutmix_alpha > 0
+        use_cutmix = float(torch.rand(())) < self.switch_prob and self.cutmix_alpha > 0
         if use_cutmix:
-            lam = float(np.random.beta(self.cutmix_alpha, self.cutmix_alpha))
+            lam = float(torch.distributions.Beta(
+                self.cutmix_alpha, self.cutmix_alpha).sample())
             H, W = images.shape[-2:]
             rh, rw = int(H * np.sqrt(1 - lam)), int(W * np.sqrt(1 - lam))
-            cy, cx = np.random.randint(H), np.random.randint(W)
+            cy, cx = int(torch.randint(H, ())), int(torch.randint(W, ()))
             y1, y2 = max(cy - rh // 2, 0), min(cy + rh // 2, H)
             x1, x2 = max(cx - rw // 2, 0), min(cx + rw // 2, W)
             images[..., y1:y2, x1:x2] = images[perm][..., y1:y2, x1:x2]
             lam = 1.0 - (y2 - y1) * (x2 - x1) / (H * W)
         else:
-            lam = float(np.random.beta(self.mixup_alpha, self.mixup_alpha))
+            lam = float(torch.distributions.Beta(
+                self.mixup_alpha, self.mixup_alpha).sample())
             images = lam * images + (1 - lam) * images[perm]
         return images, self._one_hot(labels, lam, perm)

@@ -8,8 +8,10 @@ else; ``merge_lora`` folds the update back into the base weight.
 
 TP note: for a column-parallel base the B factor is sharded on its output
 dim (A replicated); for a row-parallel base the A factor is sharded on its
-input dim (B replicated) — matching the base layer's shard geometry so the
-LoRA path needs no extra collectives beyond the base layer's own.
+input dim (B replicated) — matching the base layer's shard geometry.  The
+rank-r bottleneck needs one extra TP collective on the [.., r] activation:
+all-reduce forward for the row case (partial input-dim sum) and all-reduce
+backward for the col case (partial grads toward replicated A).
 """
 
 import math
@@ -19,6 +21,10 @@ import torch
 from torch import nn
 
 from .layers.linear import Linear1D
+from .parallel.comm import (
+    copy_to_tensor_parallel_region,
+    reduce_from_tensor_parallel_region,
+)
 from .utils import distributed as du
 
 __all__ = ["LoRALinear", "apply_lora", "merge_lora", "mark_only_lora_as_trainable"]
@@ -61,7 +67,18 @@ class LoRALinear(nn.Module):
 
     def forward(self, x):
         out = self.base(x)
-        delta = self.lora_dropout(x) @ self.lora_A.t() @ self.lora_B.t() * self.scaling
+        h = self.lora_dropout(x) @ self.lora_A.t()
+        if self.base.parallel == "row":
+            # A is sharded on the input dim, so h is a PARTIAL sum over TP:
+            # all-reduce in forward (identity backward) before the replicated
+            # B projection — mirrors the base row-linear's C1 collective.
+            h = reduce_from_tensor_parallel_region(h)
+        elif self.base.parallel == "col":
+            # h is replicated but the grad arriving from the sharded-B branch
+            # is partial per rank; identity-fwd/all-reduce-bwd here keeps the
+            # replicated lora_A (and x) gradients coherent across TP ranks.
+            h = copy_to_tensor_parallel_region(h)
+        delta = h @ self.lora_B.t() * self.scaling
         if isinstance(out, tuple):  # skip_bias_add base
             return out[0] + delta, out[1]
         return out + delta

@@ -23,6 +23,8 @@ libai/optim/build.py:86-126):
     fused update — one pass over grads total.
 """
 
+import logging
+
 import torch
 import torch.distributed as dist
 
@@ -178,6 +180,22 @@ class FusedAdamW(torch.optim.Optimizer):
     def zero_grad(self, set_to_none=False):
         for _, b in self.buckets:
             b.flat_grad.zero_()
+
+    @torch.no_grad()
+    def resync_masters(self):
+        """Refresh the fp32 masters from the (possibly just-loaded) bf16
+        params.  Buckets alias ``p.data`` into ``flat_param``, so a
+        weights-only checkpoint load updates ``flat_param`` but not
+        ``flat_master`` — without this, the first step() would rewrite the
+        loaded weights from the stale random-init masters.  The Checkpointer
+        calls this after every model-weight load; a subsequent optimizer
+        state load simply overwrites the masters again."""
+        if self._buckets is None:
+            return  # not built yet: masters will be created from current params
+        for _, b in self._buckets:
+            if b.dtype != torch.float32:
+                b.flat_master.copy_(b._upd_param().float())
+            # fp32 masters alias flat_param (non-ZeRO) or its owned slice
 
     # -- DP gradient communication (all-reduce / ZeRO reduce-scatter) --------
 
@@ -466,13 +484,20 @@ class FusedAdamW(torch.optim.Optimizer):
         self._step = state_dict["step"]
         for g, saved in zip(self.param_groups, state_dict["param_groups"]):
             g.update(saved)
-        by_name = {
-            e["name"]: e for e in state_dict["per_param"] if e.get("name")
-        }
+        saved = state_dict["per_param"]
+        by_name = {e["name"]: e for e in saved if e.get("name")}
+        # Trust positional order only for legacy checkpoints without names:
+        # with names present, per_param[idx] can belong to a DIFFERENT param
+        # (pipeline-resplit resume changes the local param list), so unmatched
+        # params keep fresh state instead of loading a stranger's.
+        positional_ok = not by_name
+        logger = logging.getLogger(__name__)
         idx = 0
         for gi, b in self.buckets:
             device = b.flat_param.device
-            master = torch.zeros(b.numel, dtype=torch.float32, device=device)
+            # start masters from CURRENT weights so unmatched params keep
+            # their values (not zeros) when flat_param is rewritten below
+            master = b.flat_param.detach().float().clone()
             m = torch.zeros_like(master)
             v = torch.zeros_like(master)
             off = 0
@@ -481,9 +506,13 @@ class FusedAdamW(torch.optim.Optimizer):
                 name = self._param_names.get(id(p))
                 if name is not None and name in by_name:
                     entry = by_name[name]
-                elif idx < len(state_dict["per_param"]):
-                    entry = state_dict["per_param"][idx]
+                elif positional_ok and idx < len(saved):
+                    entry = saved[idx]
                 else:
+                    logger.warning(
+                        f"optimizer checkpoint has no state for {name!r}; "
+                        "keeping fresh state"
+                    )
                     idx += 1
                     off += n
                     continue

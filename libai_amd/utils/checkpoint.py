@@ -119,14 +119,18 @@ class Checkpointer:
             torch.save(model_state, os.path.join(ckpt_dir, "model.pt"))
 
         # other checkpointables: per-rank topology-tagged states for PP/TP-local
-        # objects (optimizer/scheduler); rank-local shard files
+        # objects (optimizer/scheduler); rank-local shard files.
+        # state_dict() may be COLLECTIVE (FusedAdamW gathers ZeRO shards over
+        # DP and TP shards over TP) — every rank must call it; only the
+        # writer rank saves the result.
         for key, obj in self.checkpointables.items():
             if not hasattr(obj, "state_dict"):
                 continue
+            state = obj.state_dict()
             if key == "optimizer" and (pp > 1 or dutil.tensor_parallel_size > 1):
                 if dutil.data_parallel_rank == 0:
                     torch.save(
-                        obj.state_dict(),
+                        state,
                         os.path.join(
                             ckpt_dir,
                             f"{key}_tp{dutil.tensor_parallel_rank}"
@@ -134,7 +138,7 @@ class Checkpointer:
                         ),
                     )
             elif du.is_main_process():
-                torch.save(obj.state_dict(), os.path.join(ckpt_dir, f"{key}.pt"))
+                torch.save(state, os.path.join(ckpt_dir, f"{key}.pt"))
 
         if du.is_main_process():
             extra = dict(kwargs)
@@ -157,6 +161,13 @@ class Checkpointer:
         missing = [m for m in missing if m in dict(self.model.named_parameters())]
         if missing:
             self.logger.warning(f"missing keys in checkpoint: {missing[:10]}...")
+
+        # The flat-bucket optimizer aliases p.data — after rewriting the
+        # weights its fp32 masters are stale; refresh them (a subsequent
+        # optimizer state load below simply overwrites them again).
+        for obj in self.checkpointables.values():
+            if hasattr(obj, "resync_masters"):
+                obj.resync_masters()
 
         keys = (
             self.checkpointables.keys() if checkpointables is None else checkpointables

@@ -89,3 +89,42 @@ def test_overlap_grad_sync_matches_plain():
     assert torch.allclose(overlapped, plain, atol=1e-6), (
         (overlapped - plain).abs().max()
     )
+
+
+def _ckpt_save_worker(rank, world, tmpdir):
+    import os
+
+    import torch
+
+    from libai_amd.optim import FusedAdamW
+    from libai_amd.utils import distributed as du
+    from libai_amd.utils.checkpoint import Checkpointer
+
+    du.setup_dist_util({})
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(torch.nn.Linear(16, 33), torch.nn.Linear(33, 8))
+    opt = FusedAdamW(model.parameters(), lr=1e-2, zero_stage=1)
+    opt.set_param_names(model.named_parameters())
+    torch.manual_seed(100 + rank)
+    opt.zero_grad()
+    model(torch.randn(4, 16)).pow(2).mean().backward()
+    opt.grad_sync()
+    opt.step()
+    # Regression: save() calls optimizer.state_dict() which
+    # all-gathers the ZeRO shards over DP — it must run on EVERY rank, or
+    # the dp0 writer deadlocks here.
+    ck = Checkpointer(model, tmpdir, optimizer=opt)
+    ck.save("step1")
+    assert os.path.exists(os.path.join(tmpdir, "step1", "optimizer.pt"))
+    # and the saved state must round-trip
+    opt2 = FusedAdamW(model.parameters(), lr=1e-2, zero_stage=1)
+    opt2.set_param_names(model.named_parameters())
+    ck2 = Checkpointer(model, tmpdir, optimizer=opt2)
+    ck2.load(os.path.join(tmpdir, "step1"))
+    assert opt2._step == opt._step
+    return True
+
+
+@pytest.mark.timeout(300)
+def test_zero_checkpoint_save_no_deadlock(tmp_path):
+    run_dist(_ckpt_save_worker, 2, args=(str(tmp_path),))

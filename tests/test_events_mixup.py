@@ -108,3 +108,25 @@ def test_trainer_applies_mixup():
                           "labels": torch.randint(0, 10, (4,))})
     assert batch["labels"].shape == (4, 10)  # soft labels
     tr.train(0, 2)  # steps run with soft-label CE
+
+
+def test_mixup_randomness_is_torch_seeded():
+    """TP ranks share the torch seed (same_seed_for_tp_group) but have
+    different numpy seeds — mixup must depend only on the torch stream so
+    the replicated batch is mixed identically across a TP group."""
+    import numpy as np
+    import torch
+
+    from libai_amd.data.mixup import Mixup
+
+    imgs = torch.randn(8, 3, 16, 16)
+    labels = torch.randint(0, 10, (8,))
+    outs = []
+    for np_seed in (1, 999):  # simulated per-rank numpy seeds
+        np.random.seed(np_seed)
+        torch.manual_seed(5)
+        mix = Mixup(prob=1.0, switch_prob=0.5, num_classes=10)
+        x, y = mix(imgs.clone(), labels.clone())
+        outs.append((x, y))
+    assert torch.equal(outs[0][0], outs[1][0])
+    assert torch.equal(outs[0][1], outs[1][1])
