@@ -154,6 +154,10 @@ class DefaultTrainer(TrainerBase):
         self.optimizer = self.build_optimizer(cfg, self.model)
         if hasattr(self.optimizer, "set_param_names"):
             self.optimizer.set_param_names(self.model.named_parameters())
+        if getattr(self.optimizer, "zero_stage", 0) == 3:
+            from ..parallel.zero import setup_zero3
+
+            self._zero3_manager = setup_zero3(self.model, self.optimizer)
         self.lr_scheduler = self.build_lr_scheduler(cfg, self.optimizer)
 
         # diverge the RNG stream per DP rank now that init is done (dropout

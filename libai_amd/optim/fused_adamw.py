@@ -40,12 +40,13 @@ def _pad_to(n, mult):
 
 class _Bucket:
     def __init__(self, params, dtype, device, weight_decay_on, tp_sharded, dp_size,
-                 dp_rank, zero_stage):
+                 dp_rank, zero_stage, unit=-1):
         self.params = params
         self.dtype = dtype
         self.weight_decay_on = weight_decay_on
         self.tp_sharded = tp_sharded
         self.zero = zero_stage
+        self.unit = unit  # ZeRO-3 gather/release granularity (module unit id)
         self.dp_size = dp_size
         self.dp_rank = dp_rank
         raw = sum(p.numel() for p in params)
@@ -67,11 +68,21 @@ class _Bucket:
 
         # optimizer state exists only for the local shard under ZeRO
         owned = self.flat_param[self.shard_off : self.shard_off + self.shard]
-        self.flat_master = (
-            owned.float() if dtype != torch.float32 else owned
-        ) if zero_stage > 0 else (
-            self.flat_param.float() if dtype != torch.float32 else self.flat_param
-        )
+        if zero_stage == 3:
+            # persistent shard storage; the full flat buffers are released
+            # between materializations (finalize_zero3 after the DP broadcast)
+            self.shard_param = owned.detach().clone()
+            self.shard_grad = torch.zeros(self.shard, dtype=dtype, device=device)
+            self.flat_master = (
+                self.shard_param.float() if dtype != torch.float32
+                else self.shard_param
+            )
+        elif zero_stage > 0:
+            self.flat_master = owned.float() if dtype != torch.float32 else owned
+        else:
+            self.flat_master = (
+                self.flat_param.float() if dtype != torch.float32 else self.flat_param
+            )
         state_n = self.shard if zero_stage > 0 else self.numel
         self.exp_avg = torch.zeros(state_n, dtype=torch.float32, device=device)
         self.exp_avg_sq = torch.zeros(state_n, dtype=torch.float32, device=device)
@@ -80,12 +91,75 @@ class _Bucket:
 
     # region updated by the fused kernel (local shard under ZeRO)
     def _upd_param(self):
+        if self.zero == 3:
+            return self.shard_param
         return self.flat_param[self.shard_off : self.shard_off + self.shard] \
             if self.zero > 0 else self.flat_param
 
     def _upd_grad(self):
+        if self.zero == 3:
+            return self.shard_grad
         return self.flat_grad[self.shard_off : self.shard_off + self.shard] \
             if self.zero > 0 else self.flat_grad
+
+    # -- ZeRO-3 param/grad lifecycle ----------------------------------------
+    # flat_param/flat_grad keep their Storage objects forever (p.data/p.grad
+    # are views into them) but the backing memory is freed between uses via
+    # untyped_storage().resize_(0) and re-allocated on materialize — the
+    # FSDP idiom, sized here for per-transformer-layer units.
+
+    def params_live(self):
+        return self.flat_param.untyped_storage().size() > 0
+
+    def grads_live(self):
+        return self.flat_grad.untyped_storage().size() > 0
+
+    def finalize_zero3(self):
+        """After flatten + DP broadcast: snapshot the owned shard and free
+        the full buffers (the first forward re-gathers them)."""
+        self.shard_param.copy_(
+            self.flat_param[self.shard_off : self.shard_off + self.shard])
+        if self.dtype != torch.float32:
+            self.flat_master.copy_(self.shard_param.float())
+        self.release_params()
+        self.flat_grad.untyped_storage().resize_(0)
+
+    def materialize_params(self, group):
+        if self.params_live():
+            return
+        esz = self.flat_param.element_size()
+        self.flat_param.untyped_storage().resize_(self.numel * esz)
+        if dist.is_initialized() and self.dp_size > 1:
+            dist.all_gather_into_tensor(self.flat_param, self.shard_param,
+                                        group=group)
+        else:
+            self.flat_param[self.shard_off : self.shard_off + self.shard].copy_(
+                self.shard_param)
+
+    def release_params(self):
+        self.flat_param.untyped_storage().resize_(0)
+
+    def materialize_grads(self):
+        """(Re-)allocate the full grad buffer zeroed, and re-arm the
+        pending-param set for this backward."""
+        if not self.grads_live():
+            esz = self.flat_grad.element_size()
+            self.flat_grad.untyped_storage().resize_(self.numel * esz)
+            self.flat_grad.zero_()
+        self.pending = {id(p) for p in self.params}
+
+    def reduce_release_grads(self, group, dp):
+        """Average-reduce-scatter the full grads into the owned shard
+        (accumulating across micro-batches) and free the full buffer."""
+        self.flat_grad.div_(dp)
+        if dist.is_initialized() and self.dp_size > 1:
+            tmp = torch.empty_like(self.shard_grad)
+            dist.reduce_scatter_tensor(tmp, self.flat_grad, group=group)
+            self.shard_grad.add_(tmp)
+        else:
+            self.shard_grad.add_(
+                self.flat_grad[self.shard_off : self.shard_off + self.shard])
+        self.flat_grad.untyped_storage().resize_(0)
 
     def adam_desc(self, chunk):
         if self._adam_desc is None:
@@ -156,13 +230,16 @@ class FusedAdamW(torch.optim.Optimizer):
             for p in group["params"]:
                 if not p.requires_grad:
                     continue
-                key = (p.dtype, p.device, _is_tp_sharded(p))
+                # stage 3 additionally splits buckets by module UNIT so the
+                # gather/release lifecycle is per-transformer-layer
+                unit = getattr(p, "_zero3_unit", -1) if zero == 3 else -1
+                key = (p.dtype, p.device, _is_tp_sharded(p), unit)
                 by_key.setdefault(key, []).append(p)
             wd_on = group["weight_decay"] > 0
-            for (dtype, device, tp_sharded), plist in by_key.items():
+            for (dtype, device, tp_sharded, unit), plist in by_key.items():
                 self._buckets.append(
                     (gi, _Bucket(plist, dtype, device, wd_on, tp_sharded, dp, dpr,
-                                 zero))
+                                 zero, unit=unit))
                 )
         # DP replicas must start bit-identical; broadcast once from dp rank 0
         if dp > 1 and dist.is_initialized():
@@ -170,6 +247,9 @@ class FusedAdamW(torch.optim.Optimizer):
                 src_rank = dist.get_global_rank(dutil.data_parallel_group, 0)
                 dist.broadcast(b.flat_param, src=src_rank,
                                group=dutil.data_parallel_group)
+        if zero == 3:
+            for _, b in self._buckets:
+                b.finalize_zero3()
 
     @property
     def buckets(self):
@@ -179,7 +259,10 @@ class FusedAdamW(torch.optim.Optimizer):
 
     def zero_grad(self, set_to_none=False):
         for _, b in self.buckets:
-            b.flat_grad.zero_()
+            if b.zero == 3:
+                b.shard_grad.zero_()  # flat grads are zeroed on materialize
+            else:
+                b.flat_grad.zero_()
 
     @torch.no_grad()
     def resync_masters(self):
@@ -211,6 +294,16 @@ class FusedAdamW(torch.optim.Optimizer):
         group = dutil.data_parallel_group
         if getattr(self, "_overlap_active", False):
             self._finish_overlap(group, dp)
+            return
+        if self._zero_eff == 3:
+            # the backward hooks reduce-scatter each unit as its grads land;
+            # finish stragglers (buckets whose pending set never emptied)
+            # and release any still-materialized params
+            for _, b in self.buckets:
+                if b.grads_live():
+                    b.reduce_release_grads(group, dp)
+                if b.params_live():
+                    b.release_params()
             return
         for _, b in self.buckets:
             b.flat_grad.div_(dp)
@@ -316,12 +409,41 @@ class FusedAdamW(torch.optim.Optimizer):
 
     def _gather_params(self):
         dutil = du.get_dist_util()
-        if self._zero_eff == 0 or dutil.data_parallel_size == 1:
-            return
+        if self._zero_eff in (0, 3) or dutil.data_parallel_size == 1:
+            return  # stage 3 re-gathers lazily at the next forward
         group = dutil.data_parallel_group
         for _, b in self.buckets:
             dist.all_gather_into_tensor(b.flat_param, b._upd_param().contiguous(),
                                         group=group)
+
+    # -- ZeRO-3 whole-model materialization (checkpointing / export) --------
+
+    def materialize_all_params(self):
+        """Gather every stage-3 bucket's full params (p.data becomes readable
+        again).  No-op outside stage 3."""
+        if getattr(self, "_zero_eff", None) != 3:
+            return
+        group = du.get_dist_util().data_parallel_group
+        for _, b in self.buckets:
+            b.materialize_params(group)
+
+    def release_all_params(self):
+        if getattr(self, "_zero_eff", None) != 3:
+            return
+        for _, b in self.buckets:
+            b.release_params()
+
+    @torch.no_grad()
+    def refresh_shards_from_params(self):
+        """After an external write into p.data (checkpoint load), re-snapshot
+        the owned shards + masters from the materialized flat params."""
+        if getattr(self, "_zero_eff", None) != 3:
+            return
+        for _, b in self.buckets:
+            b.shard_param.copy_(
+                b.flat_param[b.shard_off : b.shard_off + b.shard])
+            if b.dtype != torch.float32:
+                b.flat_master.copy_(b.shard_param.float())
 
     # -- grad norm / clip ---------------------------------------------------
 
@@ -495,6 +617,8 @@ class FusedAdamW(torch.optim.Optimizer):
         idx = 0
         for gi, b in self.buckets:
             device = b.flat_param.device
+            if b.zero == 3:  # collective: all DP ranks load together
+                b.materialize_params(du.get_dist_util().data_parallel_group)
             # start masters from CURRENT weights so unmatched params keep
             # their values (not zeros) when flat_param is rewritten below
             master = b.flat_param.detach().float().clone()
@@ -530,7 +654,12 @@ class FusedAdamW(torch.optim.Optimizer):
             b.flat_master.copy_(master[sl])
             b.exp_avg.copy_(m[sl])
             b.exp_avg_sq.copy_(v[sl])
-            if b.dtype != torch.float32:
+            if b.zero == 3:
+                # full flat_param storage is released at rest; write the
+                # persistent shard (next forward gathers the fresh values)
+                b.shard_param.copy_(master[sl].to(b.dtype))
+                b.release_params()
+            elif b.dtype != torch.float32:
                 b.flat_param.copy_(master.to(b.dtype))
             else:
                 b.flat_param.copy_(master)

@@ -1,15 +1,34 @@
-"""ZeRO configuration helper (reference: cfg.train.zero_optimization,
-configs/common/train.py:61-67 / graph_base.py:69-70).
+"""ZeRO configuration + stage-3 parameter-sharding manager.
 
-The sharding itself lives in FusedAdamW (flat buckets make the DP slices
-contiguous): stage 1 shards optimizer state, stage 2 additionally
-reduce-scatters gradients so each rank only materializes its own grad slice
-reduction.  This module just maps config -> optimizer kwargs.
+Reference capability: cfg.train.zero_optimization (reference
+configs/common/train.py:61-67, graph_base.py:69-70; stage 3 exercised
+end-to-end in reference tests/models/test_gpt.py:186-199 — OneFlow's
+``enable_zero`` shards params transparently in the compiled graph).
+
+MI355X-native design: stages 1/2 live inside FusedAdamW's flat buckets
+(contiguous DP slices).  Stage 3 adds an FSDP-style lifecycle on top of the
+same buckets, at per-transformer-layer granularity:
+
+  * params at rest are a 1/dp shard per bucket (``shard_param``); the full
+    flat buffer's storage is freed (``untyped_storage().resize_(0)``) —
+    p.data stays a view into that storage and becomes valid again on gather
+  * a forward pre-hook on each unit all-gathers its buckets, a forward
+    post-hook frees them again (they are re-gathered by the backward
+    pre-hook); the root's leftover params (embeddings/head/final LN) are
+    gathered at model forward and kept until grad reduction
+  * per-param post-accumulate-grad hooks track a pending set per bucket;
+    when a unit's grads are complete they are average-reduce-scattered into
+    the owned shard (accumulating across micro-batches) and both the full
+    grad AND param buffers are freed
+  * the optimizer update then touches only the owned shard (exactly the
+    stage-1/2 code path), and the next forward re-gathers fresh params.
 """
+
+import torch
 
 from ..config import try_get_key
 
-__all__ = ["zero_stage_from_config"]
+__all__ = ["zero_stage_from_config", "setup_zero3", "ZeRO3Manager"]
 
 
 def zero_stage_from_config(cfg):
@@ -17,8 +36,138 @@ def zero_stage_from_config(cfg):
     if not enabled:
         return 0
     stage = int(try_get_key(cfg, "train.zero_optimization.stage", default=1))
-    if stage > 2:
-        raise NotImplementedError(
-            "ZeRO stage 3 (parameter sharding) is not implemented yet; use 1 or 2"
-        )
+    if stage not in (1, 2, 3):
+        raise ValueError(f"ZeRO stage must be 1, 2 or 3, got {stage}")
     return stage
+
+
+def _default_units(model):
+    """One unit per transformer block; everything else (embeddings, head,
+    final norm) belongs to the root unit (-1)."""
+    from ..layers.transformer_layer import TransformerLayer
+
+    units = [m for m in model.modules() if isinstance(m, TransformerLayer)]
+    if not units:
+        # non-transformer model: treat every direct child as a unit
+        units = [m for m in model.children()
+                 if sum(p.numel() for p in m.parameters()) > 0]
+    return units
+
+
+class ZeRO3Manager:
+    def __init__(self, model, optimizer, units=None, reshard_after_forward=True):
+        from ..utils import distributed as du
+
+        dutil = du.get_dist_util()
+        if dutil.pipeline_parallel_size > 1:
+            raise NotImplementedError(
+                "ZeRO stage 3 is not composed with pipeline parallelism yet; "
+                "use stage 1/2 with PP"
+            )
+        assert optimizer._buckets is None, (
+            "setup_zero3 must run before the optimizer builds its buckets "
+            "(i.e. before the first step/overlap-hook registration)"
+        )
+        self.model = model
+        self.optimizer = optimizer
+        self.reshard_after_forward = reshard_after_forward
+        self._du = du
+
+        units = list(units) if units is not None else _default_units(model)
+        unit_params = set()
+        for ui, unit in enumerate(units):
+            for p in unit.parameters():
+                p._zero3_unit = ui
+                unit_params.add(id(p))
+        for p in model.parameters():
+            if id(p) not in unit_params:
+                p._zero3_unit = -1
+
+        optimizer.zero_stage = 3
+        _ = optimizer.buckets  # build now, with units tagged
+        if optimizer._zero_eff != 3:
+            return  # dp == 1: degraded to plain, no hooks needed
+
+        self._unit_buckets = {}
+        for _, b in optimizer.buckets:
+            self._unit_buckets.setdefault(b.unit, []).append(b)
+        self._param_bucket = {}
+        for _, b in optimizer.buckets:
+            for p in b.params:
+                self._param_bucket[id(p)] = b
+
+        for ui, unit in enumerate(units):
+            if ui not in self._unit_buckets:
+                continue  # unit had no trainable params
+            unit.register_forward_pre_hook(self._make_fwd_pre(ui))
+            if reshard_after_forward:
+                unit.register_forward_hook(self._make_fwd_post(ui))
+            unit.register_full_backward_pre_hook(self._make_bwd_pre(ui))
+        # root params live for the whole fwd+bwd (used at both ends)
+        model.register_forward_pre_hook(self._root_fwd_pre)
+
+        for p in model.parameters():
+            if p.requires_grad and id(p) in self._param_bucket:
+                p.register_post_accumulate_grad_hook(self._grad_hook)
+
+        model._zero3_optimizer = optimizer  # checkpointer discovery
+
+    # -- hooks --------------------------------------------------------------
+
+    def _group(self):
+        return self._du.get_dist_util().data_parallel_group
+
+    def _make_fwd_pre(self, ui):
+        def hook(module, args):
+            g = self._group()
+            for b in self._unit_buckets[ui]:
+                b.materialize_params(g)
+            # grads are materialized by the backward pre-hook, so only the
+            # units currently in backward hold full-size grad buffers
+        return hook
+
+    def _make_fwd_post(self, ui):
+        def hook(module, args, output):
+            # free after forward; the backward pre-hook re-gathers.  During
+            # activation-checkpoint recompute this releases again and the
+            # (second) backward pre-hook of the recomputed graph re-gathers.
+            for b in self._unit_buckets[ui]:
+                b.release_params()
+        return hook
+
+    def _make_bwd_pre(self, ui):
+        def hook(module, grad_output):
+            g = self._group()
+            for b in self._unit_buckets[ui]:
+                b.materialize_params(g)
+                if not b.grads_live():
+                    b.materialize_grads()
+        return hook
+
+    def _root_fwd_pre(self, module, args):
+        g = self._group()
+        for b in self._unit_buckets.get(-1, []):
+            b.materialize_params(g)
+            if torch.is_grad_enabled() and module.training:
+                b.materialize_grads()
+
+    def _grad_hook(self, p):
+        b = self._param_bucket.get(id(p))
+        if b is None or not hasattr(b, "pending"):
+            return
+        b.pending.discard(id(p))
+        if not b.pending and b.grads_live():
+            dutil = self._du.get_dist_util()
+            b.reduce_release_grads(dutil.data_parallel_group,
+                                   dutil.data_parallel_size)
+            b.release_params()
+
+
+def setup_zero3(model, optimizer, units=None, reshard_after_forward=True):
+    """Enable ZeRO-3 parameter sharding on ``model``/``optimizer``.
+
+    Must be called after the model is on its final device and before the
+    optimizer's buckets are built.  Returns the manager (kept alive by the
+    hook closures)."""
+    return ZeRO3Manager(model, optimizer, units=units,
+                        reshard_after_forward=reshard_after_forward)

@@ -42,7 +42,10 @@ def _consolidate_model_state(model):
             out[name] = tp_merge(shards, dim,
                                  getattr(p, "tp_fused_chunks", 1)).cpu()
         else:
-            out[name] = t.detach().cpu()
+            t = t.detach()
+            # clone on CPU: .cpu() would alias the flat bucket storage, which
+            # ZeRO-3 releases before torch.save serializes the state
+            out[name] = t.clone() if t.device.type == "cpu" else t.cpu()
     return out
 
 
@@ -87,6 +90,13 @@ class Checkpointer:
     def add_checkpointable(self, key, obj):
         self.checkpointables[key] = obj
 
+    def _zero3_opt(self):
+        """The ZeRO-3 optimizer managing this model's released param storage
+        (p.data is unreadable at rest; gather around state_dict/load)."""
+        opt = self.checkpointables.get("optimizer") or getattr(
+            self.model, "_zero3_optimizer", None)
+        return opt if hasattr(opt, "materialize_all_params") else None
+
     def save(self, name, **kwargs):
         if not self.save_dir:
             return
@@ -97,7 +107,12 @@ class Checkpointer:
         du.synchronize()
 
         # model: consolidate over TP; PP stages write stage files merged below
+        z3 = self._zero3_opt()
+        if z3 is not None:
+            z3.materialize_all_params()
         model_state = _consolidate_model_state(self.model)
+        if z3 is not None:
+            z3.release_all_params()
         pp = dutil.pipeline_parallel_size
         writer = dutil.data_parallel_rank == 0 and dutil.tensor_parallel_rank == 0
         if pp > 1:
@@ -157,7 +172,13 @@ class Checkpointer:
         model_file = os.path.join(path, "model.pt")
         full_state = torch.load(model_file, map_location="cpu", weights_only=False)
         local = _shard_for_load(self.model, full_state)
+        z3 = self._zero3_opt()
+        if z3 is not None:
+            z3.materialize_all_params()  # p.data must be writable
         missing, unexpected = self.model.load_state_dict(local, strict=False)
+        if z3 is not None:
+            z3.refresh_shards_from_params()
+            z3.release_all_params()
         missing = [m for m in missing if m in dict(self.model.named_parameters())]
         if missing:
             self.logger.warning(f"missing keys in checkpoint: {missing[:10]}...")
