@@ -80,12 +80,14 @@ void ce_bwd_bf16(const void*, const int64_t*, const float*, const float*, const 
                  void*, int64_t, int64_t, int64_t, int64_t, hipStream_t);
 void ce_bwd_f32(const void*, const int64_t*, const float*, const float*, const float*,
                 void*, int64_t, int64_t, int64_t, int64_t, hipStream_t);
-void flash_fwd_bf16(const void*, const void*, const void*, void*, float*, int64_t,
+void flash_fwd_bf16(const void*, const void*, const void*, void*, float*,
+                    const int*, int64_t, int64_t, int64_t, int64_t, int64_t,
                     int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
-                    int64_t, int64_t, int64_t, int64_t, int, int, int, int, int, float,
-                    float, uint64_t, int, hipStream_t);
+                    int, int, int, int, int, float, float, uint64_t, int,
+                    hipStream_t);
 void flash_bwd_bf16(const void*, const void*, const void*, const void*, const void*,
-                    const float*, float*, void*, void*, void*, int64_t, int64_t,
+                    const float*, float*, const int*, void*, void*, void*, int64_t,
+                    int64_t,
                     int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
                     int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
                     int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
@@ -95,6 +97,14 @@ void attn_dropout_apply_bf16(void*, int64_t, int64_t, int64_t, float, uint64_t,
                              hipStream_t);
 void attn_dropout_apply_f32(void*, int64_t, int64_t, int64_t, float, uint64_t,
                             hipStream_t);
+void embedding_fwd_bf16(const int64_t*, const void*, void*, int64_t, int64_t,
+                        int64_t, int64_t, hipStream_t);
+void embedding_fwd_f32(const int64_t*, const void*, void*, int64_t, int64_t,
+                       int64_t, int64_t, hipStream_t);
+void embedding_bwd_bf16(const int64_t*, const void*, float*, int64_t, int64_t,
+                        int64_t, int64_t, int64_t, hipStream_t);
+void embedding_bwd_f32(const int64_t*, const void*, float*, int64_t, int64_t,
+                       int64_t, int64_t, int64_t, hipStream_t);
 void rope_fwd_bf16(const void*, void*, const float*, const float*, int64_t, int64_t,
                    int64_t, int64_t, int64_t, int64_t, int, int, int, int, int,
                    hipStream_t);
@@ -321,13 +331,68 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets, torch::Tensor 
   return dlogits;
 }
 
+
+// ---------------------------------------------------------------------------
+// embedding gather/scatter (K10)
+// ---------------------------------------------------------------------------
+torch::Tensor embedding_fwd(torch::Tensor ids, torch::Tensor w, int64_t vocab_start) {
+  CHECK_IN(ids);
+  CHECK_IN(w);
+  TORCH_CHECK(ids.scalar_type() == torch::kInt64, "ids must be int64");
+  const int64_t H = w.size(1);
+  TORCH_CHECK((H * w.element_size()) % 16 == 0, "hidden dim must be 16B-aligned");
+  auto sizes = ids.sizes().vec();
+  sizes.push_back(H);
+  auto out = torch::empty(sizes, w.options());
+  const int64_t N = ids.numel();
+  if (is_bf16(w))
+    embedding_fwd_bf16(ids.data_ptr<int64_t>(), w.data_ptr(), out.data_ptr(), N, H,
+                       vocab_start, w.size(0), cur_stream());
+  else
+    embedding_fwd_f32(ids.data_ptr<int64_t>(), w.data_ptr(), out.data_ptr(), N, H,
+                      vocab_start, w.size(0), cur_stream());
+  check_launch("embedding_fwd");
+  return out;
+}
+
+torch::Tensor embedding_bwd(torch::Tensor ids, torch::Tensor dout,
+                            int64_t vocab_local, int64_t vocab_start,
+                            int64_t padding_idx, torch::ScalarType grad_dtype) {
+  CHECK_IN(ids);
+  auto douts = dout.contiguous();
+  const int64_t H = douts.size(-1);
+  const int64_t N = ids.numel();
+  auto ws = torch::zeros({vocab_local, H},
+                         douts.options().dtype(torch::kFloat32));
+  const int64_t padding_row =
+      (padding_idx >= 0) ? padding_idx - vocab_start : -1;
+  if (is_bf16(douts))
+    embedding_bwd_bf16(ids.data_ptr<int64_t>(), douts.data_ptr(),
+                       ws.data_ptr<float>(), N, H, vocab_start, vocab_local,
+                       padding_row, cur_stream());
+  else
+    embedding_bwd_f32(ids.data_ptr<int64_t>(), douts.data_ptr(),
+                      ws.data_ptr<float>(), N, H, vocab_start, vocab_local,
+                      padding_row, cur_stream());
+  check_launch("embedding_bwd");
+  return grad_dtype == torch::kFloat32 ? ws : ws.to(grad_dtype);
+}
+
 // ---------------------------------------------------------------------------
 // flash attention
 // ---------------------------------------------------------------------------
 std::tuple<torch::Tensor, torch::Tensor> flash_fwd(torch::Tensor q, torch::Tensor k,
                                                    torch::Tensor v, double scale,
                                                    double p_drop, int64_t seed,
-                                                   bool causal) {
+                                                   bool causal,
+                                                   c10::optional<torch::Tensor> kv_len) {
+  const int* kvl = nullptr;
+  if (kv_len.has_value()) {
+    TORCH_CHECK(kv_len->scalar_type() == torch::kInt32 && kv_len->is_cuda() &&
+                    kv_len->is_contiguous() && kv_len->numel() == q.size(0),
+                "kv_len must be a contiguous int32 cuda tensor of size B");
+    kvl = kv_len->data_ptr<int>();
+  }
   // q/k/v: [B, S, H, D] (strided views into a fused qkv buffer are fine; the
   // last dim must be contiguous and 16B-aligned)
   TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
@@ -341,7 +406,7 @@ std::tuple<torch::Tensor, torch::Tensor> flash_fwd(torch::Tensor q, torch::Tenso
   auto o = torch::empty({B, Sq, H, D}, q.options());
   auto lse = torch::empty({B, H, Sq}, q.options().dtype(torch::kFloat32));
   flash_fwd_bf16(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                 lse.data_ptr<float>(), q.stride(0), q.stride(1), q.stride(2),
+                 lse.data_ptr<float>(), kvl, q.stride(0), q.stride(1), q.stride(2),
                  k.stride(0), k.stride(1), k.stride(2), v.stride(0), v.stride(1),
                  v.stride(2), o.stride(0), o.stride(1), o.stride(2), B, H, Sq, Sk, D,
                  (float)scale, (float)p_drop, (uint64_t)seed, causal ? 1 : 0,
@@ -353,7 +418,15 @@ std::tuple<torch::Tensor, torch::Tensor> flash_fwd(torch::Tensor q, torch::Tenso
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> flash_bwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor o,
     torch::Tensor dout, torch::Tensor lse, double scale, double p_drop, int64_t seed,
-    bool causal, c10::optional<torch::Tensor> dqkv) {
+    bool causal, c10::optional<torch::Tensor> dqkv,
+    c10::optional<torch::Tensor> kv_len) {
+  const int* kvl = nullptr;
+  if (kv_len.has_value()) {
+    TORCH_CHECK(kv_len->scalar_type() == torch::kInt32 && kv_len->is_cuda() &&
+                    kv_len->is_contiguous() && kv_len->numel() == q.size(0),
+                "kv_len must be a contiguous int32 cuda tensor of size B");
+    kvl = kv_len->data_ptr<int>();
+  }
   const int B = (int)q.size(0), Sq = (int)q.size(1), H = (int)q.size(2),
             D = (int)q.size(3);
   const int Sk = (int)k.size(1);
@@ -373,8 +446,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> flash_bwd(
   auto drow = torch::empty({B, H, Sq}, q.options().dtype(torch::kFloat32));
   flash_bwd_bf16(
       q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(), dout.data_ptr(),
-      lse.data_ptr<float>(), drow.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
-      dv.data_ptr(), q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),
+      lse.data_ptr<float>(), drow.data_ptr<float>(), kvl, dq.data_ptr(),
+      dk.data_ptr(), dv.data_ptr(), q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),
       k.stride(2), v.stride(0), v.stride(1), v.stride(2), o.stride(0), o.stride(1),
       o.stride(2), dout.stride(0), dout.stride(1), dout.stride(2), dq.stride(0),
       dq.stride(1), dq.stride(2), dk.stride(0), dk.stride(1), dk.stride(2),
@@ -469,6 +542,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_dropout_res_bwd", &bias_dropout_res_bwd);
   m.def("softmax_fwd", &softmax_fwd);
   m.def("softmax_bwd", &softmax_bwd);
+  m.def("embedding_fwd", &embedding_fwd);
+  m.def("embedding_bwd", &embedding_bwd);
   m.def("flash_fwd", &flash_fwd);
   m.def("flash_bwd", &flash_bwd);
   m.def("attn_dropout_apply", &attn_dropout_apply);

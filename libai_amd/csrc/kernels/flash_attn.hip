@@ -179,10 +179,10 @@ template <int D>
 __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_fwd_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, bf16_t* __restrict__ o, float* __restrict__ lse,
-    int64_t q_sb, int64_t q_ss, int64_t q_sh, int64_t k_sb, int64_t k_ss,
-    int64_t k_sh, int64_t v_sb, int64_t v_ss, int64_t v_sh, int64_t o_sb,
-    int64_t o_ss, int64_t o_sh, int H, int Sq, int Sk, float scale, float p_drop,
-    uint64_t seed, int causal) {
+    const int* __restrict__ kv_len, int64_t q_sb, int64_t q_ss, int64_t q_sh,
+    int64_t k_sb, int64_t k_ss, int64_t k_sh, int64_t v_sb, int64_t v_ss,
+    int64_t v_sh, int64_t o_sb, int64_t o_ss, int64_t o_sh, int H, int Sq, int Sk,
+    float scale, float p_drop, uint64_t seed, int causal) {
   constexpr int DT = D / 32;
   constexpr int KC = D / 16;
 
@@ -227,7 +227,11 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_fwd_kernel(
   for (int t = 0; t < DT; ++t) oacc[t] = f32x16_t{};
   float m_run = -3.0e38f, l_run = 0.f;
 
-  const int kv_end = causal ? min(Sk, q_block + QBLK) : Sk;
+  // right-padding support: keys at/after kv_len[b] are masked out (BERT-style
+  // per-sequence valid length; replaces the reference's additive -10000 pad
+  // mask, attention.py:221-226)
+  const int sk_eff = (kv_len != nullptr) ? kv_len[b] : Sk;
+  const int kv_end = causal ? min(sk_eff, q_block + QBLK) : sk_eff;
   const int n_rounds = CDIV(kv_end, 2 * KVB);
   const float ks = (p_drop > 0.f) ? 1.0f / (1.0f - p_drop) : 1.0f;
   const int qg = q_base + l31;
@@ -262,8 +266,8 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_fwd_kernel(
       int kva = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
       int kvb = kva + 32;
       float a = s0[r], bb = s1[r];
-      if (kva >= Sk || (causal && kva > qg)) a = -3.0e38f;
-      if (kvb >= Sk || (causal && kvb > qg)) bb = -3.0e38f;
+      if (kva >= sk_eff || (causal && kva > qg)) a = -3.0e38f;
+      if (kvb >= sk_eff || (causal && kvb > qg)) bb = -3.0e38f;
       sv[0][r] = a;
       sv[1][r] = bb;
       pmax = fmaxf(pmax, fmaxf(a, bb));
@@ -380,7 +384,8 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
-    bf16_t* __restrict__ dk, bf16_t* __restrict__ dv, int64_t q_sb, int64_t q_ss,
+    const int* __restrict__ kv_len, bf16_t* __restrict__ dk,
+    bf16_t* __restrict__ dv, int64_t q_sb, int64_t q_ss,
     int64_t q_sh, int64_t k_sb, int64_t k_ss, int64_t k_sh, int64_t v_sb,
     int64_t v_ss, int64_t v_sh, int64_t do_sb, int64_t do_ss, int64_t do_sh,
     int64_t dk_sb, int64_t dk_ss, int64_t dk_sh, int64_t dv_sb, int64_t dv_ss,
@@ -439,8 +444,12 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
   }
 
   const float ks = (p_drop > 0.f) ? 1.0f / (1.0f - p_drop) : 1.0f;
+  const int sk_eff = (kv_len != nullptr) ? kv_len[b] : Sk;
   const int qt_start = causal ? (kv_block / QTILE) : 0;
-  const int n_qtiles = CDIV(Sq, QTILE);
+  // fully-padded kv block: accumulators stay zero, skip straight to the
+  // (zero) writes.  kv_block is uniform across the block, so this does not
+  // break the __syncthreads inside the loop.
+  const int n_qtiles = (kv_block < sk_eff) ? CDIV(Sq, QTILE) : 0;
 
   for (int qt = qt_start; qt < n_qtiles; ++qt) {
     const int q0 = qt * QTILE;
@@ -504,7 +513,8 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
             const int k = r4 * 4 + j;
             const int r = c16 * 8 + k;
             const int qrow = qbase + j;
-            const bool valid = qrow < Sq && kvg < Sk && (!causal || qrow >= kvg);
+            const bool valid =
+                qrow < Sq && kvg < sk_eff && (!causal || qrow >= kvg);
             float p = 0.f;
             if (valid) p = __expf(s[r] * scale - lse_lds[qrow - q0]);
             float keep = valid ? 1.f : 0.f;
@@ -560,7 +570,8 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
-    bf16_t* __restrict__ dq, int64_t q_sb, int64_t q_ss, int64_t q_sh, int64_t k_sb,
+    const int* __restrict__ kv_len, bf16_t* __restrict__ dq, int64_t q_sb,
+    int64_t q_ss, int64_t q_sh, int64_t k_sb,
     int64_t k_ss, int64_t k_sh, int64_t v_sb, int64_t v_ss, int64_t v_sh,
     int64_t do_sb, int64_t do_ss, int64_t do_sh, int64_t dq_sb, int64_t dq_ss,
     int64_t dq_sh, int H, int Sq, int Sk, float scale, float p_drop, uint64_t seed,
@@ -618,7 +629,8 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
   for (int t = 0; t < DT; ++t) dq_acc[t] = f32x16_t{};
 
   const float ks = (p_drop > 0.f) ? 1.0f / (1.0f - p_drop) : 1.0f;
-  const int kv_end = causal ? min(Sk, q_block + QBLK) : Sk;
+  const int sk_eff = (kv_len != nullptr) ? kv_len[b] : Sk;
+  const int kv_end = causal ? min(sk_eff, q_block + QBLK) : sk_eff;
   const int n_tiles = CDIV(kv_end, KVTILE);
 
   for (int tile = 0; tile < n_tiles; ++tile) {
@@ -659,7 +671,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
         for (int j = 0; j < 4; ++j) {
           const int r = r4 * 4 + j;
           const int kv = kvb + j;
-          const bool valid = kv < Sk && qg < Sq && (!causal || kv <= qg);
+          const bool valid = kv < sk_eff && qg < Sq && (!causal || kv <= qg);
           float p = valid ? __expf(st[r] * scale - lse_lane) : 0.f;
           float keep = (p_drop > 0.f && valid) ? drop_keep_byte(h, j, thr8, ks)
                                                : (valid ? 1.f : 0.f);
@@ -718,8 +730,9 @@ __global__ void attn_dropout_apply_kernel(typename E::T* __restrict__ x, int64_t
 }  // namespace
 
 extern "C" void flash_fwd_bf16(const void* q, const void* k, const void* v, void* o,
-                               float* lse, int64_t q_sb, int64_t q_ss, int64_t q_sh,
-                               int64_t k_sb, int64_t k_ss, int64_t k_sh, int64_t v_sb,
+                               float* lse, const int* kv_len, int64_t q_sb,
+                               int64_t q_ss, int64_t q_sh, int64_t k_sb,
+                               int64_t k_ss, int64_t k_sh, int64_t v_sb,
                                int64_t v_ss, int64_t v_sh, int64_t o_sb, int64_t o_ss,
                                int64_t o_sh, int B, int H, int Sq, int Sk, int D,
                                float scale, float p_drop, uint64_t seed, int causal,
@@ -728,19 +741,20 @@ extern "C" void flash_fwd_bf16(const void* q, const void* k, const void* v, void
   dim3 block(256);
   if (D == 64)
     flash_fwd_kernel<64><<<grid, block, 0, stream>>>(
-        (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (bf16_t*)o, lse, q_sb,
-        q_ss, q_sh, k_sb, k_ss, k_sh, v_sb, v_ss, v_sh, o_sb, o_ss, o_sh, H, Sq, Sk,
-        scale, p_drop, seed, causal);
+        (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (bf16_t*)o, lse,
+        kv_len, q_sb, q_ss, q_sh, k_sb, k_ss, k_sh, v_sb, v_ss, v_sh, o_sb, o_ss,
+        o_sh, H, Sq, Sk, scale, p_drop, seed, causal);
   else if (D == 128)
     flash_fwd_kernel<128><<<grid, block, 0, stream>>>(
-        (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (bf16_t*)o, lse, q_sb,
-        q_ss, q_sh, k_sb, k_ss, k_sh, v_sb, v_ss, v_sh, o_sb, o_ss, o_sh, H, Sq, Sk,
-        scale, p_drop, seed, causal);
+        (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (bf16_t*)o, lse,
+        kv_len, q_sb, q_ss, q_sh, k_sb, k_ss, k_sh, v_sb, v_ss, v_sh, o_sb, o_ss,
+        o_sh, H, Sq, Sk, scale, p_drop, seed, causal);
 }
 
 extern "C" void flash_bwd_bf16(
     const void* q, const void* k, const void* v, const void* o, const void* dout,
-    const float* lse, float* drow_ws, void* dq, void* dk, void* dv, int64_t q_sb,
+    const float* lse, float* drow_ws, const int* kv_len, void* dq, void* dk,
+    void* dv, int64_t q_sb,
     int64_t q_ss, int64_t q_sh, int64_t k_sb, int64_t k_ss, int64_t k_sh,
     int64_t v_sb, int64_t v_ss, int64_t v_sh, int64_t o_sb, int64_t o_ss,
     int64_t o_sh, int64_t do_sb, int64_t do_ss, int64_t do_sh, int64_t dq_sb,
@@ -765,14 +779,14 @@ extern "C" void flash_bwd_bf16(
   }
 #define BWD_ARGS_KV                                                                  \
   (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (const bf16_t*)dout, lse,    \
-      drow_ws, (bf16_t*)dk, (bf16_t*)dv, q_sb, q_ss, q_sh, k_sb, k_ss, k_sh, v_sb,   \
-      v_ss, v_sh, do_sb, do_ss, do_sh, dk_sb, dk_ss, dk_sh, dv_sb, dv_ss, dv_sh, H,  \
-      Sq, Sk, scale, p_drop, seed, causal
+      drow_ws, kv_len, (bf16_t*)dk, (bf16_t*)dv, q_sb, q_ss, q_sh, k_sb, k_ss,       \
+      k_sh, v_sb, v_ss, v_sh, do_sb, do_ss, do_sh, dk_sb, dk_ss, dk_sh, dv_sb,       \
+      dv_ss, dv_sh, H, Sq, Sk, scale, p_drop, seed, causal
 #define BWD_ARGS_Q                                                                   \
   (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (const bf16_t*)dout, lse,    \
-      drow_ws, (bf16_t*)dq, q_sb, q_ss, q_sh, k_sb, k_ss, k_sh, v_sb, v_ss, v_sh,    \
-      do_sb, do_ss, do_sh, dq_sb, dq_ss, dq_sh, H, Sq, Sk, scale, p_drop, seed,      \
-      causal
+      drow_ws, kv_len, (bf16_t*)dq, q_sb, q_ss, q_sh, k_sb, k_ss, k_sh, v_sb, v_ss,  \
+      v_sh, do_sb, do_ss, do_sh, dq_sb, dq_ss, dq_sh, H, Sq, Sk, scale, p_drop,      \
+      seed, causal
   dim3 block(256);
   if (D == 64) {
     flash_bwd_kv_kernel<64>

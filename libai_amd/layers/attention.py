@@ -115,6 +115,7 @@ class MultiheadAttention(nn.Module):
             from ..ops.attention import (
                 flash_attention_available,
                 flash_attention_qkv,
+                mask_kv_len,
             )
 
             b, s, _ = hidden_states.shape
@@ -130,6 +131,7 @@ class MultiheadAttention(nn.Module):
                     p_drop=self.attention_dropout_prob,
                     causal=self.attn_mask_type == AttnMaskType.causal,
                     training=self.training,
+                    kv_len=mask_kv_len(attention_mask),
                 )
                 context = o.reshape(b, s, self.num_heads_local * self.head_size)
                 out, bias = self.dense(context)

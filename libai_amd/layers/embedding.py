@@ -37,6 +37,10 @@ class Embedding(nn.Module):
                 self.weight[padding_idx].fill_(0)
 
     def forward(self, ids):
+        from ..ops.embedding import fused_embedding, fused_embedding_available
+
+        if fused_embedding_available(self.weight):
+            return fused_embedding(ids, self.weight, 0, self.padding_idx)
         return F.embedding(ids, self.weight, padding_idx=self.padding_idx)
 
 
@@ -64,7 +68,14 @@ class VocabEmbedding(nn.Module):
         init_tp_shard_(self.weight, (num_embeddings, embedding_dim), init_method, 0)
 
     def forward(self, ids):
+        from ..ops.embedding import fused_embedding, fused_embedding_available
+
         tp = du.get_dist_util().tensor_parallel_size
+        if fused_embedding_available(self.weight):
+            # OOV ids -> zero rows in-kernel (no sub/clamp/mask-mul passes)
+            out = fused_embedding(ids, self.weight, self.vocab_start,
+                                  self.padding_idx)
+            return reduce_from_tensor_parallel_region(out) if tp > 1 else out
         if tp == 1:
             return F.embedding(ids, self.weight, padding_idx=self.padding_idx)
         local = ids - self.vocab_start

@@ -64,14 +64,22 @@ class BertEmbeddings(nn.Module):
 
 def extended_attn_mask(attention_mask):
     """[b, s] (1=visible) -> [b, sq, sk] uint8 (1=MASKED) via the outer
-    product of visibilities (reference: bert_model.py:40-60)."""
+    product of visibilities (reference: bert_model.py:40-60).
+
+    When the visibility is pure right-padding (a contiguous prefix of 1s)
+    the per-sequence valid length is attached as ``mask._kv_len`` — the
+    flash-attention kernel consumes that instead of the O(s^2) additive
+    mask, so padded BERT/RoBERTa batches stay on the fused path."""
     if attention_mask is None:
         return None
     if attention_mask.dim() == 3:  # pre-built [b, sq, sk] (1=MASKED)
         return attention_mask.to(torch.uint8)
     m = attention_mask.to(torch.uint8)
     visible = m.unsqueeze(1) * m.unsqueeze(2)  # [b, s, s]
-    return (1 - visible).to(torch.uint8)
+    mask = (1 - visible).to(torch.uint8)
+    if m.dim() == 2 and (m[:, :-1] >= m[:, 1:]).all():
+        mask._kv_len = m.sum(dim=-1, dtype=torch.int32)
+    return mask
 
 
 class BertPooler(nn.Module):

@@ -1,11 +1,12 @@
 """Fused flash-style attention (bf16, MFMA) with recompute backward.
 
-Forward: one HIP kernel (csrc/kernels/flash_attn.hip) — the S x S score
-matrix never touches HBM; O and the per-row logsumexp are saved.
+Forward: one HIP kernel (libai_amd/csrc/kernels/flash_attn.hip) — the S x S
+score matrix never touches HBM; O and the per-row logsumexp are saved.
 
-Backward (v1): recompute composition — S = QK^T (rocBLAS), P = exp(S*scale -
-lse) with the philox dropout mask regenerated in-kernel, then the four
-gradient GEMMs.  A fully-fused HIP backward is the planned v2.
+Padding masks: BERT/RoBERTa-style right-padded batches are expressed as a
+per-sequence valid key length (int32 [B]); the kernels mask kv >= kv_len[b]
+in-register, replacing the reference's additive -10000 [b, s, s] mask
+(reference libai/layers/attention.py:221-226) without materializing scores.
 
 Replaces the reference's K2-K5 chain (reference
 libai/layers/attention.py:211-253).
@@ -15,7 +16,14 @@ import torch
 
 from ._ext import draw_seed, ext
 
-__all__ = ["flash_attention", "flash_attention_qkv", "flash_attention_available"]
+__all__ = ["flash_attention", "flash_attention_qkv", "flash_attention_available",
+           "mask_kv_len"]
+
+
+def mask_kv_len(pad_mask):
+    """The per-sequence valid-length tensor attached to a padding mask by
+    ``extended_attn_mask`` when the mask is pure right-padding, else None."""
+    return getattr(pad_mask, "_kv_len", None) if pad_mask is not None else None
 
 
 def flash_attention_available(head_dim, dtype, device, sq, sk, pad_mask):
@@ -23,7 +31,7 @@ def flash_attention_available(head_dim, dtype, device, sq, sk, pad_mask):
         device.type == "cuda"
         and dtype == torch.bfloat16
         and head_dim in (64, 128)
-        and pad_mask is None
+        and (pad_mask is None or mask_kv_len(pad_mask) is not None)
         and sk % 8 == 0
         and sq == sk  # training self-attention (no KV-cache decode)
     )
@@ -38,13 +46,14 @@ class _FlashAttnQKVFn(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, qkv5, scale, p_drop, causal):
+    def forward(ctx, qkv5, scale, p_drop, causal, kv_len):
         q = qkv5[..., 0, :]
         k = qkv5[..., 1, :]
         v = qkv5[..., 2, :]
         seed = draw_seed() if p_drop > 0 else 0
-        o, lse = ext().flash_fwd(q, k, v, scale, p_drop, seed, causal)
+        o, lse = ext().flash_fwd(q, k, v, scale, p_drop, seed, causal, kv_len)
         ctx.save_for_backward(qkv5, o, lse)
+        ctx.kv_len = kv_len
         ctx.meta = (scale, p_drop, seed, causal)
         return o
 
@@ -55,23 +64,28 @@ class _FlashAttnQKVFn(torch.autograd.Function):
         dqkv = torch.empty_like(qkv5)
         ext().flash_bwd(
             qkv5[..., 0, :], qkv5[..., 1, :], qkv5[..., 2, :], o, do.contiguous(),
-            lse, scale, p_drop, seed, causal, dqkv,
+            lse, scale, p_drop, seed, causal, dqkv, ctx.kv_len,
         )
-        return dqkv, None, None, None
+        return dqkv, None, None, None, None
 
 
-def flash_attention_qkv(qkv5, scale, p_drop=0.0, causal=True, training=True):
-    """qkv5: packed [B, S, H, 3, D] bf16 -> O [B, S, H, D] (zero-copy in/out)."""
-    return _FlashAttnQKVFn.apply(qkv5, scale, p_drop if training else 0.0, causal)
+def flash_attention_qkv(qkv5, scale, p_drop=0.0, causal=True, training=True,
+                        kv_len=None):
+    """qkv5: packed [B, S, H, 3, D] bf16 -> O [B, S, H, D] (zero-copy in/out).
+
+    kv_len: optional int32 [B] — keys at/after kv_len[b] are masked out."""
+    return _FlashAttnQKVFn.apply(qkv5, scale, p_drop if training else 0.0, causal,
+                                 kv_len)
 
 
 class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, scale, p_drop, causal):
+    def forward(ctx, q, k, v, scale, p_drop, causal, kv_len):
         # q, k, v: [B, S, H, D] (may be strided views of the fused qkv buffer)
         seed = draw_seed() if p_drop > 0 else 0
-        o, lse = ext().flash_fwd(q, k, v, scale, p_drop, seed, causal)
+        o, lse = ext().flash_fwd(q, k, v, scale, p_drop, seed, causal, kv_len)
         ctx.save_for_backward(q, k, v, o, lse)
+        ctx.kv_len = kv_len
         ctx.meta = (scale, p_drop, seed, causal)
         return o
 
@@ -80,11 +94,14 @@ class _FlashAttnFn(torch.autograd.Function):
         q, k, v, o, lse = ctx.saved_tensors
         scale, p_drop, seed, causal = ctx.meta
         dq, dk, dv = ext().flash_bwd(
-            q, k, v, o, do.contiguous(), lse, scale, p_drop, seed, causal, None
+            q, k, v, o, do.contiguous(), lse, scale, p_drop, seed, causal, None,
+            ctx.kv_len,
         )
-        return dq, dk, dv, None, None, None
+        return dq, dk, dv, None, None, None, None
 
 
-def flash_attention(q, k, v, scale, p_drop=0.0, causal=True, training=True):
+def flash_attention(q, k, v, scale, p_drop=0.0, causal=True, training=True,
+                    kv_len=None):
     """q, k, v: [B, S, H, D] bf16 -> O [B, S, H, D]."""
-    return _FlashAttnFn.apply(q, k, v, scale, p_drop if training else 0.0, causal)
+    return _FlashAttnFn.apply(q, k, v, scale, p_drop if training else 0.0, causal,
+                              kv_len)

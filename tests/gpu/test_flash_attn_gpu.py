@@ -157,3 +157,77 @@ def test_flash_qkv_packed_matches_view_path():
     o1.backward(g)
     o2.backward(g)
     assert torch.equal(q1.grad, q2.grad)
+
+
+def _ref_attention_padded(q, k, v, scale, causal, kv_len):
+    qh = q.float().permute(0, 2, 1, 3)
+    kh = k.float().permute(0, 2, 1, 3)
+    vh = v.float().permute(0, 2, 1, 3)
+    s = torch.matmul(qh, kh.transpose(-1, -2)) * scale
+    sq, sk = s.shape[-2], s.shape[-1]
+    pad = torch.arange(sk, device=q.device)[None, None, None, :] >= kv_len[:, None, None, None]
+    s = s.masked_fill(pad, float("-inf"))
+    if causal:
+        cm = torch.ones(sq, sk, dtype=torch.bool, device=q.device).tril_(sk - sq)
+        s = s.masked_fill(~cm, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    return torch.matmul(p, vh).permute(0, 2, 1, 3)
+
+
+@pytest.mark.parametrize("d", [64, 128])
+@pytest.mark.parametrize("causal", [False, True])
+def test_flash_kv_len_padding_matches_reference(d, causal):
+    """Per-sequence kv_len masking (BERT right-padding) fwd+bwd vs fp32."""
+    from libai_amd.ops.attention import flash_attention
+
+    torch.manual_seed(1)
+    b, s, h = 4, 384, 4
+    lens = torch.tensor([384, 200, 57, 1], device="cuda", dtype=torch.int32)
+    q = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    scale = 1.0 / math.sqrt(d)
+    o = flash_attention(q, k, v, scale, p_drop=0.0, causal=causal, kv_len=lens)
+
+    qr = q.detach().float().requires_grad_(True)
+    kr = k.detach().float().requires_grad_(True)
+    vr = v.detach().float().requires_grad_(True)
+    ref = _ref_attention_padded(qr, kr, vr, scale, causal, lens)
+    err = (o.float() - ref).abs().max().item()
+    assert err < 2e-2, f"padded flash fwd max err {err}"
+
+    # upstream grad zeroed at padded QUERY rows (loss-mask contract: padded
+    # positions never contribute to the loss)
+    g = torch.randn_like(o, dtype=torch.float32)
+    qmask = (torch.arange(s, device="cuda")[None, :] < lens[:, None]).to(torch.float32)
+    g = g * qmask[:, :, None, None]
+    o.backward(g.to(torch.bfloat16))
+    ref.backward(g)
+    for got, want, name in ((q.grad, qr.grad, "dq"), (k.grad, kr.grad, "dk"),
+                            (v.grad, vr.grad, "dv")):
+        err = (got.float() - want).abs().max().item()
+        assert err < 5e-2, f"padded flash {name} max err {err}"
+    # grads wrt padded keys must be exactly zero
+    kpad = (torch.arange(s, device="cuda")[None, :] >= lens[:, None])
+    assert k.grad.float()[kpad].abs().max().item() == 0.0
+    assert v.grad.float()[kpad].abs().max().item() == 0.0
+
+
+def test_bert_padded_batch_takes_flash_path():
+    """extended_attn_mask attaches _kv_len for right-padding and the layer
+    dispatches to the fused kernel (no [b,nh,s,s] scores materialized)."""
+    from libai_amd.models.bert_model import extended_attn_mask
+    from libai_amd.ops.attention import flash_attention_available
+
+    lens = torch.tensor([128, 96, 17], device="cuda")
+    vis = (torch.arange(128, device="cuda")[None, :] < lens[:, None]).to(torch.int64)
+    mask = extended_attn_mask(vis)
+    assert getattr(mask, "_kv_len", None) is not None
+    assert torch.equal(mask._kv_len.long(), lens)
+    assert flash_attention_available(64, torch.bfloat16, mask.device, 128, 128, mask)
+    # non-prefix visibility must NOT claim a kv_len
+    vis2 = vis.clone()
+    vis2[1, 0] = 0
+    mask2 = extended_attn_mask(vis2)
+    assert getattr(mask2, "_kv_len", None) is None
+    assert not flash_attention_available(64, torch.bfloat16, mask2.device, 128, 128, mask2)
