@@ -294,11 +294,10 @@ class DefaultTrainer(TrainerBase):
             ),
         ]
         eval_period = try_get_key(cfg, "train.evaluation.eval_period", default=0)
-        if du.get_dist_util().pipeline_parallel_size > 1:
-            eval_period = 0  # eval loop runs the full forward; stages are pruned
         if eval_period and try_get_key(cfg, "train.evaluation.enabled", default=True):
             def _eval():
-                return self.test(self.cfg, model=self.model)
+                return self.test(self.cfg, model=self.model,
+                                 pipeline_scheduler=self.pipeline_scheduler)
 
             ret.append(hooks.EvalHook(eval_period, _eval))
         if du.is_main_process():
@@ -328,8 +327,12 @@ class DefaultTrainer(TrainerBase):
         return self._trainer.get_batch(data)
 
     @classmethod
-    def test(cls, cfg, model=None, evaluator=None):
-        """Run evaluators over the test loaders (reference: default.py:781-848)."""
+    def test(cls, cfg, model=None, evaluator=None, pipeline_scheduler=None):
+        """Run evaluators over the test loaders (reference: default.py:781-848).
+
+        With ``pipeline_scheduler`` the eval forward runs through the 1F1B
+        engine's run_eval (every stage participates; reference evaluates
+        pipelined graphs too, evaluator.py:119)."""
         from ..evaluation import inference_on_dataset
 
         test_loaders = cls.build_test_loader(cfg)
@@ -342,5 +345,6 @@ class DefaultTrainer(TrainerBase):
             )
             if ev is None:
                 continue
-            results[f"dataset_{i}"] = inference_on_dataset(model, loader, ev)
+            results[f"dataset_{i}"] = inference_on_dataset(
+                model, loader, ev, pipeline_scheduler=pipeline_scheduler)
         return results

@@ -65,13 +65,30 @@ def flatten_results_dict(results):
 
 
 def inference_on_dataset(model, data_loader, evaluator, eval_iter=None,
-                         get_batch=None):
-    """Eval loop: model(**batch) per batch; evaluator.process on rank 0
-    after DP gather (reference: evaluator.py:119-278)."""
+                         get_batch=None, pipeline_scheduler=None):
+    """Eval loop: model(**batch) per batch; evaluator.process on the writer
+    rank after DP gather (reference: evaluator.py:119-278).
+
+    Under pipeline parallelism pass ``pipeline_scheduler``: every stage
+    participates in the P2P forward (PipelineScheduler.run_eval), outputs
+    exist on the LAST stage, and the evaluator runs on that stage's
+    (dp0, tp0) rank; the final results dict is broadcast world-wide
+    (reference runs test() on pipelined graphs the same way,
+    evaluator.py:119-278)."""
     logger = logging.getLogger(__name__)
     if evaluator is None:
         return {}
     evaluator.reset()
+    dutil = du.get_dist_util()
+    pp = dutil.pipeline_parallel_size
+    on_last_stage = pp == 1 or dutil.pipeline_parallel_rank == pp - 1
+    # global rank that runs the evaluator: rank 0 for pp==1 (main process),
+    # else the last stage's (dp0, tp0) rank (stage-major rank layout)
+    writer_rank = 0 if pp == 1 else (
+        (pp - 1) * dutil.data_parallel_size * dutil.tensor_parallel_size)
+    is_writer = (du.is_main_process() if pp == 1 else
+                 (on_last_stage and dutil.data_parallel_rank == 0
+                  and dutil.tensor_parallel_rank == 0))
     total = len(data_loader) if hasattr(data_loader, "__len__") else None
     if eval_iter is not None and total is not None:
         total = min(total, eval_iter)
@@ -99,12 +116,17 @@ def inference_on_dataset(model, data_loader, evaluator, eval_iter=None,
                 for k, v in data.items()
             }
             t0 = time.perf_counter()
-            outputs = model(**data)
+            if pipeline_scheduler is not None:
+                outputs = pipeline_scheduler.run_eval(data)
+            else:
+                outputs = model(**data)
             if torch.cuda.is_available():
                 torch.cuda.synchronize()
             total_compute_time += time.perf_counter() - t0
 
-            # gather DP-sharded outputs + labels to rank 0
+            if not on_last_stage:
+                continue  # outputs live on the last stage only
+            # gather DP-sharded outputs + labels across the stage's DP group
             gathered_out = {
                 k: du.tensor_to_rank0(v) if torch.is_tensor(v) else v
                 for k, v in outputs.items()
@@ -113,7 +135,7 @@ def inference_on_dataset(model, data_loader, evaluator, eval_iter=None,
                 k: du.tensor_to_rank0(v) if torch.is_tensor(v) else v
                 for k, v in data.items()
             }
-            if du.is_main_process():
+            if is_writer:
                 evaluator.process(gathered_in, gathered_out)
 
     total_time = time.perf_counter() - start_time
@@ -122,8 +144,8 @@ def inference_on_dataset(model, data_loader, evaluator, eval_iter=None,
             f"Total inference time: {datetime.timedelta(seconds=total_time)} "
             f"({total_compute_time:.3f}s compute)"
         )
-    results = evaluator.evaluate() if du.is_main_process() else None
-    results = du.broadcast_py_object(results, src=0)
+    results = evaluator.evaluate() if is_writer else None
+    results = du.broadcast_py_object(results, src=writer_rank)
     return results if results is not None else {}
 
 

@@ -104,6 +104,60 @@ class RobertaForPreTraining(nn.Module):
     def set_activation_checkpoint(self, enabled=True):
         self.roberta.set_activation_checkpoint(enabled)
 
+    # -- pipeline-parallel protocol (reference assigns RoBERTa stage ids in
+    # roberta_model.py; single-tensor boundary like BERT) -------------------
+
+    def _embed(self, b):
+        rb = self.roberta
+        input_ids = b["input_ids"]
+        pos_ids = create_position_ids_from_input_ids(input_ids, rb.pad_token_id)
+        emb = rb.embeddings.vocab_embeddings(input_ids)
+        emb = emb + rb.embeddings.position_embeddings(
+            pos_ids.clamp(max=rb.embeddings.position_embeddings.num_embeddings - 1)
+        )
+        if rb.embeddings.tokentype_embeddings is not None:
+            tt = b.get("tokentype_ids")
+            if tt is None:
+                tt = torch.zeros_like(input_ids)
+            emb = emb + rb.embeddings.tokentype_embeddings(tt)
+        return rb.embeddings.embedding_dropout(emb)
+
+    def pipeline_units(self):
+        from .bert_model import extended_attn_mask
+
+        rb = self.roberta
+        units = [(0, "embeddings", lambda h, b: self._embed(b))]
+        for i, layer in enumerate(rb.layers):
+            units.append(
+                (i, f"layer_{i}",
+                 (lambda lyr: lambda h, b: rb._run_layer(
+                     lyr, h, extended_attn_mask(b.get("attention_mask"))
+                 ))(layer))
+            )
+
+        def head(h, b):
+            h = rb.final_layernorm(h)
+            hh = self.lm_head(h)
+            logits = self.lm_logits(hh, rb.word_embeddings_weight)
+            if b.get("lm_labels") is not None and b.get("loss_mask") is not None:
+                return self.loss_func(logits, b["lm_labels"], b["loss_mask"])
+            return {"prediction_scores": logits}
+
+        units.append((-1, "head", head))
+        return units
+
+    def pipeline_stage_modules(self):
+        rb = self.roberta
+        m = {0: [rb.embeddings]}
+        for i, layer in enumerate(rb.layers):
+            m.setdefault(i, []).append(layer)
+        last = [rb.final_layernorm, self.lm_head, self.lm_logits, self.loss_func,
+                rb.embeddings.vocab_embeddings]
+        if rb.pooler is not None:  # unused by the pretraining loss, but owned
+            last.append(rb.pooler)  # by the last stage so other stages free it
+        m.setdefault(-1, []).extend(last)
+        return m
+
 
 class RobertaForCausalLM(RobertaForPreTraining):
     """RoBERTa as a left-to-right LM (reference: roberta_model.py's CLM

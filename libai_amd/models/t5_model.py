@@ -86,6 +86,7 @@ class T5Model(nn.Module):
         init_method = init_method_normal(initializer_range)
         scaled_init = scaled_init_method_normal(initializer_range, hidden_layers)
         self.hidden_layers = hidden_layers
+        self.hidden_size = hidden_size
 
         self.embedding = T5Embedding(
             vocab_size, hidden_size, max_position_embeddings,
@@ -248,3 +249,102 @@ class T5ForPreTraining(nn.Module):
 
     def set_activation_checkpoint(self, enabled=True):
         self.t5_model.set_activation_checkpoint(enabled)
+
+    # -- pipeline-parallel protocol -----------------------------------------
+    # layer_idx space: encoder 0..n-1 (embedding at 0, enc final LN at n-1),
+    # decoder n..2n-1 (decoder embedding at n), head at -1 — so
+    # cfg.train.dist.pipeline_num_layers must be 2*hidden_layers (reference
+    # t5_model.py:450+ assigns stage ids the same way via OneFlow placements).
+    # The boundary state is (enc_out,) inside/after the encoder and
+    # (enc_out, dec_hidden) between decoder stages — the multi-tensor case
+    # the tuple-boundary scheduler exists for.
+
+    @staticmethod
+    def _enc_mask(b):
+        if "_enc_mask" not in b:
+            b["_enc_mask"] = extended_attn_mask(b.get("encoder_attn_mask"))
+        return b["_enc_mask"]
+
+    @staticmethod
+    def _cross_mask(b):
+        if "_cross_mask" not in b:
+            enc = b.get("encoder_attn_mask")
+            if enc is None:
+                b["_cross_mask"] = None
+            else:
+                dm = b.get("decoder_attn_mask")
+                if dm is None:
+                    dm = torch.ones_like(b["decoder_input_ids"])
+                b["_cross_mask"] = cross_attn_mask(dm, enc)
+        return b["_cross_mask"]
+
+    def pipeline_units(self):
+        t5 = self.t5_model
+        n = t5.hidden_layers
+        units = [
+            (0, "enc_embedding",
+             lambda h, b: t5.embedding(b["encoder_input_ids"]))
+        ]
+        for i, layer in enumerate(t5.encoder_layers):
+            units.append(
+                (i, f"enc_{i}",
+                 (lambda lyr: lambda h, b: t5._run(lyr, h, self._enc_mask(b)))(layer))
+            )
+        units.append(
+            (n - 1, "enc_final_ln", lambda h, b: t5.encoder_final_layernorm(h))
+        )
+        units.append(
+            (n, "dec_embedding",
+             lambda h, b: (h, t5.embedding(b["decoder_input_ids"])))
+        )
+
+        def dec_fn(lyr):
+            def fn(state, b):
+                enc, dec = state
+                dec = t5._run(
+                    lyr, dec, attention_mask=None, encoder_states=enc,
+                    encoder_attention_mask=self._cross_mask(b),
+                )
+                return (enc, dec)
+            return fn
+
+        for i, layer in enumerate(t5.decoder_layers):
+            units.append((n + i, f"dec_{i}", dec_fn(layer)))
+
+        def head(state, b):
+            _, dec = state
+            h = t5.decoder_final_layernorm(dec)
+            logits = t5.lm_head(h, t5.embedding.word_embeddings.weight)
+            if b.get("lm_labels") is not None:
+                return self.loss_func(logits, b["lm_labels"], b["loss_mask"])
+            return {"prediction_scores": logits}
+
+        units.append((-1, "head", head))
+        return units
+
+    def pipeline_stage_modules(self):
+        t5 = self.t5_model
+        n = t5.hidden_layers
+        m = {0: [t5.embedding]}
+        for i, layer in enumerate(t5.encoder_layers):
+            m.setdefault(i, []).append(layer)
+        m.setdefault(n - 1, []).append(t5.encoder_final_layernorm)
+        # the SHARED embedding is owned by the decoder-embedding stage and the
+        # head stage too (tied logits); the engine keeps a replica per owning
+        # stage and all-reduces its grad over the tied group
+        m.setdefault(n, []).append(t5.embedding)
+        for i, layer in enumerate(t5.decoder_layers):
+            m.setdefault(n + i, []).append(layer)
+        m.setdefault(-1, []).extend(
+            [t5.decoder_final_layernorm, t5.lm_head, t5.embedding]
+        )
+        return m
+
+    def pipeline_boundary_shapes(self, batch, first_idx):
+        b, s_enc = batch["encoder_input_ids"].shape
+        s_dec = batch["decoder_input_ids"].shape[1]
+        h = self.t5_model.hidden_size
+        n = self.t5_model.hidden_layers
+        if first_idx == -1 or first_idx > n:
+            return [(b, s_enc, h), (b, s_dec, h)]
+        return [(b, s_enc, h)]

@@ -158,6 +158,20 @@ class PipelineScheduler:
                 dist.all_reduce(p.grad.data, group=group)
 
     # -- p2p ----------------------------------------------------------------
+    #
+    # A pipeline boundary carries a TUPLE of activation tensors (usually one;
+    # enc-dec models like T5 carry (encoder_out, decoder_hidden) between
+    # decoder stages — reference t5_model.py:450+ relies on OneFlow moving
+    # both global tensors).  Single-tensor models keep the legacy
+    # fn(hidden, batch) unit signature; tuple-state models take/return tuples.
+
+    @staticmethod
+    def _to_tuple(x):
+        if x is None:
+            return None
+        if isinstance(x, (tuple, list)):
+            return tuple(x)
+        return (x,)
 
     def _default_activation_shape(self, batch):
         ref = batch.get("input_ids")
@@ -170,10 +184,19 @@ class PipelineScheduler:
             )
         return (*ref.shape[:2], h)
 
-    def _comm(self, send_prev=None, send_next=None, recv_prev_shape=None,
-              recv_next_shape=None, device=None):
+    def boundary_shapes(self, batch):
+        """Shapes of the activation tuple entering this stage's first unit."""
+        if hasattr(self.model, "pipeline_boundary_shapes"):
+            first_idx = self.local_units[0][0]
+            return [tuple(s) for s in
+                    self.model.pipeline_boundary_shapes(batch, first_idx)]
+        return [tuple(self.activation_shape(batch))]
+
+    def _comm(self, send_prev=None, send_next=None, recv_prev_shapes=None,
+              recv_next_shapes=None, device=None):
         """One fused P2P exchange (RCCL group): any subset of
-        {send to prev, send to next, recv from prev, recv from next}.
+        {send to prev, send to next, recv from prev, recv from next}; each
+        side is a LIST of tensors/shapes (the activation tuple).
 
         Fusing the steady-state send_forward+recv_backward (and the mirror
         send_backward+recv_forward) into ONE batch_isend_irecv is what makes
@@ -182,22 +205,24 @@ class PipelineScheduler:
         """
         ops = []
         recv_prev = recv_next = None
-        if recv_prev_shape is not None:
-            recv_prev = torch.empty(recv_prev_shape, dtype=self.dtype, device=device)
-            ops.append(dist.P2POp(dist.irecv, recv_prev, self.dutil.prev_pipeline_rank()))
-        if recv_next_shape is not None:
-            recv_next = torch.empty(recv_next_shape, dtype=self.dtype, device=device)
-            ops.append(dist.P2POp(dist.irecv, recv_next, self.dutil.next_pipeline_rank()))
+        if recv_prev_shapes is not None:
+            recv_prev = [torch.empty(s, dtype=self.dtype, device=device)
+                         for s in recv_prev_shapes]
+            ops += [dist.P2POp(dist.irecv, t, self.dutil.prev_pipeline_rank())
+                    for t in recv_prev]
+        if recv_next_shapes is not None:
+            recv_next = [torch.empty(s, dtype=self.dtype, device=device)
+                         for s in recv_next_shapes]
+            ops += [dist.P2POp(dist.irecv, t, self.dutil.next_pipeline_rank())
+                    for t in recv_next]
         if send_prev is not None:
-            ops.append(
-                dist.P2POp(dist.isend, send_prev.contiguous(),
-                           self.dutil.prev_pipeline_rank())
-            )
+            ops += [dist.P2POp(dist.isend, t.contiguous(),
+                               self.dutil.prev_pipeline_rank())
+                    for t in send_prev]
         if send_next is not None:
-            ops.append(
-                dist.P2POp(dist.isend, send_next.contiguous(),
-                           self.dutil.next_pipeline_rank())
-            )
+            ops += [dist.P2POp(dist.isend, t.contiguous(),
+                               self.dutil.next_pipeline_rank())
+                    for t in send_next]
         if ops:
             for req in dist.batch_isend_irecv(ops):
                 req.wait()
@@ -205,8 +230,11 @@ class PipelineScheduler:
 
     # -- schedule ------------------------------------------------------------
 
-    def _run_units(self, x, batch):
-        h = x
+    def _run_units(self, xs, batch):
+        """xs: tuple of recv'd activations (or None on the first stage).
+        Units take/return a bare tensor when the state is a single tensor
+        (legacy protocol) and a tuple otherwise."""
+        h = xs[0] if (xs is not None and len(xs) == 1) else xs
         for idx, name, fn in self.local_units:
             h = fn(h, batch)
         return h
@@ -215,6 +243,10 @@ class PipelineScheduler:
         losses = {k: v for k, v in out.items() if v.requires_grad}
         total = sum(losses.values()) / num_micro
         total.backward()
+
+    @staticmethod
+    def _grads_of(xs):
+        return [t.grad if t.grad is not None else torch.zeros_like(t) for t in xs]
 
     def run_1f1b(self, micro_batches):
         """Run fwd+bwd over the micro-batches; returns the averaged loss dict
@@ -237,9 +269,12 @@ class PipelineScheduler:
         def recv_fwd(batch):
             if self.is_first:
                 return None
-            x, _ = self._comm(recv_prev_shape=self.activation_shape(batch),
-                              device=device)
-            return x.requires_grad_(True)
+            xs, _ = self._comm(recv_prev_shapes=self.boundary_shapes(batch),
+                               device=device)
+            return tuple(t.requires_grad_(True) for t in xs)
+
+        def detached(out):
+            return [t.detach() for t in self._to_tuple(out)]
 
         # warmup forwards
         for _ in range(num_warmup):
@@ -248,7 +283,7 @@ class PipelineScheduler:
             out = self._run_units(x, batch)
             accumulate_loss(out)
             if not self.is_last:
-                self._comm(send_next=out.detach())
+                self._comm(send_next=detached(out))
             in_flight.append((x, out))
 
         # steady 1F1B
@@ -264,13 +299,16 @@ class PipelineScheduler:
             if self.is_last:
                 grad_out = None
             else:
-                _, grad_out = self._comm(send_next=out.detach(),
-                                         recv_next_shape=out.shape, device=device)
+                out_t = self._to_tuple(out)
+                _, grad_out = self._comm(send_next=detached(out),
+                                         recv_next_shapes=[t.shape for t in out_t],
+                                         device=device)
             bx, bout = in_flight.pop(0)
             if self.is_last:
                 self._loss_backward(bout, m)
             else:
-                torch.autograd.backward(bout, grad_tensors=grad_out)
+                torch.autograd.backward(self._to_tuple(bout),
+                                        grad_tensors=grad_out)
             last_iter = i == num_steady - 1
             if self.is_first:
                 if not last_iter:
@@ -278,15 +316,15 @@ class PipelineScheduler:
                     x_next = None
             else:
                 if last_iter:
-                    self._comm(send_prev=bx.grad)
+                    self._comm(send_prev=self._grads_of(bx))
                 else:
                     x_next_batch = next(it)
                     x_next, _ = self._comm(
-                        send_prev=bx.grad,
-                        recv_prev_shape=self.activation_shape(x_next_batch),
+                        send_prev=self._grads_of(bx),
+                        recv_prev_shapes=self.boundary_shapes(x_next_batch),
                         device=device,
                     )
-                    x_next = x_next.requires_grad_(True)
+                    x_next = tuple(t.requires_grad_(True) for t in x_next)
 
         # cooldown backwards
         while in_flight:
@@ -294,10 +332,12 @@ class PipelineScheduler:
             if self.is_last:
                 self._loss_backward(bout, m)
             else:
-                _, grad_out = self._comm(recv_next_shape=bout.shape, device=device)
-                torch.autograd.backward(bout, grad_tensors=grad_out)
+                bout_t = self._to_tuple(bout)
+                _, grad_out = self._comm(
+                    recv_next_shapes=[t.shape for t in bout_t], device=device)
+                torch.autograd.backward(bout_t, grad_tensors=grad_out)
             if not self.is_first:
-                self._comm(send_prev=bx.grad)
+                self._comm(send_prev=self._grads_of(bx))
 
         self.sync_tied_grads()
         return loss_acc if self.is_last else None
@@ -311,9 +351,11 @@ class PipelineScheduler:
         device = du.get_device()
         x = None
         if not self.is_first:
-            x, _ = self._comm(recv_prev_shape=self.activation_shape(batch), device=device)
+            x, _ = self._comm(recv_prev_shapes=self.boundary_shapes(batch),
+                              device=device)
+            x = tuple(x)
         h = self._run_units(x, batch)
         if not self.is_last:
-            self._comm(send_next=h)
+            self._comm(send_next=list(self._to_tuple(h)))
             return None
         return h
