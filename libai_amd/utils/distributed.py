@@ -106,6 +106,11 @@ class _DistributeUtil:
         world = _env_int("WORLD_SIZE", 1)
         if world <= 1:
             return  # single-process mode: no groups needed
+        # Defensive RCCL launch env: surface collective failures as errors
+        # instead of silent hangs (a single wedged rank otherwise stalls the
+        # whole job until the 15-min timeout), and keep dmabuf IPC on ROCm.
+        os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
+        os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
         backend = "nccl" if torch.cuda.is_available() else "gloo"
         if torch.cuda.is_available():
             torch.cuda.set_device(_env_int("LOCAL_RANK", 0))

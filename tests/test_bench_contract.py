@@ -41,3 +41,26 @@ def test_bench_two_rank_json_contract():
     assert d["config"]["parallelism"] == "dp2"
     assert d["config"]["global_batch"] == 4  # micro 2 x dp 2
     assert d["value"] > 0
+
+
+def test_bench_eight_rank_json_contract():
+    """8-rank torchrun launch (the driver's SCALE run shape) on gloo CPU."""
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29783", "bench.py", "--gpus", "8", "--steps", "1",
+         "--warmup", "0", "--micro-batch", "1", "--layers", "2",
+         "--seq-len", "64", "--hidden", "64", "--heads", "4",
+         "--vocab", "512"],
+        capture_output=True, text=True, cwd=REPO, env=env, timeout=900,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 8
+    assert d["config"]["parallelism"] == "dp8"
+    assert d["config"]["global_batch"] == 8
+    assert d["value"] > 0

@@ -8,6 +8,9 @@ CONFIG=$2
 GPUS=${3:-1}
 shift 3 || shift 2
 export HSA_ENABLE_IPC_MODE_LEGACY=${HSA_ENABLE_IPC_MODE_LEGACY:-0}
+# defensive RCCL env: fail fast on wedged collectives, log comm setup issues
+export TORCH_NCCL_ASYNC_ERROR_HANDLING=${TORCH_NCCL_ASYNC_ERROR_HANDLING:-1}
+export NCCL_DEBUG=${NCCL_DEBUG:-WARN}
 NNODES=${NNODES:-1}
 NODE_RANK=${NODE_RANK:-0}
 MASTER_ADDR=${MASTER_ADDR:-127.0.0.1}
