@@ -7,5 +7,8 @@ from .t5_pretrain import dataloader, model, optim, train  # noqa: F401
 # instance, so the common-file `cfg` alias would be a different object)
 model.cfg.mlp_type = "gated"
 model.cfg.activation = "gelu"
+# MT5/T5.1.1 uses bucketed relative-position biases instead of absolute
+# positions (reference projects/MT5/layers/attention_layer.py:118-123)
+model.cfg.relative_attention = True
 
 train.update(output_dir="./output/mt5_pretrain")

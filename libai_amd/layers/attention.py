@@ -99,8 +99,15 @@ class MultiheadAttention(nn.Module):
         past_key_value=None,
         use_cache=False,
         residual=None,
+        position_bias=None,
     ):
         """attention_mask: [b, sq, sk] bool/uint8, 1 = MASKED (reference semantics).
+
+        ``position_bias``: optional additive score bias broadcastable to
+        [b, nh_local, sq, sk] — ALiBi slopes (BLOOM, reference
+        projects/BLOOM/modeling/) or T5 relative-position bias (reference
+        projects/MT5/layers/attention_layer.py:118-123).  Added AFTER the
+        1/sqrt(hs) scaling, matching both conventions.
 
         If ``residual`` is given, returns hidden + dropout(out + bias) (the
         fused TransformerLayer path); otherwise applies bias+dropout only.
@@ -111,6 +118,7 @@ class MultiheadAttention(nn.Module):
             not self.is_cross_attention
             and past_key_value is None
             and not use_cache
+            and position_bias is None
         ):
             from ..ops.attention import (
                 flash_attention_available,
@@ -161,6 +169,11 @@ class MultiheadAttention(nn.Module):
         # fold the whole 1/sqrt(hs) (incl. coeff round-trip = identity) into the
         # fused softmax's scale argument.
         scale = self.norm_factor * (self.coeff if self.coeff else 1.0)
+        if position_bias is not None:
+            # additive score bias (ALiBi / T5 relative positions): fold the
+            # scale here so the fused softmax runs with scale=1
+            scores = scores * scale + position_bias
+            scale = 1.0
         causal = self.attn_mask_type == AttnMaskType.causal and past_key_value is None
         probs = fused_scale_mask_softmax(
             scores,

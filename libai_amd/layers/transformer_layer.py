@@ -106,6 +106,7 @@ class TransformerLayer(nn.Module):
         encoder_attention_mask=None,
         past_key_value=None,
         use_cache=False,
+        position_bias=None,
     ):
         if past_key_value is not None:
             if self.is_decoder:
@@ -119,7 +120,7 @@ class TransformerLayer(nn.Module):
         residual = ln1 if self.apply_residual_post_layernorm else hidden_states
         attn_out = self.self_attention(
             ln1, attention_mask=attention_mask, past_key_value=self_past,
-            use_cache=use_cache, residual=residual,
+            use_cache=use_cache, residual=residual, position_bias=position_bias,
         )
         if use_cache:
             attn_out, self_present = attn_out

@@ -1,7 +1,9 @@
 from .bert_model import BertForPreTraining, BertModel
+from .bloom import BloomForCausalLM, BloomModel
 from .build import build_model
 from .gpt_model import GPTForPreTraining, GPTModel
 from .llama import LlamaForCausalLM, LlamaModel
+from .mae import MAEForPreTraining
 from .resmlp import ResMLP
 from .roberta_model import RobertaForCausalLM, RobertaForPreTraining, RobertaModel
 from .swin_transformer import SwinTransformer
@@ -25,5 +27,8 @@ __all__ = [
     "SwinTransformer",
     "SwinTransformerV2",
     "ResMLP",
+    "BloomModel",
+    "BloomForCausalLM",
+    "MAEForPreTraining",
     "build_model",
 ]

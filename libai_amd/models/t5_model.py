@@ -36,14 +36,18 @@ def cross_attn_mask(dec_mask, enc_mask):
 
 class T5Embedding(nn.Module):
     def __init__(self, vocab_size, hidden_size, max_position_embeddings,
-                 embedding_dropout_prob, init_method, *, layer_idx=0):
+                 embedding_dropout_prob, init_method, *, layer_idx=0,
+                 use_position_embeddings=True):
         super().__init__()
         self.word_embeddings = VocabEmbedding(vocab_size, hidden_size,
                                               init_method=init_method,
                                               layer_idx=layer_idx)
-        self.position_embeddings = Embedding(max_position_embeddings, hidden_size,
-                                             init_method=init_method,
-                                             layer_idx=layer_idx)
+        # T5 with relative attention biases has NO absolute positions
+        self.position_embeddings = (
+            Embedding(max_position_embeddings, hidden_size,
+                      init_method=init_method, layer_idx=layer_idx)
+            if use_position_embeddings else None
+        )
         self.dropout = nn.Dropout(embedding_dropout_prob)
         self.register_buffer(
             "position_ids", torch.arange(max_position_embeddings).unsqueeze(0),
@@ -51,11 +55,12 @@ class T5Embedding(nn.Module):
         )
 
     def forward(self, input_ids, past_length=0):
-        s = input_ids.size(1)
-        pos = self.position_ids[:, past_length : past_length + s]
-        return self.dropout(
-            self.word_embeddings(input_ids) + self.position_embeddings(pos)
-        )
+        emb = self.word_embeddings(input_ids)
+        if self.position_embeddings is not None:
+            s = input_ids.size(1)
+            pos = self.position_ids[:, past_length : past_length + s]
+            emb = emb + self.position_embeddings(pos)
+        return self.dropout(emb)
 
 
 class T5Model(nn.Module):
@@ -81,6 +86,9 @@ class T5Model(nn.Module):
         amp_enabled=False,
         mlp_type="dense",
         activation="gelu",
+        relative_attention=False,
+        relative_attention_num_buckets=32,
+        relative_attention_max_distance=128,
     ):
         super().__init__()
         init_method = init_method_normal(initializer_range)
@@ -91,7 +99,25 @@ class T5Model(nn.Module):
         self.embedding = T5Embedding(
             vocab_size, hidden_size, max_position_embeddings,
             embedding_dropout_prob, init_method, layer_idx=0,
+            use_position_embeddings=not relative_attention,
         )
+        # MT5/T5.1.1 relative-position bias (reference
+        # projects/MT5/layers/attention_layer.py:118-123): one learned
+        # bucketed bias per stack, shared by every layer in it
+        self.enc_rel_bias = self.dec_rel_bias = None
+        if relative_attention:
+            from ..layers.position_bias import T5RelativePositionBias
+
+            self.enc_rel_bias = T5RelativePositionBias(
+                num_attention_heads, relative_attention_num_buckets,
+                relative_attention_max_distance, bidirectional=True,
+                init_method=init_method, layer_idx=0,
+            )
+            self.dec_rel_bias = T5RelativePositionBias(
+                num_attention_heads, relative_attention_num_buckets,
+                relative_attention_max_distance, bidirectional=False,
+                init_method=init_method, layer_idx=hidden_layers,
+            )
 
         def make_layer(i, is_decoder):
             return TransformerLayer(
@@ -151,6 +177,11 @@ class T5Model(nn.Module):
             "amp_enabled": cfg.get("amp_enabled", False),
             "mlp_type": cfg.get("mlp_type", "dense"),
             "activation": cfg.get("activation", "gelu"),
+            "relative_attention": cfg.get("relative_attention", False),
+            "relative_attention_num_buckets": cfg.get(
+                "relative_attention_num_buckets", 32),
+            "relative_attention_max_distance": cfg.get(
+                "relative_attention_max_distance", 128),
         }
 
     def _run(self, layer, *args, **kw):
@@ -161,8 +192,10 @@ class T5Model(nn.Module):
     def encode(self, encoder_input_ids, encoder_attn_mask=None):
         mask = extended_attn_mask(encoder_attn_mask)
         h = self.embedding(encoder_input_ids)
+        s = h.size(1)
+        bias = self.enc_rel_bias(s, s, h.device) if self.enc_rel_bias else None
         for layer in self.encoder_layers:
-            h = self._run(layer, h, mask)
+            h = self._run(layer, h, mask, position_bias=bias)
         return self.encoder_final_layernorm(h)
 
     def forward(
@@ -188,6 +221,11 @@ class T5Model(nn.Module):
             past_key_values[0][0][0].shape[2] if past_key_values is not None else 0
         )
         h = self.embedding(decoder_input_ids, past_len)
+        dec_bias = None
+        if self.dec_rel_bias is not None:
+            sk = h.size(1) + past_len
+            dec_bias = self.dec_rel_bias(h.size(1), sk, h.device) if past_len == 0 \
+                else self.dec_rel_bias(sk, sk, h.device)[:, :, past_len:, :]
         presents = [] if use_cache else None
         for i, layer in enumerate(self.decoder_layers):
             past = past_key_values[i] if past_key_values is not None else None
@@ -198,6 +236,7 @@ class T5Model(nn.Module):
                 encoder_attention_mask=encoder_decoder_attn_mask,
                 past_key_value=past,
                 use_cache=use_cache,
+                position_bias=dec_bias,
             )
             if use_cache:
                 h, p = out
@@ -285,11 +324,30 @@ class T5ForPreTraining(nn.Module):
             (0, "enc_embedding",
              lambda h, b: t5.embedding(b["encoder_input_ids"]))
         ]
+        def enc_bias(b, s, device):
+            if t5.enc_rel_bias is None:
+                return None
+            # one bias tensor per micro-batch forward (shared autograd node
+            # across layers, like the eager path)
+            if "_enc_bias" not in b:
+                b["_enc_bias"] = t5.enc_rel_bias(s, s, device)
+            return b["_enc_bias"]
+
+        def dec_bias(b, s, device):
+            if t5.dec_rel_bias is None:
+                return None
+            if "_dec_bias" not in b:
+                b["_dec_bias"] = t5.dec_rel_bias(s, s, device)
+            return b["_dec_bias"]
+
+        def enc_fn(lyr):
+            def fn(h, b):
+                return t5._run(lyr, h, self._enc_mask(b),
+                               position_bias=enc_bias(b, h.size(1), h.device))
+            return fn
+
         for i, layer in enumerate(t5.encoder_layers):
-            units.append(
-                (i, f"enc_{i}",
-                 (lambda lyr: lambda h, b: t5._run(lyr, h, self._enc_mask(b)))(layer))
-            )
+            units.append((i, f"enc_{i}", enc_fn(layer)))
         units.append(
             (n - 1, "enc_final_ln", lambda h, b: t5.encoder_final_layernorm(h))
         )
@@ -304,6 +362,7 @@ class T5ForPreTraining(nn.Module):
                 dec = t5._run(
                     lyr, dec, attention_mask=None, encoder_states=enc,
                     encoder_attention_mask=self._cross_mask(b),
+                    position_bias=dec_bias(b, dec.size(1), dec.device),
                 )
                 return (enc, dec)
             return fn
@@ -328,6 +387,8 @@ class T5ForPreTraining(nn.Module):
         m = {0: [t5.embedding]}
         for i, layer in enumerate(t5.encoder_layers):
             m.setdefault(i, []).append(layer)
+            if t5.enc_rel_bias is not None:  # shared bias: every enc stage owns it
+                m[i].append(t5.enc_rel_bias)
         m.setdefault(n - 1, []).append(t5.encoder_final_layernorm)
         # the SHARED embedding is owned by the decoder-embedding stage and the
         # head stage too (tied logits); the engine keeps a replica per owning
@@ -335,6 +396,8 @@ class T5ForPreTraining(nn.Module):
         m.setdefault(n, []).append(t5.embedding)
         for i, layer in enumerate(t5.decoder_layers):
             m.setdefault(n + i, []).append(layer)
+            if t5.dec_rel_bias is not None:
+                m[n + i].append(t5.dec_rel_bias)
         m.setdefault(-1, []).extend(
             [t5.decoder_final_layernorm, t5.lm_head, t5.embedding]
         )

@@ -126,3 +126,47 @@ def test_t5_gated_gpu_bf16():
         lm_labels=torch.randint(0, 1024, (b, 32), device="cuda"),
         loss_mask=torch.ones(b, 32, dtype=torch.long, device="cuda"),
     ))
+
+
+def test_bloom_alibi_gpu_bf16():
+    from libai_amd.models import BloomForCausalLM
+
+    torch.manual_seed(0)
+    m = BloomForCausalLM(vocab_size=1024, hidden_size=256, hidden_layers=2,
+                         num_attention_heads=4).to(torch.bfloat16).cuda()
+    ids = torch.randint(0, 1024, (2, 65), device="cuda")
+    _step(m, dict(input_ids=ids[:, :-1], labels=ids[:, 1:]))
+
+
+def test_mae_gpu_bf16():
+    from libai_amd.models import MAEForPreTraining
+
+    torch.manual_seed(0)
+    m = MAEForPreTraining(img_size=64, patch_size=16, embed_dim=256, depth=2,
+                          num_heads=4, decoder_embed_dim=128, decoder_depth=1,
+                          decoder_num_heads=4).to(torch.bfloat16).cuda()
+    _step(m, dict(images=torch.randn(2, 3, 64, 64, device="cuda",
+                                     dtype=torch.bfloat16)))
+
+
+def test_t5_relative_position_bias_gpu_bf16():
+    from libai_amd.models import T5ForPreTraining
+
+    torch.manual_seed(0)
+    m = T5ForPreTraining(
+        vocab_size=1024, hidden_size=256, hidden_layers=2, num_attention_heads=4,
+        intermediate_size=512, max_position_embeddings=128,
+        relative_attention=True,
+    ).to(torch.bfloat16).cuda()
+    b = 2
+    _step(m, dict(
+        encoder_input_ids=torch.randint(0, 1024, (b, 64), device="cuda"),
+        decoder_input_ids=torch.randint(0, 1024, (b, 32), device="cuda"),
+        encoder_attn_mask=torch.ones(b, 64, dtype=torch.uint8, device="cuda"),
+        lm_labels=torch.randint(0, 1024, (b, 32), device="cuda"),
+        loss_mask=torch.ones(b, 32, dtype=torch.long, device="cuda"),
+    ))
+    # relative attention: no absolute positions, bias params got grads
+    assert m.t5_model.embedding.position_embeddings is None
+    assert m.t5_model.enc_rel_bias.weight.grad is not None
+    assert m.t5_model.dec_rel_bias.weight.grad is not None
