@@ -32,6 +32,15 @@ bool is_bf16(const torch::Tensor& t) {
 
 }  // namespace
 
+// hipBLASLt epilogue-fused GEMMs (csrc/gemm_lt.cpp)
+std::tuple<torch::Tensor, torch::Tensor> lt_gelu_aux_bias(torch::Tensor x,
+                                                          torch::Tensor w,
+                                                          torch::Tensor bias);
+std::tuple<torch::Tensor, torch::Tensor> lt_dgelu_bgrad(torch::Tensor dy,
+                                                        torch::Tensor w2,
+                                                        torch::Tensor aux);
+bool lt_epilogues_available();
+
 // ---------------------------------------------------------------------------
 // extern "C" launchers from csrc/kernels/*.hip
 // ---------------------------------------------------------------------------
@@ -542,6 +551,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_dropout_res_bwd", &bias_dropout_res_bwd);
   m.def("softmax_fwd", &softmax_fwd);
   m.def("softmax_bwd", &softmax_bwd);
+  m.def("lt_gelu_aux_bias", &lt_gelu_aux_bias);
+  m.def("lt_dgelu_bgrad", &lt_dgelu_bgrad);
+  m.def("lt_epilogues_available", &lt_epilogues_available);
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);
   m.def("flash_fwd", &flash_fwd);

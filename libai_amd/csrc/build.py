@@ -67,27 +67,28 @@ def build(force=False, verbose=True):
             ]
         )
 
-    bind_src = os.path.join(CSRC, "bindings.cpp")
-    bind_obj = os.path.join(BUILD, "bindings.o")
     torch_includes = cpp_ext.include_paths()
     py_include = sysconfig.get_paths()["include"]
-    if force or not _newer(bind_obj, bind_src):
-        cmd = [
-            "g++",
-            "-O2",
-            "-std=c++17",
-            "-fPIC",
-            "-D__HIP_PLATFORM_AMD__=1",
-            "-DUSE_ROCM=1",
-            "-DTORCH_EXTENSION_NAME=_C",
-            "-DTORCH_API_INCLUDE_EXTENSION_H",
-        ]
-        for p in torch_includes:
-            cmd += ["-isystem", p]
-        cmd += ["-isystem", "/opt/rocm/include", "-isystem", py_include]
-        cmd += ["-c", bind_src, "-o", bind_obj]
-        _run(cmd)
-    objs.append(bind_obj)
+    for cpp_name in ("bindings.cpp", "gemm_lt.cpp"):
+        cpp_src = os.path.join(CSRC, cpp_name)
+        cpp_obj = os.path.join(BUILD, cpp_name[:-4] + ".o")
+        if force or not _newer(cpp_obj, cpp_src):
+            cmd = [
+                "g++",
+                "-O2",
+                "-std=c++17",
+                "-fPIC",
+                "-D__HIP_PLATFORM_AMD__=1",
+                "-DUSE_ROCM=1",
+                "-DTORCH_EXTENSION_NAME=_C",
+                "-DTORCH_API_INCLUDE_EXTENSION_H",
+            ]
+            for p in torch_includes:
+                cmd += ["-isystem", p]
+            cmd += ["-isystem", "/opt/rocm/include", "-isystem", py_include]
+            cmd += ["-c", cpp_src, "-o", cpp_obj]
+            _run(cmd)
+        objs.append(cpp_obj)
 
     # standalone pybind11 data-helpers extension (no torch dependency)
     dh_src = os.path.join(CSRC, "data_helpers.cpp")
@@ -120,6 +121,7 @@ def build(force=False, verbose=True):
                 "-ltorch_python",
                 "-L/opt/rocm/lib",
                 "-lamdhip64",
+                "-lhipblaslt",
                 "-o",
                 OUT_SO,
             ]

@@ -48,6 +48,25 @@ class MLP(nn.Module):
 
     def forward(self, hidden_states, residual=None):
         if self.bias_gelu_fusion:
+            from ..ops.fused_mlp import fused_mlp, fused_mlp_available
+
+            if fused_mlp_available(hidden_states):
+                # epilogue-fused: gelu(xW1+b1) in the GEMM, dgelu+db1 in the
+                # backward GEMM — no separate elementwise kernels
+                from ..parallel.comm import (
+                    copy_to_tensor_parallel_region,
+                    reduce_from_tensor_parallel_region,
+                )
+
+                x = copy_to_tensor_parallel_region(hidden_states)
+                out = fused_mlp(x, self.dense_h_to_4h.weight,
+                                self.dense_h_to_4h.bias,
+                                self.dense_4h_to_h.weight)
+                out = reduce_from_tensor_parallel_region(out)
+                return bias_dropout_add(
+                    out, bias=self.dense_4h_to_h.bias, residual=residual,
+                    p=self.output_dropout_prob, training=self.training,
+                )
             inter, bias = self.dense_h_to_4h(hidden_states)
             inter = bias_gelu(inter, bias)
         else:

@@ -287,3 +287,69 @@ def test_fused_embedding_vocab_shard_and_padding():
     o = fused_embedding(ids_pad, w, padding_idx=7)
     o.sum().backward()
     assert w.grad.abs().max().item() == 0.0
+
+
+def test_lt_fused_mlp_matches_reference():
+    """hipBLASLt epilogue MLP (gelu-aux fwd, dgelu+bgrad bwd) vs fp32 ref."""
+    from libai_amd.ops.fused_mlp import fused_mlp, fused_mlp_available
+
+    x_probe = torch.randn(2, 2, device="cuda", dtype=torch.bfloat16)
+    if not fused_mlp_available(x_probe):
+        pytest.skip("hipBLASLt epilogues unavailable on this stack")
+
+    torch.manual_seed(0)
+    M, H, F = 512, 256, 1024
+    x = torch.randn(4, M // 4, H, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    w1 = (torch.randn(F, H, device="cuda") * 0.05).to(torch.bfloat16).requires_grad_(True)
+    b1 = (torch.randn(F, device="cuda") * 0.05).to(torch.bfloat16).requires_grad_(True)
+    w2 = (torch.randn(H, F, device="cuda") * 0.05).to(torch.bfloat16).requires_grad_(True)
+
+    y = fused_mlp(x, w1, b1, w2)
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    xr = x.detach().float().requires_grad_(True)
+    w1r = w1.detach().float().requires_grad_(True)
+    b1r = b1.detach().float().requires_grad_(True)
+    w2r = w2.detach().float().requires_grad_(True)
+    # hipBLASLt GELU is the tanh approximation
+    ref = torch.nn.functional.gelu(xr @ w1r.t() + b1r, approximate="tanh") @ w2r.t()
+    ref.backward(g.float())
+
+    assert (y.float() - ref).abs().max().item() < 0.15, \
+        (y.float() - ref).abs().max().item()
+    for got, want, name, tol in (
+        (x.grad, xr.grad, "dx", 0.3),
+        (w1.grad, w1r.grad, "dw1", 0.5),
+        (b1.grad, b1r.grad, "db1", 0.3),
+        (w2.grad, w2r.grad, "dw2", 0.5),
+    ):
+        rel = (got.float() - want).abs().max() / want.abs().max().clamp(min=1e-3)
+        assert rel.item() < tol, f"{name} rel err {rel.item()}"
+
+
+def test_mlp_layer_uses_fused_path_gpu():
+    """The MLP module's fused path trains and matches the unfused path
+    closely (different gelu flavor: tanh vs erf)."""
+    import libai_amd.ops.fused_mlp as fm
+    from libai_amd.layers.mlp import MLP
+
+    if not fm.fused_mlp_available(
+        torch.randn(2, 2, device="cuda", dtype=torch.bfloat16)
+    ):
+        pytest.skip("hipBLASLt epilogues unavailable")
+
+    torch.manual_seed(0)
+    mlp = MLP(256, 1024, output_dropout_prob=0.0).to(torch.bfloat16).cuda()
+    x = torch.randn(2, 64, 256, device="cuda", dtype=torch.bfloat16)
+    res = torch.zeros_like(x)
+    out_fused = mlp(x, residual=res)
+    saved = fm._PROBED
+    try:
+        fm._PROBED = False  # force the unfused erf path
+        out_ref = mlp(x, residual=res)
+    finally:
+        fm._PROBED = saved
+    # tanh-vs-erf gelu differ by <3e-3 on typical activations (bf16 noise)
+    assert (out_fused.float() - out_ref.float()).abs().max().item() < 0.1

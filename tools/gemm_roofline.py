@@ -1,0 +1,109 @@
+#!/usr/bin/env python3
+"""Per-shape GEMM roofline for the GPT-2 345M Linear1D sites vs hipBLASLt.
+
+Measures fwd + both bwd GEMMs for each of the five hot linear shapes
+(qkv, attn-out, h->4h, 4h->h, lm-logits) at the bench micro-batch, plus the
+epilogue-fused MLP pair, and reports TF/s and % of the 2.5 PFLOP/s bf16
+dense peak.  Writes markdown to stdout (redirect into profiles/).
+
+Usage (GPU box): python tools/gemm_roofline.py [--mb 48] [--seq 1024]
+"""
+
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+PEAK_TFLOPS = 2500.0  # MI355X dense bf16 (no sparsity)
+
+
+def bench(fn, iters=20, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def gemm_tf(M, N, K, secs):
+    return 2.0 * M * N * K / secs / 1e12
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--mb", type=int, default=48)
+    p.add_argument("--seq", type=int, default=1024)
+    p.add_argument("--hidden", type=int, default=1024)
+    p.add_argument("--vocab", type=int, default=50304)
+    args = p.parse_args()
+
+    import bench as bench_mod  # noqa: F401  (loads the tunableop table)
+
+    bench_mod._enable_tuned_gemms()
+    torch.cuda.init()
+
+    M = args.mb * args.seq
+    H = args.hidden
+    shapes = [
+        ("qkv (h->3h)", M, 3 * H, H),
+        ("attn-out (h->h)", M, H, H),
+        ("mlp h->4h", M, 4 * H, H),
+        ("mlp 4h->h", M, H, 4 * H),
+        ("lm-logits (h->v)", M, args.vocab, H),
+    ]
+    print(f"# GEMM roofline — GPT-2 345M shapes, mb{args.mb} seq{args.seq} bf16\n")
+    print(f"M = {M} tokens; peak = {PEAK_TFLOPS:.0f} TF/s dense bf16 "
+          "(AMD headline /2, no sparsity)\n")
+    print("| site | GEMM | M x N x K | ms | TF/s | % peak |")
+    print("|---|---|---|---|---|---|")
+    for name, m, n, k in shapes:
+        x = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16)
+        dy = torch.randn(m, n, device="cuda", dtype=torch.bfloat16)
+        t_f = bench(lambda: torch.matmul(x, w.t()))
+        t_dx = bench(lambda: torch.matmul(dy, w))
+        t_dw = bench(lambda: torch.matmul(dy.t(), x))
+        for g, t in (("fwd", t_f), ("bwd dX", t_dx), ("bwd dW", t_dw)):
+            tf = gemm_tf(m, n, k, t)
+            print(f"| {name} | {g} | {m}x{n}x{k} | {t * 1e3:.3f} | "
+                  f"{tf:.0f} | {100 * tf / PEAK_TFLOPS:.1f}% |")
+        del x, w, dy
+
+    # epilogue-fused MLP A/B (bias+gelu in the GEMM vs separate kernels)
+    from libai_amd.ops.fused_bias import bias_gelu
+    from libai_amd.ops.fused_mlp import fused_mlp_available
+    from libai_amd.ops._ext import ext
+    from libai_amd.utils import distributed as du
+
+    du.setup_dist_util({})
+    x = torch.randn(M, H, device="cuda", dtype=torch.bfloat16)
+    w1 = torch.randn(4 * H, H, device="cuda", dtype=torch.bfloat16)
+    b1 = torch.randn(4 * H, device="cuda", dtype=torch.bfloat16)
+    w2 = torch.randn(H, 4 * H, device="cuda", dtype=torch.bfloat16)
+    print("\n## MLP epilogue fusion A/B (fwd path)\n")
+    if fused_mlp_available(x):
+        t_sep = bench(lambda: torch.matmul(bias_gelu(torch.matmul(x, w1.t()), b1),
+                                           w2.t()))
+        t_fus = bench(lambda: torch.matmul(
+            ext().lt_gelu_aux_bias(x, w1, b1)[0], w2.t()))
+        print(f"separate (GEMM + bias_gelu + GEMM): {t_sep * 1e3:.3f} ms")
+        print(f"fused (GELU_AUX_BIAS GEMM + GEMM):  {t_fus * 1e3:.3f} ms "
+              f"({100 * (t_sep - t_fus) / t_sep:+.1f}%)")
+        dy = torch.randn(M, H, device="cuda", dtype=torch.bfloat16)
+        _, aux = ext().lt_gelu_aux_bias(x, w1, b1)
+        t_b = bench(lambda: ext().lt_dgelu_bgrad(dy, w2, aux))
+        print(f"bwd DGELU_BGRAD GEMM: {t_b * 1e3:.3f} ms "
+              f"(vs plain dX GEMM above + separate dgelu + colsum)")
+    else:
+        print("hipBLASLt epilogues unavailable on this stack")
+
+
+if __name__ == "__main__":
+    main()
