@@ -22,3 +22,39 @@ def test_onnx_export_gpt(tmp_path):
     except Exception as e:  # torch.onnx availability varies per build
         pytest.skip(f"torch.onnx export unavailable: {e}")
     assert os.path.exists(path) and os.path.getsize(path) > 1000
+
+
+def test_t5_onnx_export(tmp_path):
+    from libai_amd.models.t5_model import T5ForPreTraining
+    from libai_amd.onnx_export.export import export_t5_onnx_model
+
+    torch.manual_seed(0)
+    m = T5ForPreTraining(vocab_size=64, hidden_size=32, hidden_layers=2,
+                         num_attention_heads=4, intermediate_size=64,
+                         hidden_dropout_prob=0.0,
+                         attention_probs_dropout_prob=0.0,
+                         embedding_dropout_prob=0.0)
+    out = str(tmp_path / "t5.onnx")
+    try:
+        export_t5_onnx_model(m, enc_ids=torch.randint(0, 64, (1, 8)),
+                             dec_ids=torch.randint(0, 64, (1, 6)),
+                             output_path=out, check=False)
+    except Exception as e:  # torch.onnx needs the onnx package in this build
+        pytest.skip(f"torch.onnx export unavailable: {e}")
+    assert os.path.getsize(out) > 1000
+
+
+def test_onnxruntime_verification_if_available(tmp_path):
+    pytest.importorskip("onnxruntime")
+    from libai_amd.models import GPTForPreTraining
+    from libai_amd.onnx_export.export import export_onnx_model, verify_onnx_model
+
+    torch.manual_seed(0)
+    m = GPTForPreTraining(hidden_layers=2, vocab_size=64, hidden_size=32,
+                          ffn_hidden_size=64, num_attention_heads=4,
+                          max_seq_length=32, embedding_dropout_prob=0.0,
+                          attention_dropout_prob=0.0, output_dropout_prob=0.0)
+    out = str(tmp_path / "gpt.onnx")
+    ids = torch.randint(0, 64, (1, 8))
+    export_onnx_model(m, sample_input=ids, output_path=out)
+    assert verify_onnx_model(out, m, {"input_ids": ids})
