@@ -60,6 +60,9 @@ def build(force=False, verbose=True):
                 "-O3",
                 "-std=c++17",
                 "-fPIC",
+            ]
+            + os.environ.get("EXTRA_HIP_FLAGS", "").split()
+            + [
                 "-c",
                 src,
                 "-o",

@@ -40,6 +40,17 @@ typedef float f32x16_t __attribute__((ext_vector_type(16)));
 #define QBLK (QB * NW)
 #define KVB 64   // kv rows per LDS tile (forward)
 
+// A/B experiment knobs (guide T5: s_setprio around MFMA clusters, +4-7% on
+// attention when wave roles diverge; FLASH_STAGGER decorrelates the waves'
+// subtile order so they don't hit the same stall points in lockstep).
+#ifdef FLASH_SETPRIO
+#define PRIO_HI() __builtin_amdgcn_s_setprio(1)
+#define PRIO_LO() __builtin_amdgcn_s_setprio(0)
+#else
+#define PRIO_HI()
+#define PRIO_LO()
+#endif
+
 // native bf16 converts (v_cvt_pk_bf16_f32-class) instead of the bit-manip
 // RNE helper: the repack runs per score element and the manual rounding was
 // ~6 VALU per value.
@@ -250,6 +261,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_fwd_kernel(
 
     // S = K x Q^T : two 32x32 tiles
     f32x16_t s0{}, s1{};
+    PRIO_HI();
 #pragma unroll
     for (int c = 0; c < KC; ++c) {
       bf16x8_t ka = row_img_frag<D>(k_lds, ro + l31, c, hi);
@@ -257,6 +269,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_fwd_kernel(
       bf16x8_t kb = row_img_frag<D>(k_lds, ro + l31 + 32, c, hi);
       s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kb, qfrag[c], s1, 0, 0, 0);
     }
+    PRIO_LO();
 
     // scores are already in the exp2 domain (Q pre-scale); m/l run in it too
     float sv[2][16];
@@ -309,6 +322,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_fwd_kernel(
     }
 
     // PV: O^T[d][q] += V^T x P
+    PRIO_HI();
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
 #pragma unroll
@@ -323,6 +337,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_fwd_kernel(
         }
       }
     }
+    PRIO_LO();
     }  // sub
   }
 
@@ -466,7 +481,14 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
     if (causal && q0 + QTILE - 1 < kv_base) continue;  // entirely above diag
 
 #pragma unroll
-    for (int sub = 0; sub < NSUB; ++sub) {
+    for (int sub_i = 0; sub_i < NSUB; ++sub_i) {
+#ifdef FLASH_STAGGER
+      // per-wave subtile order: decorrelates the waves' stall points
+      // (subtiles are independent; accumulation order is irrelevant)
+      const int sub = (sub_i + wave) % NSUB;
+#else
+      const int sub = sub_i;
+#endif
       const int q0s = q0 + sub * QB;
       if (causal && q0s + QB - 1 < kv_base) continue;
       const int ro = sub * QB;  // row offset inside the staged images
@@ -474,6 +496,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
       // S[q][kv] = Q x K^T  (lane col = kv)
       f32x16_t s{};
       f32x16_t dpd{};
+      PRIO_HI();
 #pragma unroll
       for (int c = 0; c < KC; ++c) {
         bf16x8_t qa = row_img_frag<D>(q_row, ro + l31, c, hi);
@@ -481,6 +504,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
         bf16x8_t da = row_img_frag<D>(do_row, ro + l31, c, hi);
         dpd = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, get_vf(c), dpd, 0, 0, 0);
       }
+      PRIO_LO();
 
       // per-chunk: compute Pd/dS for 8 regs, repack, feed the MFMAs — the
       // short lifetimes keep the kernel at 3 waves/SIMD (full pd[16]/ds[16]
@@ -529,6 +553,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
         bf16x8_t pdf = repack_chunk(pd8);
         bf16x8_t dsf = repack_chunk(ds8);
         const int qc = ro + c16 * 16;
+        PRIO_HI();
 #pragma unroll
         for (int dt = 0; dt < DT; ++dt) {
           bf16x8_t dof = tr_img_frag<D>(do_tr, qc, dt, lane);
@@ -538,6 +563,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
           dk_acc[dt] =
               __builtin_amdgcn_mfma_f32_32x32x16_bf16(qtf, dsf, dk_acc[dt], 0, 0, 0);
         }
+        PRIO_LO();
       }
     }
   }
@@ -651,6 +677,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
       // S^T[kv][q] = K x Q^T ;  dPd^T[kv][q] = V x dO^T   (lane col = q)
       f32x16_t st{};
       f32x16_t dpdt{};
+      PRIO_HI();
 #pragma unroll
       for (int c = 0; c < KC; ++c) {
         bf16x8_t ka = row_img_frag<D>(k_row, ro + l31, c, hi);
@@ -658,6 +685,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
         bf16x8_t va = row_img_frag<D>(v_row, ro + l31, c, hi);
         dpdt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, get_dof(c), dpdt, 0, 0, 0);
       }
+      PRIO_LO();
 
       float ds[16];
       const uint32_t thr8 = drop_threshold_u8(p_drop);
@@ -680,6 +708,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
       }
 
       // dQ^T[d][q] += K^T x dS^T
+      PRIO_HI();
 #pragma unroll
       for (int c16 = 0; c16 < 2; ++c16) {
         bf16x8_t dsf = repack_chunk(&ds[c16 * 8]);
@@ -691,6 +720,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
               __builtin_amdgcn_mfma_f32_32x32x16_bf16(ktf, dsf, dq_acc[dt], 0, 0, 0);
         }
       }
+      PRIO_LO();
     }
   }
 
