@@ -114,6 +114,8 @@ void embedding_bwd_bf16(const int64_t*, const void*, float*, int64_t, int64_t,
                         int64_t, int64_t, int64_t, hipStream_t);
 void embedding_bwd_f32(const int64_t*, const void*, float*, int64_t, int64_t,
                        int64_t, int64_t, int64_t, hipStream_t);
+void gemm_dw_bf16(const void*, const void*, float*, void*, int, int64_t, int64_t,
+                  int64_t, int64_t, hipStream_t);
 void rope_fwd_bf16(const void*, void*, const float*, const float*, int64_t, int64_t,
                    int64_t, int64_t, int64_t, int64_t, int, int, int, int, int,
                    hipStream_t);
@@ -341,6 +343,32 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets, torch::Tensor 
 }
 
 
+
+// ---------------------------------------------------------------------------
+// split-K weight-gradient GEMM: dW[N, K] = dY^T @ X (K1 bwd dW)
+// ---------------------------------------------------------------------------
+torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x, int64_t splits) {
+  CHECK_IN(dy);
+  CHECK_IN(x);
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16 &&
+              x.scalar_type() == torch::kBFloat16, "gemm_dw is bf16-only");
+  const int64_t N = dy.size(-1), K = x.size(-1);
+  const int64_t M = dy.numel() / N;
+  TORCH_CHECK(x.numel() / K == M, "dY/X row mismatch");
+  TORCH_CHECK(N % 128 == 0 && K % 128 == 0, "N and K must be 128-aligned");
+  if (splits <= 0) {  // heuristic: >=512 workgroups to fill 8 XCDs
+    const int64_t tiles = (N / 128) * (K / 128);
+    splits = std::max<int64_t>(1, std::min<int64_t>(16, 512 / tiles));
+  }
+  auto ws = torch::empty({splits, N, K},
+                         dy.options().dtype(torch::kFloat32));
+  auto out = torch::empty({N, K}, dy.options());
+  gemm_dw_bf16(dy.data_ptr(), x.data_ptr(), ws.data_ptr<float>(),
+               out.data_ptr(), 1, M, N, K, splits, cur_stream());
+  check_launch("gemm_dw");
+  return out;
+}
+
 // ---------------------------------------------------------------------------
 // embedding gather/scatter (K10)
 // ---------------------------------------------------------------------------
@@ -554,6 +582,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lt_gelu_aux_bias", &lt_gelu_aux_bias);
   m.def("lt_dgelu_bgrad", &lt_dgelu_bgrad);
   m.def("lt_epilogues_available", &lt_epilogues_available);
+  m.def("gemm_dw", &gemm_dw);
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);
   m.def("flash_fwd", &flash_fwd);
