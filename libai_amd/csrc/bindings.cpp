@@ -92,7 +92,7 @@ void ce_bwd_f32(const void*, const int64_t*, const float*, const float*, const f
 void flash_fwd_bf16(const void*, const void*, const void*, void*, float*,
                     const int*, int64_t, int64_t, int64_t, int64_t, int64_t,
                     int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
-                    int, int, int, int, int, float, float, uint64_t, int,
+                    int, int, int, int, int, float, float, uint64_t, int, int,
                     hipStream_t);
 void flash_bwd_bf16(const void*, const void*, const void*, const void*, const void*,
                     const float*, float*, const int*, void*, void*, void*, int64_t,
@@ -101,7 +101,7 @@ void flash_bwd_bf16(const void*, const void*, const void*, const void*, const vo
                     int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
                     int64_t, int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
                     int64_t, int, int, int, int, int, float, float, uint64_t, int,
-                    hipStream_t);
+                    int, hipStream_t);
 void attn_dropout_apply_bf16(void*, int64_t, int64_t, int64_t, float, uint64_t,
                              hipStream_t);
 void attn_dropout_apply_f32(void*, int64_t, int64_t, int64_t, float, uint64_t,
@@ -437,9 +437,12 @@ std::tuple<torch::Tensor, torch::Tensor> flash_fwd(torch::Tensor q, torch::Tenso
   TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1);
   const int B = (int)q.size(0), Sq = (int)q.size(1), H = (int)q.size(2),
             D = (int)q.size(3);
-  const int Sk = (int)k.size(1);
+  const int Sk = (int)k.size(1), Hkv = (int)k.size(2);
   TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
   TORCH_CHECK(Sk % 8 == 0, "Sk must be a multiple of 8");
+  TORCH_CHECK(Hkv > 0 && H % Hkv == 0 && (int)v.size(2) == Hkv,
+              "GQA head counts: H divisible by Hkv, v matches k");
+  const int kv_group = H / Hkv;
   auto o = torch::empty({B, Sq, H, D}, q.options());
   auto lse = torch::empty({B, H, Sq}, q.options().dtype(torch::kFloat32));
   flash_fwd_bf16(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
@@ -447,7 +450,7 @@ std::tuple<torch::Tensor, torch::Tensor> flash_fwd(torch::Tensor q, torch::Tenso
                  k.stride(0), k.stride(1), k.stride(2), v.stride(0), v.stride(1),
                  v.stride(2), o.stride(0), o.stride(1), o.stride(2), B, H, Sq, Sk, D,
                  (float)scale, (float)p_drop, (uint64_t)seed, causal ? 1 : 0,
-                 cur_stream());
+                 kv_group, cur_stream());
   check_launch("flash_fwd");
   return {o, lse};
 }
@@ -466,19 +469,21 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> flash_bwd(
   }
   const int B = (int)q.size(0), Sq = (int)q.size(1), H = (int)q.size(2),
             D = (int)q.size(3);
-  const int Sk = (int)k.size(1);
+  const int Sk = (int)k.size(1), Hkv = (int)k.size(2);
+  const int kv_group = H / Hkv;
   TORCH_CHECK(dout.is_contiguous() && o.is_contiguous());
   torch::Tensor dq, dk, dv;
   if (dqkv.has_value()) {
     // [B, S, H, 3, D] fused grad buffer: write q/k/v grads in place
+    TORCH_CHECK(kv_group == 1, "packed dqkv implies MHA");
     auto g = dqkv.value();
     dq = g.select(3, 0);
     dk = g.select(3, 1);
     dv = g.select(3, 2);
   } else {
     dq = torch::empty_like(q, q.options());
-    dk = torch::empty({B, Sk, H, D}, q.options());
-    dv = torch::empty({B, Sk, H, D}, q.options());
+    dk = torch::empty({B, Sk, Hkv, D}, q.options());
+    dv = torch::empty({B, Sk, Hkv, D}, q.options());
   }
   auto drow = torch::empty({B, H, Sq}, q.options().dtype(torch::kFloat32));
   flash_bwd_bf16(
@@ -489,7 +494,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> flash_bwd(
       o.stride(2), dout.stride(0), dout.stride(1), dout.stride(2), dq.stride(0),
       dq.stride(1), dq.stride(2), dk.stride(0), dk.stride(1), dk.stride(2),
       dv.stride(0), dv.stride(1), dv.stride(2), B, H, Sq, Sk, D, (float)scale,
-      (float)p_drop, (uint64_t)seed, causal ? 1 : 0, cur_stream());
+      (float)p_drop, (uint64_t)seed, causal ? 1 : 0, kv_group, cur_stream());
   check_launch("flash_bwd");
   return {dq, dk, dv};
 }

@@ -193,7 +193,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_fwd_kernel(
     const int* __restrict__ kv_len, int64_t q_sb, int64_t q_ss, int64_t q_sh,
     int64_t k_sb, int64_t k_ss, int64_t k_sh, int64_t v_sb, int64_t v_ss,
     int64_t v_sh, int64_t o_sb, int64_t o_ss, int64_t o_sh, int H, int Sq, int Sk,
-    float scale, float p_drop, uint64_t seed, int causal) {
+    float scale, float p_drop, uint64_t seed, int causal, int kv_group) {
   constexpr int DT = D / 32;
   constexpr int KC = D / 16;
 
@@ -213,9 +213,10 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_fwd_kernel(
   const int q_block = blockIdx.x * QBLK;
   const int q_base = q_block + wave * QB;
 
+  const int hk = h / kv_group;  // GQA: query head -> its kv head
   const bf16_t* qp = q + b * q_sb + h * q_sh;
-  const bf16_t* kp = k + b * k_sb + h * k_sh;
-  const bf16_t* vp = v + b * v_sb + h * v_sh;
+  const bf16_t* kp = k + b * k_sb + hk * k_sh;
+  const bf16_t* vp = v + b * v_sb + hk * v_sh;
 
   // Q fragments pre-scaled by scale*log2e: scores come out of the MFMA
   // already in the exp2 domain, saving a mul per score element and the
@@ -405,7 +406,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
     int64_t v_ss, int64_t v_sh, int64_t do_sb, int64_t do_ss, int64_t do_sh,
     int64_t dk_sb, int64_t dk_ss, int64_t dk_sh, int64_t dv_sb, int64_t dv_ss,
     int64_t dv_sh, int H, int Sq, int Sk, float scale, float p_drop, uint64_t seed,
-    int causal) {
+    int causal, int kv_group) {
   constexpr int DT = D / 32;
   constexpr int KC = D / 16;
   // D=64: 64-row staging tiles + persistent K/V fragments.  D=128: 32-row
@@ -427,16 +428,17 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
   const int l31 = lane & 31;
   const int hi = lane >> 5;
 
-  const int bh = blockIdx.y;
-  const int b = bh / H, h = bh % H;
+  // GQA: grid.y spans B * H_kv; each kv head's dK/dV accumulates the
+  // contributions of its kv_group query heads (looped below)
+  const int Hkv = H / kv_group;
+  const int bhk = blockIdx.y;
+  const int b = bhk / Hkv, hk = bhk % Hkv;
   const int kv_block = blockIdx.x * QBLK;   // 128 kv rows per block
   const int kv_base = kv_block + wave * QB; // this wave's 32 kv rows
   const int kvg = kv_base + l31;            // lane's kv row
 
-  const bf16_t* qp = q + b * q_sb + h * q_sh;
-  const bf16_t* kp = k + b * k_sb + h * k_sh;
-  const bf16_t* vp = v + b * v_sb + h * v_sh;
-  const bf16_t* dop = dout + b * do_sb + h * do_sh;
+  const bf16_t* kp = k + b * k_sb + hk * k_sh;
+  const bf16_t* vp = v + b * v_sb + hk * v_sh;
 
   // per-lane K and V row fragments (B-operands: lane j = kv).  Persistent in
   // registers for D=64; re-read from global (L2) per use for D=128.
@@ -466,6 +468,11 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
   // break the __syncthreads inside the loop.
   const int n_qtiles = (kv_block < sk_eff) ? CDIV(Sq, QTILE) : 0;
 
+  for (int g = 0; g < kv_group; ++g) {
+  const int h = hk * kv_group + g;
+  const int bh = b * H + h;  // q-head index: lse/drow rows + dropout key
+  const bf16_t* qp = q + b * q_sb + h * q_sh;
+  const bf16_t* dop = dout + b * do_sb + h * do_sh;
   for (int qt = qt_start; qt < n_qtiles; ++qt) {
     const int q0 = qt * QTILE;
     __syncthreads();
@@ -568,10 +575,12 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
     }
   }
 
+  }  // group loop (GQA)
+
   // write dK, dV rows (lane owns kv row kvg; d in 4-element runs)
   if (kvg < Sk) {
-    bf16_t* dkp = dk + b * dk_sb + h * dk_sh + (int64_t)kvg * dk_ss;
-    bf16_t* dvp = dv + b * dv_sb + h * dv_sh + (int64_t)kvg * dv_ss;
+    bf16_t* dkp = dk + b * dk_sb + hk * dk_sh + (int64_t)kvg * dk_ss;
+    bf16_t* dvp = dv + b * dv_sb + hk * dv_sh + (int64_t)kvg * dv_ss;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt)
 #pragma unroll
@@ -601,7 +610,7 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
     int64_t k_ss, int64_t k_sh, int64_t v_sb, int64_t v_ss, int64_t v_sh,
     int64_t do_sb, int64_t do_ss, int64_t do_sh, int64_t dq_sb, int64_t dq_ss,
     int64_t dq_sh, int H, int Sq, int Sk, float scale, float p_drop, uint64_t seed,
-    int causal) {
+    int causal, int kv_group) {
   constexpr int DT = D / 32;
   constexpr int KC = D / 16;
   constexpr int KVTILE = (D == 64) ? 2 * QB : QB;
@@ -623,9 +632,10 @@ __global__ __launch_bounds__(256, 2) void flash_bwd_q_kernel(
   const int q_base = q_block + wave * QB;
   const int qg = q_base + l31;
 
+  const int hk = h / kv_group;  // GQA
   const bf16_t* qp = q + b * q_sb + h * q_sh;
-  const bf16_t* kp = k + b * k_sb + h * k_sh;
-  const bf16_t* vp = v + b * v_sb + h * v_sh;
+  const bf16_t* kp = k + b * k_sb + hk * k_sh;
+  const bf16_t* vp = v + b * v_sb + hk * v_sh;
   const bf16_t* dop = dout + b * do_sb + h * do_sh;
 
   // per-lane Q and dO row fragments (B-operands: lane j = q); persistent for
@@ -766,19 +776,19 @@ extern "C" void flash_fwd_bf16(const void* q, const void* k, const void* v, void
                                int64_t v_ss, int64_t v_sh, int64_t o_sb, int64_t o_ss,
                                int64_t o_sh, int B, int H, int Sq, int Sk, int D,
                                float scale, float p_drop, uint64_t seed, int causal,
-                               hipStream_t stream) {
+                               int kv_group, hipStream_t stream) {
   dim3 grid(CDIV(Sq, QBLK), B * H);
   dim3 block(256);
   if (D == 64)
     flash_fwd_kernel<64><<<grid, block, 0, stream>>>(
         (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (bf16_t*)o, lse,
         kv_len, q_sb, q_ss, q_sh, k_sb, k_ss, k_sh, v_sb, v_ss, v_sh, o_sb, o_ss,
-        o_sh, H, Sq, Sk, scale, p_drop, seed, causal);
+        o_sh, H, Sq, Sk, scale, p_drop, seed, causal, kv_group);
   else if (D == 128)
     flash_fwd_kernel<128><<<grid, block, 0, stream>>>(
         (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (bf16_t*)o, lse,
         kv_len, q_sb, q_ss, q_sh, k_sb, k_ss, k_sh, v_sb, v_ss, v_sh, o_sb, o_ss,
-        o_sh, H, Sq, Sk, scale, p_drop, seed, causal);
+        o_sh, H, Sq, Sk, scale, p_drop, seed, causal, kv_group);
 }
 
 extern "C" void flash_bwd_bf16(
@@ -790,7 +800,8 @@ extern "C" void flash_bwd_bf16(
     int64_t o_sh, int64_t do_sb, int64_t do_ss, int64_t do_sh, int64_t dq_sb,
     int64_t dq_ss, int64_t dq_sh, int64_t dk_sb, int64_t dk_ss, int64_t dk_sh,
     int64_t dv_sb, int64_t dv_ss, int64_t dv_sh, int B, int H, int Sq, int Sk, int D,
-    float scale, float p_drop, uint64_t seed, int causal, hipStream_t stream) {
+    float scale, float p_drop, uint64_t seed, int causal, int kv_group,
+    hipStream_t stream) {
   // Drow = rowsum(dO * O)  (dO and O share layout; use dO's strides for both:
   // the wrapper guarantees o was allocated with the same layout)
   {
@@ -811,21 +822,22 @@ extern "C" void flash_bwd_bf16(
   (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (const bf16_t*)dout, lse,    \
       drow_ws, kv_len, (bf16_t*)dk, (bf16_t*)dv, q_sb, q_ss, q_sh, k_sb, k_ss,       \
       k_sh, v_sb, v_ss, v_sh, do_sb, do_ss, do_sh, dk_sb, dk_ss, dk_sh, dv_sb,       \
-      dv_ss, dv_sh, H, Sq, Sk, scale, p_drop, seed, causal
+      dv_ss, dv_sh, H, Sq, Sk, scale, p_drop, seed, causal, kv_group
 #define BWD_ARGS_Q                                                                   \
   (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v, (const bf16_t*)dout, lse,    \
       drow_ws, kv_len, (bf16_t*)dq, q_sb, q_ss, q_sh, k_sb, k_ss, k_sh, v_sb, v_ss,  \
       v_sh, do_sb, do_ss, do_sh, dq_sb, dq_ss, dq_sh, H, Sq, Sk, scale, p_drop,      \
-      seed, causal
+      seed, causal, kv_group
   dim3 block(256);
+  const int Hkv = H / kv_group;  // bwd_kv grid spans the KV heads
   if (D == 64) {
     flash_bwd_kv_kernel<64>
-        <<<dim3(CDIV(Sk, QBLK), B * H), block, 0, stream>>>(BWD_ARGS_KV);
+        <<<dim3(CDIV(Sk, QBLK), B * Hkv), block, 0, stream>>>(BWD_ARGS_KV);
     flash_bwd_q_kernel<64>
         <<<dim3(CDIV(Sq, QBLK), B * H), block, 0, stream>>>(BWD_ARGS_Q);
   } else if (D == 128) {
     flash_bwd_kv_kernel<128>
-        <<<dim3(CDIV(Sk, QBLK), B * H), block, 0, stream>>>(BWD_ARGS_KV);
+        <<<dim3(CDIV(Sk, QBLK), B * Hkv), block, 0, stream>>>(BWD_ARGS_KV);
     flash_bwd_q_kernel<128>
         <<<dim3(CDIV(Sq, QBLK), B * H), block, 0, stream>>>(BWD_ARGS_Q);
   }

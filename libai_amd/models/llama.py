@@ -24,38 +24,70 @@ __all__ = ["LlamaModel", "LlamaForCausalLM"]
 
 
 class LlamaAttention(nn.Module):
+    """Self-attention with optional grouped-query KV heads (GQA — the
+    Llama-2-70B / Qwen2 / ChatGLM2 sharing scheme; reference capability:
+    projects/Qwen, projects/ChatGLM).  num_key_value_heads == num_heads is
+    plain MHA with the fused qkv projection; fewer kv heads use a separate
+    q projection + fused [k|v] pair, and the flash kernels map each query
+    head to kv head h // (Hq/Hkv) in-kernel."""
+
     def __init__(self, hidden_size, num_heads, max_position_embeddings,
-                 init_method, output_init_method, rope_theta=10000.0, *, layer_idx=0):
+                 init_method, output_init_method, rope_theta=10000.0,
+                 num_key_value_heads=None, *, layer_idx=0):
         super().__init__()
         self.hidden_size = hidden_size
         self.num_heads = num_heads
         self.head_dim = hidden_size // num_heads
+        self.num_kv_heads = num_key_value_heads or num_heads
+        assert num_heads % self.num_kv_heads == 0, (num_heads, self.num_kv_heads)
         dutil = du.get_dist_util()
         tp = dutil.tensor_parallel_size
         # the fused qkv weight is per-head interleaved; whole heads must land
         # on each TP rank for the contiguous column shard to stay coherent
         assert num_heads % tp == 0, (num_heads, tp)
+        assert self.num_kv_heads % tp == 0, (self.num_kv_heads, tp)
         self.num_heads_local = num_heads // tp
+        self.num_kv_local = self.num_kv_heads // tp
         self.max_pos = max_position_embeddings
         self.rope_theta = rope_theta
         self.layer_idx = layer_idx
-        self.query_key_value = Linear1D(hidden_size, 3 * hidden_size, bias=False,
-                                        parallel="col", init_method=init_method,
-                                        layer_idx=layer_idx)
+        if self.num_kv_heads == num_heads:
+            self.query_key_value = Linear1D(hidden_size, 3 * hidden_size,
+                                            bias=False, parallel="col",
+                                            init_method=init_method,
+                                            layer_idx=layer_idx)
+        else:
+            self.q_proj = Linear1D(hidden_size, hidden_size, bias=False,
+                                   parallel="col", init_method=init_method,
+                                   layer_idx=layer_idx)
+            self.kv_proj = Linear1D(
+                hidden_size, 2 * self.num_kv_heads * self.head_dim, bias=False,
+                parallel="col", init_method=init_method, fused_chunks=2,
+                layer_idx=layer_idx)
         self.o_proj = Linear1D(hidden_size, hidden_size, bias=False, parallel="row",
                                init_method=output_init_method, skip_bias_add=True,
                                layer_idx=layer_idx)
         self.scale = 1.0 / math.sqrt(self.head_dim)
 
+    def _project(self, hidden_states):
+        b, s, _ = hidden_states.shape
+        if self.num_kv_heads == self.num_heads:
+            qkv = self.query_key_value(hidden_states)
+            qkv5 = qkv.view(b, s, self.num_heads_local, 3, self.head_dim)
+            return qkv5[..., 0, :], qkv5[..., 1, :], qkv5[..., 2, :]
+        q = self.q_proj(hidden_states).view(b, s, self.num_heads_local,
+                                            self.head_dim)
+        kv = self.kv_proj(hidden_states).view(b, s, 2, self.num_kv_local,
+                                              self.head_dim)
+        return q, kv[:, :, 0], kv[:, :, 1]
+
     def forward(self, hidden_states, past_key_value=None, use_cache=False,
                 residual=None):
         b, s, _ = hidden_states.shape
-        qkv = self.query_key_value(hidden_states)
-        qkv5 = qkv.view(b, s, self.num_heads_local, 3, self.head_dim)
+        q, k, v = self._project(hidden_states)
         pos0 = past_key_value[0].shape[2] if past_key_value is not None else 0
-        q = apply_rotary_pos_emb(qkv5[..., 0, :], self.max_pos, self.rope_theta, pos0)
-        k = apply_rotary_pos_emb(qkv5[..., 1, :], self.max_pos, self.rope_theta, pos0)
-        v = qkv5[..., 2, :]
+        q = apply_rotary_pos_emb(q, self.max_pos, self.rope_theta, pos0)
+        k = apply_rotary_pos_emb(k, self.max_pos, self.rope_theta, pos0)
 
         if (
             past_key_value is None
@@ -67,6 +99,7 @@ class LlamaAttention(nn.Module):
             context = o.reshape(b, s, self.num_heads_local * self.head_dim)
         else:
             # unfused path (decode / CPU): [b, nh, s, hs]
+            group = self.num_heads_local // self.num_kv_local
             qh = q.permute(0, 2, 1, 3)
             kh = k.permute(0, 2, 1, 3)
             vh = v.permute(0, 2, 1, 3)
@@ -74,7 +107,10 @@ class LlamaAttention(nn.Module):
                 pk, pv = past_key_value
                 kh = torch.cat([pk, kh], dim=2)
                 vh = torch.cat([pv, vh], dim=2)
-            present = (kh, vh) if use_cache else None
+            present = (kh, vh) if use_cache else None  # cache holds KV heads
+            if group > 1:
+                kh = kh.repeat_interleave(group, dim=1)
+                vh = vh.repeat_interleave(group, dim=1)
             scores = torch.matmul(qh, kh.transpose(-1, -2))
             causal = past_key_value is None
             probs = fused_scale_mask_softmax(scores, scale=self.scale, causal=causal,
@@ -116,7 +152,8 @@ class LlamaMLP(nn.Module):
 class LlamaDecoderLayer(nn.Module):
     def __init__(self, hidden_size, intermediate_size, num_heads,
                  max_position_embeddings, rms_norm_eps, init_method,
-                 output_init_method, rope_theta=10000.0, *, layer_idx=0):
+                 output_init_method, rope_theta=10000.0,
+                 num_key_value_heads=None, *, layer_idx=0):
         super().__init__()
         self.layer_idx = layer_idx
         self.input_layernorm = RMSLayerNorm(hidden_size, eps=rms_norm_eps,
@@ -124,6 +161,7 @@ class LlamaDecoderLayer(nn.Module):
         self.self_attn = LlamaAttention(hidden_size, num_heads,
                                         max_position_embeddings, init_method,
                                         output_init_method, rope_theta,
+                                        num_key_value_heads,
                                         layer_idx=layer_idx)
         self.post_attention_layernorm = RMSLayerNorm(hidden_size, eps=rms_norm_eps,
                                                      layer_idx=layer_idx)
@@ -159,6 +197,7 @@ class LlamaModel(nn.Module):
         use_scaled_init_for_output_weights=False,
         tie_word_embeddings=False,
         rope_theta=10000.0,
+        num_key_value_heads=None,
         amp_enabled=False,
     ):
         super().__init__()
@@ -175,7 +214,8 @@ class LlamaModel(nn.Module):
                 LlamaDecoderLayer(
                     hidden_size, intermediate_size, num_attention_heads,
                     max_position_embeddings, rms_norm_eps, init_method,
-                    output_init_method, rope_theta, layer_idx=i,
+                    output_init_method, rope_theta, num_key_value_heads,
+                    layer_idx=i,
                 )
                 for i in range(hidden_layers)
             ]
@@ -200,6 +240,7 @@ class LlamaModel(nn.Module):
             ),
             "tie_word_embeddings": cfg.get("tie_word_embeddings", False),
             "rope_theta": cfg.get("rope_theta", 10000.0),
+            "num_key_value_heads": cfg.get("num_key_value_heads", None),
             "amp_enabled": cfg.get("amp_enabled", False),
         }
 

@@ -190,3 +190,26 @@ def test_roberta_causal_lm():
         ids2[:, -1] = 3
         b = m(input_ids=ids2)["prediction_scores"]
     assert torch.allclose(a[:, :-1], b[:, :-1], atol=1e-5)
+
+
+def test_llama_gqa_fwd_bwd_and_decode_parity():
+    """GQA (num_key_value_heads < num_heads): train step + incremental
+    decode == full forward; the KV cache stores only the KV heads."""
+    from libai_amd.models import LlamaForCausalLM
+
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(hidden_layers=2, vocab_size=128, hidden_size=64,
+                         intermediate_size=128, num_attention_heads=8,
+                         num_key_value_heads=2,
+                         max_position_embeddings=64).eval()
+    ids = torch.randint(0, 128, (2, 18))
+    out = m(input_ids=ids[:, :-1], labels=ids[:, 1:])
+    out["lm_loss"].backward()
+    assert torch.isfinite(out["lm_loss"])
+    with torch.no_grad():
+        full = m(input_ids=ids)["prediction_scores"]
+        o = m(input_ids=ids[:, :10], use_cache=True)
+        st = m(input_ids=ids[:, 10:11],
+               past_key_values=o["past_key_values"], use_cache=True)
+    assert (st["prediction_scores"][:, 0] - full[:, 10]).abs().max() < 1e-4
+    assert o["past_key_values"][0][0].shape[1] == 2  # kv heads only
