@@ -43,15 +43,21 @@ def zero_stage_from_config(cfg):
 
 def _default_units(model):
     """One unit per transformer block; everything else (embeddings, head,
-    final norm) belongs to the root unit (-1)."""
-    from ..layers.transformer_layer import TransformerLayer
+    final norm) belongs to the root unit (-1), which stays materialized for
+    the whole forward+backward (safe for direct weight reads like tied
+    lm_head projections).  Models can override via ``zero3_unit_modules()``.
 
-    units = [m for m in model.modules() if isinstance(m, TransformerLayer)]
-    if not units:
-        # non-transformer model: treat every direct child as a unit
-        units = [m for m in model.children()
-                 if sum(p.numel() for p in m.parameters()) > 0]
-    return units
+    Unknown architectures get NO per-layer units — the entire model becomes
+    the root unit (params sharded between steps, full during fwd+bwd),
+    which is always correct; per-layer gather/release is an optimization
+    for the known block classes."""
+    if hasattr(model, "zero3_unit_modules"):
+        return list(model.zero3_unit_modules())
+    from ..layers.transformer_layer import TransformerLayer
+    from ..models.llama import LlamaDecoderLayer
+
+    return [m for m in model.modules()
+            if isinstance(m, (TransformerLayer, LlamaDecoderLayer))]
 
 
 class ZeRO3Manager:
@@ -103,8 +109,11 @@ class ZeRO3Manager:
             if reshard_after_forward:
                 unit.register_forward_hook(self._make_fwd_post(ui))
             unit.register_full_backward_pre_hook(self._make_bwd_pre(ui))
-        # root params live for the whole fwd+bwd (used at both ends)
+        # root params live for the whole fwd+bwd (used at both ends);
+        # inference forwards (no grads to trigger the release-at-reduce)
+        # release them at model-forward exit instead
         model.register_forward_pre_hook(self._root_fwd_pre)
+        model.register_forward_hook(self._root_fwd_post)
 
         for p in model.parameters():
             if p.requires_grad and id(p) in self._param_bucket:
@@ -150,6 +159,11 @@ class ZeRO3Manager:
             b.materialize_params(g)
             if torch.is_grad_enabled() and module.training:
                 b.materialize_grads()
+
+    def _root_fwd_post(self, module, args, output):
+        if not torch.is_grad_enabled():  # eval/no-grad: nothing will reduce
+            for b in self._unit_buckets.get(-1, []):
+                b.release_params()
 
     def _grad_hook(self, p):
         b = self._param_bucket.get(id(p))

@@ -174,3 +174,57 @@ def _zero3_ckpt_worker(rank, world, tmpdir):
 @pytest.mark.timeout(300)
 def test_zero3_checkpoint_roundtrip(tmp_path):
     run_dist(_zero3_ckpt_worker, 2, args=(str(tmp_path),))
+
+
+def _llama_mem_worker(rank, world):
+    """Llama-shaped ZeRO-3 memory accounting: live param bytes at rest are
+    ~1/dp of the replicated model (the VERDICT's Llama-shape memory check,
+    CPU-sized shapes; the sharding arithmetic is size-independent)."""
+    import torch
+
+    from libai_amd.models import LlamaForCausalLM
+    from libai_amd.optim import FusedAdamW
+    from libai_amd.parallel.zero import setup_zero3
+    from libai_amd.utils import distributed as du
+
+    du.setup_dist_util({})
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(hidden_layers=4, vocab_size=1024, hidden_size=512,
+                             intermediate_size=1408, num_attention_heads=8,
+                             max_position_embeddings=128)
+    total = sum(p.numel() for p in model.parameters())
+    opt = FusedAdamW(model.parameters(), lr=1e-3)
+    setup_zero3(model, opt)
+
+    def live_param_elems():
+        n = 0
+        for _, b in opt.buckets:
+            n += b.shard_param.numel()
+            n += b.flat_param.untyped_storage().size() // b.flat_param.element_size()
+        return n
+
+    at_rest = live_param_elems()
+    assert at_rest <= total / world * 1.05, (at_rest, total)
+
+    # one step: grads at rest are sharded too; peak full buffers are freed
+    ids = torch.randint(0, 1024, (2, 33))
+    opt.zero_grad()
+    model(input_ids=ids[:, :-1], labels=ids[:, 1:])["lm_loss"].backward()
+    opt.grad_sync()
+    opt.step()
+    assert live_param_elems() <= total / world * 1.05
+    grad_live = sum(
+        b.shard_grad.numel()
+        + b.flat_grad.untyped_storage().size() // b.flat_grad.element_size()
+        for _, b in opt.buckets
+    )
+    assert grad_live <= total / world * 1.05
+    # optimizer state (masters + moments) is sharded by construction
+    state = sum(b.flat_master.numel() + b.exp_avg.numel() + b.exp_avg_sq.numel()
+                for _, b in opt.buckets)
+    assert state <= 3 * total / world * 1.05
+    return True
+
+
+def test_zero3_llama_shape_memory():
+    run_dist(_llama_mem_worker, 2)
