@@ -259,12 +259,24 @@ class LlamaLoaderHuggerFace(ModelLoaderHuggerFace):
                     qkv.setdefault(i, {})["u"] = v
                 elif rest == "mlp.down_proj.weight":
                     out[base + "mlp.down_proj.weight"] = v
+        nkv = None
+        if cfg is not None:
+            nkv = getattr(cfg, "num_key_value_heads", None)
+            if nkv is None and hasattr(cfg, "get"):
+                nkv = cfg.get("num_key_value_heads", None)
+            nkv = nkv or nh
         for i, d in qkv.items():
             base = f"model.layers.{i}."
             if all(x in d for x in "qkv"):
-                stacked = torch.cat([d["q"], d["k"], d["v"]], dim=0)
-                out[base + "self_attn.query_key_value.weight"] = \
-                    GPT2LoaderHuggerFace._interleave_qkv(stacked, nh)
+                if nkv is not None and nkv != nh:
+                    # GQA (Llama-70B/Qwen2 class): separate q + fused [k|v]
+                    out[base + "self_attn.q_proj.weight"] = d["q"]
+                    out[base + "self_attn.kv_proj.weight"] = \
+                        torch.cat([d["k"], d["v"]], dim=0)
+                else:
+                    stacked = torch.cat([d["q"], d["k"], d["v"]], dim=0)
+                    out[base + "self_attn.query_key_value.weight"] = \
+                        GPT2LoaderHuggerFace._interleave_qkv(stacked, nh)
             if "g" in d and "u" in d:
                 out[base + "mlp.gate_up_proj.weight"] = torch.cat([d["g"], d["u"]], 0)
         return out
