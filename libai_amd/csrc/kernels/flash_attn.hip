@@ -395,8 +395,18 @@ __global__ void rowdot_kernel(const bf16_t* __restrict__ dout,
 // ===========================================================================
 // backward over kv tiles: dK, dV   (block = 128 kv rows, wave owns 32)
 // ===========================================================================
+// FLASH_BWD_DBUF: double-buffered q/do staging (2-phase pipeline) at
+// 2 blocks/CU — next tile's global loads fly during this tile's compute,
+// one barrier per tile instead of two.
+#ifdef FLASH_BWD_DBUF
+#define BWD_KV_NBUF 2
+#define BWD_KV_OCC 2
+#else
+#define BWD_KV_NBUF 1
+#define BWD_KV_OCC 3
+#endif
 template <int D>
-__global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
+__global__ __launch_bounds__(256, D == 64 ? BWD_KV_OCC : 2) void flash_bwd_kv_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
@@ -415,12 +425,12 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
   constexpr int QTILE = (D == 64) ? 2 * QB : QB;
   constexpr int NSUB = QTILE / QB;
 
-  __shared__ __align__(16) bf16_t q_row[QTILE * D];
-  __shared__ __align__(16) bf16_t q_tr[QTILE * D];
-  __shared__ __align__(16) bf16_t do_row[QTILE * D];
-  __shared__ __align__(16) bf16_t do_tr[QTILE * D];
-  __shared__ float lse_lds[QTILE];
-  __shared__ float drow_lds[QTILE];
+  __shared__ __align__(16) bf16_t q_row[BWD_KV_NBUF][QTILE * D];
+  __shared__ __align__(16) bf16_t q_tr[BWD_KV_NBUF][QTILE * D];
+  __shared__ __align__(16) bf16_t do_row[BWD_KV_NBUF][QTILE * D];
+  __shared__ __align__(16) bf16_t do_tr[BWD_KV_NBUF][QTILE * D];
+  __shared__ float lse_lds[BWD_KV_NBUF][QTILE];
+  __shared__ float drow_lds[BWD_KV_NBUF][QTILE];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -473,19 +483,21 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
   const int bh = b * H + h;  // q-head index: lse/drow rows + dropout key
   const bf16_t* qp = q + b * q_sb + h * q_sh;
   const bf16_t* dop = dout + b * do_sb + h * do_sh;
-  for (int qt = qt_start; qt < n_qtiles; ++qt) {
+
+  auto stage_all = [&](int buf, int qt) {
     const int q0 = qt * QTILE;
-    __syncthreads();
-    stage_tile<QTILE, D, true, true>(q_row, q_tr, qp, q0, Sq, q_ss, tid);
-    stage_tile<QTILE, D, true, true>(do_row, do_tr, dop, q0, Sq, do_ss, tid);
+    stage_tile<QTILE, D, true, true>(q_row[buf], q_tr[buf], qp, q0, Sq, q_ss, tid);
+    stage_tile<QTILE, D, true, true>(do_row[buf], do_tr[buf], dop, q0, Sq, do_ss,
+                                     tid);
     if (tid < QTILE) {
       int qr = min(q0 + tid, Sq - 1);
-      lse_lds[tid] = lse[(int64_t)bh * Sq + qr];
-      drow_lds[tid] = drow[(int64_t)bh * Sq + qr];
+      lse_lds[buf][tid] = lse[(int64_t)bh * Sq + qr];
+      drow_lds[buf][tid] = drow[(int64_t)bh * Sq + qr];
     }
-    __syncthreads();
+  };
 
-    if (causal && q0 + QTILE - 1 < kv_base) continue;  // entirely above diag
+  auto compute_tile = [&](int buf, int q0) {
+    if (causal && q0 + QTILE - 1 < kv_base) return;  // entirely above diag
 
 #pragma unroll
     for (int sub_i = 0; sub_i < NSUB; ++sub_i) {
@@ -506,9 +518,9 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
       PRIO_HI();
 #pragma unroll
       for (int c = 0; c < KC; ++c) {
-        bf16x8_t qa = row_img_frag<D>(q_row, ro + l31, c, hi);
+        bf16x8_t qa = row_img_frag<D>(q_row[buf], ro + l31, c, hi);
         s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa, get_kf(c), s, 0, 0, 0);
-        bf16x8_t da = row_img_frag<D>(do_row, ro + l31, c, hi);
+        bf16x8_t da = row_img_frag<D>(do_row[buf], ro + l31, c, hi);
         dpd = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, get_vf(c), dpd, 0, 0, 0);
       }
       PRIO_LO();
@@ -547,13 +559,13 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
             const bool valid =
                 qrow < Sq && kvg < sk_eff && (!causal || qrow >= kvg);
             float p = 0.f;
-            if (valid) p = __expf(s[r] * scale - lse_lds[qrow - q0]);
+            if (valid) p = __expf(s[r] * scale - lse_lds[buf][qrow - q0]);
             float keep = valid ? 1.f : 0.f;
             if (p_drop > 0.f && valid)
               keep = (((hqj[j] >> byte_sh) & 0xFFu) >= thr8) ? ks : 0.f;
             pd8[k] = p * keep;
             ds8[k] = valid
-                         ? scale * p * (dpd[r] * keep - drow_lds[qrow - q0])
+                         ? scale * p * (dpd[r] * keep - drow_lds[buf][qrow - q0])
                          : 0.f;
           }
         }
@@ -563,17 +575,42 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void flash_bwd_kv_kernel(
         PRIO_HI();
 #pragma unroll
         for (int dt = 0; dt < DT; ++dt) {
-          bf16x8_t dof = tr_img_frag<D>(do_tr, qc, dt, lane);
+          bf16x8_t dof = tr_img_frag<D>(do_tr[buf], qc, dt, lane);
           dv_acc[dt] =
               __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, pdf, dv_acc[dt], 0, 0, 0);
-          bf16x8_t qtf = tr_img_frag<D>(q_tr, qc, dt, lane);
+          bf16x8_t qtf = tr_img_frag<D>(q_tr[buf], qc, dt, lane);
           dk_acc[dt] =
               __builtin_amdgcn_mfma_f32_32x32x16_bf16(qtf, dsf, dk_acc[dt], 0, 0, 0);
         }
         PRIO_LO();
       }
     }
+  };  // compute_tile
+
+#ifdef FLASH_BWD_DBUF
+  {
+    int cur = 0;
+    if (qt_start < n_qtiles) {
+      __syncthreads();  // previous g's reads complete before re-staging
+      stage_all(0, qt_start);
+      __syncthreads();
+    }
+    for (int qt = qt_start; qt < n_qtiles; ++qt) {
+      if (qt + 1 < n_qtiles)
+        stage_all(cur ^ 1, qt + 1);  // next tile's loads fly under compute
+      compute_tile(cur, qt * QTILE);
+      __syncthreads();
+      cur ^= 1;
+    }
   }
+#else
+  for (int qt = qt_start; qt < n_qtiles; ++qt) {
+    __syncthreads();
+    stage_all(0, qt);
+    __syncthreads();
+    compute_tile(0, qt * QTILE);
+  }
+#endif
 
   }  // group loop (GQA)
 
