@@ -14,3 +14,7 @@ train.update(
         pipeline_num_layers=gpt_cfg.hidden_layers,
     ),
 )
+
+# Megatron-style sequence parallelism (beyond the reference's feature set):
+# shard the LN/dropout regions along seq over the TP group —
+#   model.cfg.sequence_parallel = True   (requires pipeline_parallel_size=1)

@@ -48,6 +48,7 @@ class MultiheadAttention(nn.Module):
         scale_mask_softmax_fusion=True,
         apply_query_key_layer_scaling=False,
         attn_mask_type=AttnMaskType.padding,
+        sequence_parallel=False,
         *,
         layer_idx=0,
     ):
@@ -65,6 +66,9 @@ class MultiheadAttention(nn.Module):
         self.output_dropout_prob = output_dropout_prob
         self.layer_idx = layer_idx
         self.is_cross_attention = is_cross_attention
+        self.sequence_parallel = sequence_parallel
+        assert not (sequence_parallel and is_cross_attention), \
+            "sequence parallelism is wired for self-attention blocks only"
 
         self.norm_factor = 1.0 / math.sqrt(self.head_size)
         self.coeff = None
@@ -80,10 +84,14 @@ class MultiheadAttention(nn.Module):
                                       init_method=init_method, layer_idx=layer_idx)
         else:
             self.query_key_value = Linear1D(hidden_size, 3 * hidden_size, parallel="col",
-                                            init_method=init_method, layer_idx=layer_idx)
+                                            init_method=init_method,
+                                            sequence_parallel=sequence_parallel,
+                                            layer_idx=layer_idx)
         self.dense = Linear1D(hidden_size, hidden_size, parallel="row",
                               init_method=output_layer_init_method,
-                              skip_bias_add=True, layer_idx=layer_idx)
+                              skip_bias_add=True,
+                              sequence_parallel=sequence_parallel,
+                              layer_idx=layer_idx)
 
     def _split_heads(self, x, n):
         # [b, s, n*hs] -> [b, nh_local, s, hs] per chunk
@@ -127,6 +135,8 @@ class MultiheadAttention(nn.Module):
             )
 
             b, s, _ = hidden_states.shape
+            if self.sequence_parallel:  # input is the seq shard; qkv gathers
+                s = s * du.get_dist_util().tensor_parallel_size
             if flash_attention_available(
                 self.head_size, hidden_states.dtype, hidden_states.device, s, s,
                 attention_mask,

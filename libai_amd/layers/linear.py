@@ -85,6 +85,7 @@ class Linear1D(nn.Module):
         init_method=nn.init.xavier_normal_,
         skip_bias_add=False,
         fused_chunks=1,
+        sequence_parallel=False,
         *,
         layer_idx=0,
         dtype=None,
@@ -95,6 +96,7 @@ class Linear1D(nn.Module):
         self.out_features = out_features
         self.parallel = parallel
         self.skip_bias_add = skip_bias_add
+        self.sequence_parallel = sequence_parallel
         self.layer_idx = layer_idx
 
         dutil = du.get_dist_util()
@@ -128,6 +130,11 @@ class Linear1D(nn.Module):
             init_tp_shard_(self.weight, (out_features, in_features), init_method, 1)
             if bias:
                 self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype))
+                if sequence_parallel:
+                    # bias applies to seq-SHARDED outputs: its grad is a
+                    # partial sum over this rank's tokens -> TP all-reduce
+                    # at grad sync (FusedAdamW._sync_sp_grads)
+                    self.bias.sequence_parallel_grad = True
             else:
                 self.register_parameter("bias", None)
         else:
@@ -140,7 +147,14 @@ class Linear1D(nn.Module):
 
     def forward(self, x):
         if self.parallel == "col":
-            x = copy_to_tensor_parallel_region(x)
+            if self.sequence_parallel:
+                # SP: input arrives seq-sharded; all-gather replaces the
+                # copy_to (its backward reduce-scatters the partial grads)
+                from ..parallel.comm import gather_from_sequence_parallel_region
+
+                x = gather_from_sequence_parallel_region(x)
+            else:
+                x = copy_to_tensor_parallel_region(x)
             if self.skip_bias_add:
                 return F.linear(x, self.weight), self.bias
             # bias fused into the GEMM epilogue (a separate add costs a full
@@ -148,7 +162,15 @@ class Linear1D(nn.Module):
             return F.linear(x, self.weight, self.bias)
         if self.parallel == "row":
             out = F.linear(x, self.weight)
-            out = reduce_from_tensor_parallel_region(out)
+            if self.sequence_parallel:
+                # SP: reduce the TP partial sums AND scatter back to shards
+                from ..parallel.comm import (
+                    reduce_scatter_to_sequence_parallel_region,
+                )
+
+                out = reduce_scatter_to_sequence_parallel_region(out)
+            else:
+                out = reduce_from_tensor_parallel_region(out)
             if self.skip_bias_add:
                 return out, self.bias
             if self.bias is not None:

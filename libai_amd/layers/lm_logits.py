@@ -17,10 +17,12 @@ __all__ = ["LMLogits"]
 
 
 class LMLogits(nn.Module):
-    def __init__(self, vocab_size, bias=False, *, layer_idx=-1):
+    def __init__(self, vocab_size, bias=False, sequence_parallel=False, *,
+                 layer_idx=-1):
         super().__init__()
         self.vocab_size = vocab_size
         self.layer_idx = layer_idx
+        self.sequence_parallel = sequence_parallel
         if bias:
             dutil = du.get_dist_util()
             tp = dutil.tensor_parallel_size
@@ -33,6 +35,13 @@ class LMLogits(nn.Module):
             self.register_parameter("bias", None)
 
     def forward(self, hidden_states, word_embeddings_weight):
-        x = copy_to_tensor_parallel_region(hidden_states)
+        if self.sequence_parallel:
+            # SP: hidden arrives seq-sharded; the all-gather's backward
+            # reduce-scatter replaces the copy_to all-reduce
+            from ..parallel.comm import gather_from_sequence_parallel_region
+
+            x = gather_from_sequence_parallel_region(hidden_states)
+        else:
+            x = copy_to_tensor_parallel_region(hidden_states)
         logits = F.linear(x, word_embeddings_weight, self.bias)
         return logits

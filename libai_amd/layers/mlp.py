@@ -26,6 +26,7 @@ class MLP(nn.Module):
         bias_gelu_fusion=True,
         bias_dropout_fusion=True,
         activation="gelu",
+        sequence_parallel=False,
         *,
         layer_idx=0,
     ):
@@ -33,16 +34,18 @@ class MLP(nn.Module):
         self.output_dropout_prob = output_dropout_prob
         self.layer_idx = layer_idx
         self.bias_gelu_fusion = bias_gelu_fusion and activation == "gelu"
+        self.sequence_parallel = sequence_parallel
         output_layer_init_method = output_layer_init_method or init_method
 
         self.dense_h_to_4h = Linear1D(
             hidden_size, ffn_hidden_size, parallel="col", init_method=init_method,
-            skip_bias_add=self.bias_gelu_fusion, layer_idx=layer_idx,
+            skip_bias_add=self.bias_gelu_fusion,
+            sequence_parallel=sequence_parallel, layer_idx=layer_idx,
         )
         self.dense_4h_to_h = Linear1D(
             ffn_hidden_size, hidden_size, parallel="row",
             init_method=output_layer_init_method, skip_bias_add=True,
-            layer_idx=layer_idx,
+            sequence_parallel=sequence_parallel, layer_idx=layer_idx,
         )
         self.activation_func = None if self.bias_gelu_fusion else build_activation(activation)
 
@@ -55,14 +58,22 @@ class MLP(nn.Module):
                 # backward GEMM — no separate elementwise kernels
                 from ..parallel.comm import (
                     copy_to_tensor_parallel_region,
+                    gather_from_sequence_parallel_region,
                     reduce_from_tensor_parallel_region,
+                    reduce_scatter_to_sequence_parallel_region,
                 )
 
-                x = copy_to_tensor_parallel_region(hidden_states)
+                if self.sequence_parallel:
+                    x = gather_from_sequence_parallel_region(hidden_states)
+                else:
+                    x = copy_to_tensor_parallel_region(hidden_states)
                 out = fused_mlp(x, self.dense_h_to_4h.weight,
                                 self.dense_h_to_4h.bias,
                                 self.dense_4h_to_h.weight)
-                out = reduce_from_tensor_parallel_region(out)
+                if self.sequence_parallel:
+                    out = reduce_scatter_to_sequence_parallel_region(out)
+                else:
+                    out = reduce_from_tensor_parallel_region(out)
                 return bias_dropout_add(
                     out, bias=self.dense_4h_to_h.bias, residual=residual,
                     p=self.output_dropout_prob, training=self.training,

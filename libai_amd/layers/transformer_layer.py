@@ -37,6 +37,7 @@ class TransformerLayer(nn.Module):
         attn_mask_type=AttnMaskType.padding,
         mlp_type="dense",
         activation="gelu",
+        sequence_parallel=False,
         *,
         layer_idx=0,
     ):
@@ -44,7 +45,13 @@ class TransformerLayer(nn.Module):
         self.layer_idx = layer_idx
         self.is_decoder = is_decoder
         self.apply_residual_post_layernorm = apply_residual_post_layernorm
+        self.sequence_parallel = sequence_parallel
+        assert not (sequence_parallel and is_decoder), \
+            "sequence parallelism: encoder/causal self-attention blocks only"
+        assert not (sequence_parallel and mlp_type == "gated"), \
+            "sequence parallelism + gated MLP not wired yet"
         output_layer_init_method = output_layer_init_method or init_method
+        self._sp = sequence_parallel
 
         self.input_layernorm = LayerNorm(hidden_size, eps=layernorm_epsilon,
                                          layer_idx=layer_idx)
@@ -58,6 +65,7 @@ class TransformerLayer(nn.Module):
             scale_mask_softmax_fusion=scale_mask_softmax_fusion,
             apply_query_key_layer_scaling=apply_query_key_layer_scaling,
             attn_mask_type=attn_mask_type,
+            sequence_parallel=sequence_parallel,
             layer_idx=layer_idx,
         )
         self.post_attention_layernorm = LayerNorm(hidden_size, eps=layernorm_epsilon,
@@ -95,8 +103,15 @@ class TransformerLayer(nn.Module):
                 bias_gelu_fusion=bias_gelu_fusion,
                 bias_dropout_fusion=bias_dropout_fusion,
                 activation=activation,
+                sequence_parallel=sequence_parallel,
                 layer_idx=layer_idx,
             )
+        if sequence_parallel:
+            # LayerNorms run on seq shards: their affine grads are partial
+            # sums over local tokens and need a TP all-reduce at grad sync
+            for ln in (self.input_layernorm, self.post_attention_layernorm):
+                for p in ln.parameters():
+                    p.sequence_parallel_grad = True
 
     def forward(
         self,

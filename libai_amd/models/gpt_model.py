@@ -16,6 +16,7 @@ from torch import nn
 from torch.utils.checkpoint import checkpoint as act_checkpoint
 
 from ..config import configurable
+from ..utils import distributed as du
 from ..layers import (
     AttnMaskType,
     Embedding,
@@ -58,7 +59,8 @@ class Transformer(nn.Module):
                  attention_dropout_prob, output_dropout_prob, layernorm_epsilon,
                  init_method, output_layer_init_method, bias_gelu_fusion,
                  bias_dropout_fusion, scale_mask_softmax_fusion,
-                 apply_query_key_layer_scaling, apply_residual_post_layernorm):
+                 apply_query_key_layer_scaling, apply_residual_post_layernorm,
+                 sequence_parallel=False):
         super().__init__()
         self.num_layers = num_layers
         self.checkpoint_activations = False
@@ -77,6 +79,7 @@ class Transformer(nn.Module):
                     apply_query_key_layer_scaling=apply_query_key_layer_scaling,
                     apply_residual_post_layernorm=apply_residual_post_layernorm,
                     attn_mask_type=AttnMaskType.causal,
+                    sequence_parallel=sequence_parallel,
                     layer_idx=i,
                 )
                 for i in range(num_layers)
@@ -124,9 +127,11 @@ class GPTModel(nn.Module):
         scale_mask_softmax_fusion=True,
         apply_query_key_layer_scaling=False,
         apply_residual_post_layernorm=False,
+        sequence_parallel=False,
         amp_enabled=False,
     ):
         super().__init__()
+        self.sequence_parallel = sequence_parallel
         init_method = init_method_normal(initializer_range)
         output_layer_init_method = (
             scaled_init_method_normal(initializer_range, hidden_layers)
@@ -145,8 +150,13 @@ class GPTModel(nn.Module):
             init_method, output_layer_init_method, bias_gelu_fusion,
             bias_dropout_fusion, scale_mask_softmax_fusion,
             apply_query_key_layer_scaling, apply_residual_post_layernorm,
+            sequence_parallel=sequence_parallel,
         )
-        self.lm_head = LMLogits(vocab_size, bias=False, layer_idx=-1)
+        self.lm_head = LMLogits(vocab_size, bias=False,
+                                sequence_parallel=sequence_parallel, layer_idx=-1)
+        if sequence_parallel:
+            for p in self.transformer.layernorm_f.parameters():
+                p.sequence_parallel_grad = True  # runs on the seq shard
         self.hidden_layers = hidden_layers
 
     @classmethod
@@ -175,6 +185,7 @@ class GPTModel(nn.Module):
             "apply_residual_post_layernorm": cfg.get(
                 "apply_residual_post_layernorm", False
             ),
+            "sequence_parallel": cfg.get("sequence_parallel", False),
             "amp_enabled": cfg.get("amp_enabled", False),
         }
 
@@ -183,6 +194,12 @@ class GPTModel(nn.Module):
             past_key_values[0][0].shape[2] if past_key_values is not None else 0
         )
         h = self.embeddings(input_ids, past_length)
+        if self.sequence_parallel and not use_cache and past_key_values is None:
+            # SP region entry: [b, s, h] -> this rank's [b, s/tp, h] shard
+            from ..parallel.comm import scatter_to_sequence_parallel_region
+
+            assert input_ids.shape[1] % du.get_dist_util().tensor_parallel_size == 0
+            h = scatter_to_sequence_parallel_region(h)
         h = self.transformer(h, past_key_values=past_key_values, use_cache=use_cache)
         if use_cache:
             h, presents = h
@@ -238,6 +255,9 @@ class GPTForPreTraining(nn.Module):
     def pipeline_units(self):
         """Ordered (layer_idx, name, fn(hidden, batch)) units for the 1F1B
         engine; first unit ignores ``hidden``, last returns the loss dict."""
+        assert not getattr(self.GPT_model, "sequence_parallel", False), (
+            "sequence parallelism is not composed with pipeline parallelism"
+        )
         units = [
             (0, "embeddings", lambda h, b: self.GPT_model.embeddings(b["input_ids"]))
         ]

@@ -282,12 +282,27 @@ class FusedAdamW(torch.optim.Optimizer):
 
     # -- DP gradient communication (all-reduce / ZeRO reduce-scatter) --------
 
+    def _sync_sp_grads(self, dutil):
+        """Sequence parallelism: LN weights/biases and row-linear biases in
+        the SP region see only this rank's seq-shard of tokens — their grads
+        are PARTIAL and must be summed over TP (Megatron's
+        sequence_parallel param marking)."""
+        if dutil.tensor_parallel_size == 1 or not dist.is_initialized():
+            return
+        for _, b in self.buckets:
+            for p in b.params:
+                if getattr(p, "sequence_parallel_grad", False) and \
+                        p.grad is not None:
+                    dist.all_reduce(p.grad.data,
+                                    group=dutil.tensor_parallel_group)
+
     def grad_sync(self):
         """Average gradients over the DP group; with ZeRO-2, reduce-scatter so
         only the local slice is received.  If comm/compute overlap is active
         (register_overlap_hooks + begin_overlap_step), this just drains the
         in-flight chunk all-reduces and handles stragglers."""
         dutil = du.get_dist_util()
+        self._sync_sp_grads(dutil)  # BEFORE any DP reduce (full views valid)
         dp = dutil.data_parallel_size
         if dp == 1 or not dist.is_initialized():
             return

@@ -74,6 +74,13 @@ class ZeRO3Manager:
             "setup_zero3 must run before the optimizer builds its buckets "
             "(i.e. before the first step/overlap-hook registration)"
         )
+        if any(getattr(p, "sequence_parallel_grad", False)
+               for p in model.parameters()):
+            raise NotImplementedError(
+                "ZeRO-3 is not composed with sequence parallelism: the SP "
+                "grad all-reduce must precede the per-unit reduce-scatter "
+                "that stage 3's hooks run mid-backward"
+            )
         self.model = model
         self.optimizer = optimizer
         self.reshard_after_forward = reshard_after_forward
