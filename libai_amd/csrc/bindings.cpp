@@ -116,6 +116,10 @@ void embedding_bwd_f32(const int64_t*, const void*, float*, int64_t, int64_t,
                        int64_t, int64_t, int64_t, hipStream_t);
 void gemm_dw_bf16(const void*, const void*, float*, void*, int, int64_t, int64_t,
                   int64_t, int64_t, hipStream_t);
+void flash_decode_bf16(const void*, const void*, const void*, void*, const int*,
+                       int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
+                       int64_t, int64_t, int64_t, int64_t, int, int, int, int,
+                       float, int, hipStream_t);
 void rope_fwd_bf16(const void*, void*, const float*, const float*, int64_t, int64_t,
                    int64_t, int64_t, int64_t, int64_t, int, int, int, int, int,
                    hipStream_t);
@@ -369,6 +373,36 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x, int64_t splits) {
   return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// fused single-query decode attention (K16 serving path)
+// ---------------------------------------------------------------------------
+torch::Tensor flash_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                           double scale, c10::optional<torch::Tensor> kv_len) {
+  // q: [B, H, 1, D]; k/v: [B, Hkv, Skv, D] (the decode KV-cache layout)
+  TORCH_CHECK(q.dim() == 4 && q.size(2) == 1, "flash_decode: q must be [B,H,1,D]");
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "flash_decode is bf16-only");
+  TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1);
+  const int B = (int)q.size(0), H = (int)q.size(1), D = (int)q.size(3);
+  const int Skv = (int)k.size(2), Hkv = (int)k.size(1);
+  TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+  TORCH_CHECK(H % Hkv == 0 && (int)v.size(1) == Hkv);
+  const int* kvl = nullptr;
+  if (kv_len.has_value()) {
+    TORCH_CHECK(kv_len->scalar_type() == torch::kInt32 && kv_len->is_cuda() &&
+                kv_len->numel() == B);
+    kvl = kv_len->data_ptr<int>();
+  }
+  auto o = torch::empty({B, H, 1, D}, q.options());
+  flash_decode_bf16(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(), kvl,
+                    q.stride(0), q.stride(1), k.stride(0), k.stride(2),
+                    k.stride(1), v.stride(0), v.stride(2), v.stride(1),
+                    o.stride(0), o.stride(1), B, H, Skv, D, (float)scale,
+                    H / Hkv, cur_stream());
+  check_launch("flash_decode");
+  return o;
+}
+
 // ---------------------------------------------------------------------------
 // embedding gather/scatter (K10)
 // ---------------------------------------------------------------------------
@@ -588,6 +622,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lt_dgelu_bgrad", &lt_dgelu_bgrad);
   m.def("lt_epilogues_available", &lt_epilogues_available);
   m.def("gemm_dw", &gemm_dw);
+  m.def("flash_decode", &flash_decode);
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);
   m.def("flash_fwd", &flash_fwd);

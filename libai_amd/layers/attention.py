@@ -172,6 +172,30 @@ class MultiheadAttention(nn.Module):
                 v = torch.cat([pv, v], dim=2)
         present = (k, v) if use_cache else None
 
+        scale_d = self.norm_factor * (self.coeff if self.coeff else 1.0)
+        if (
+            past_key_value is not None
+            and attention_mask is None
+            and position_bias is None
+            and not self.is_cross_attention
+        ):
+            from ..ops.attention import (
+                decode_attention_available,
+                flash_decode_attn,
+            )
+
+            if decode_attention_available(q, self.head_size):
+                # fused single-token decode (K16): no [b, nh, 1, skv] scores
+                ctx = flash_decode_attn(q.contiguous(), k, v, scale_d)
+                b, nh, _, hs = ctx.shape
+                context = ctx.permute(0, 2, 1, 3).reshape(b, 1, nh * hs)
+                out, bias = self.dense(context)
+                out = bias_dropout_add(
+                    out, bias=bias, residual=residual,
+                    p=self.output_dropout_prob, training=self.training,
+                )
+                return (out, present) if use_cache else out
+
         # [b, nh, sq, hs] x [b, nh, hs, sk] -> [b, nh, sq, sk]  (rocBLAS bmm)
         scores = torch.matmul(q, k.transpose(-1, -2))
         # the reference's query_key_layer_scaling splits the scale between the

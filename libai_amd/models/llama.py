@@ -108,6 +108,21 @@ class LlamaAttention(nn.Module):
                 kh = torch.cat([pk, kh], dim=2)
                 vh = torch.cat([pv, vh], dim=2)
             present = (kh, vh) if use_cache else None  # cache holds KV heads
+            from ..ops.attention import (
+                decode_attention_available,
+                flash_decode_attn,
+            )
+
+            if past_key_value is not None and decode_attention_available(
+                qh, self.head_dim
+            ):
+                # fused single-token decode (K16); GQA mapping in-kernel
+                ctx = flash_decode_attn(qh.contiguous(), kh, vh, self.scale)
+                context = ctx.permute(0, 2, 1, 3).reshape(
+                    b, 1, self.num_heads_local * self.head_dim)
+                out, _ = self.o_proj(context)
+                out = out + residual if residual is not None else out
+                return (out, present) if use_cache else out
             if group > 1:
                 kh = kh.repeat_interleave(group, dim=1)
                 vh = vh.repeat_interleave(group, dim=1)

@@ -105,3 +105,23 @@ def flash_attention(q, k, v, scale, p_drop=0.0, causal=True, training=True,
     """q, k, v: [B, S, H, D] bf16 -> O [B, S, H, D]."""
     return _FlashAttnFn.apply(q, k, v, scale, p_drop if training else 0.0, causal,
                               kv_len)
+
+
+def decode_attention_available(q, head_dim):
+    """Fused single-query decode path (K16): inference, bf16, D in {64,128}."""
+    return (
+        q.device.type == "cuda"
+        and q.dtype == torch.bfloat16
+        and head_dim in (64, 128)
+        and not torch.is_grad_enabled()
+        and q.shape[-2] == 1
+    )
+
+
+def flash_decode_attn(q, k, v, scale, kv_len=None):
+    """q [B, H, 1, D], k/v [B, Hkv, Skv, D] (KV-cache layout) -> [B, H, 1, D].
+
+    One fused kernel: online-softmax q.K^T -> .V, GQA head mapping in-kernel
+    (replaces the decode bmm+softmax+bmm chain; reference capability
+    flow._C.fused_multi_head_attention_inference_v2)."""
+    return ext().flash_decode(q, k, v, scale, kv_len)
