@@ -213,3 +213,53 @@ def test_eval_under_pp2():
     assert results[0], "empty eval results"
     for v in results[0].values():
         assert v == v, "NaN metric"
+
+
+def _bloom_pp2_worker(rank, world):
+    import torch
+
+    from libai_amd.models import BloomForCausalLM
+    from libai_amd.parallel.pipeline import PipelineScheduler
+    from libai_amd.utils import distributed as du
+
+    du.setup_dist_util({"pipeline_parallel_size": 2, "pipeline_num_layers": 4})
+    torch.manual_seed(123)
+    model = BloomForCausalLM(vocab_size=128, hidden_size=32, hidden_layers=4,
+                             num_attention_heads=4)
+    model.hidden_size = 32
+    sched = PipelineScheduler(model, dtype=torch.float32)
+    torch.manual_seed(7)
+    batches = []
+    for _ in range(2):
+        ids = torch.randint(0, 128, (2, 17))
+        batches.append({"input_ids": ids[:, :-1], "labels": ids[:, 1:]})
+    loss_dict = sched.run_1f1b(batches)
+    grads = {n: p.grad.clone() for n, p in model.named_parameters()
+             if p.grad is not None}
+    return (float(loss_dict["lm_loss"]) if loss_dict else None), grads
+
+
+def test_bloom_pp2_matches_single_process():
+    from libai_amd.models import BloomForCausalLM
+    from libai_amd.utils import distributed as du
+
+    du._DIST_UTIL = None
+    du.setup_dist_util({})
+    torch.manual_seed(123)
+    model = BloomForCausalLM(vocab_size=128, hidden_size=32, hidden_layers=4,
+                             num_attention_heads=4)
+    torch.manual_seed(7)
+    losses = []
+    for _ in range(2):
+        ids = torch.randint(0, 128, (2, 17))
+        out = model(input_ids=ids[:, :-1], labels=ids[:, 1:])
+        (out["lm_loss"] / 2).backward()
+        losses.append(float(out["lm_loss"]))
+    ref_grads = {n: p.grad.clone() for n, p in model.named_parameters()
+                 if p.grad is not None}
+    results = run_dist(_bloom_pp2_worker, 2)
+    assert results[1][0] == pytest.approx(sum(losses) / 2, abs=1e-4)
+    for rank, (loss, grads) in enumerate(results):
+        for name, g in grads.items():
+            assert torch.allclose(g, ref_grads[name], atol=1e-4), \
+                f"grad mismatch {name} (rank {rank})"

@@ -1,0 +1,77 @@
+#!/usr/bin/env python3
+"""Serving decode micro-bench: fused flash_decode vs the unfused
+bmm+softmax chain, GPT-2 345M shape, batch x 1-token steps over a warm
+KV cache.
+
+Usage (GPU box): python tools/decode_bench.py [--batch 32] [--ctx 1024]
+"""
+
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--batch", type=int, default=32)
+    p.add_argument("--ctx", type=int, default=1024)
+    p.add_argument("--steps", type=int, default=64)
+    args = p.parse_args()
+
+    import bench as bench_mod
+
+    bench_mod._enable_tuned_gemms()
+    from libai_amd.models import GPTForPreTraining
+    from libai_amd.utils import distributed as du
+    import libai_amd.ops.attention as A
+
+    du.setup_dist_util({})
+    torch.manual_seed(0)
+    m = GPTForPreTraining(
+        hidden_layers=24, vocab_size=50304, hidden_size=1024,
+        ffn_hidden_size=4096, num_attention_heads=16, max_seq_length=2048,
+        embedding_dropout_prob=0.0, attention_dropout_prob=0.0,
+        output_dropout_prob=0.0,
+    ).to(torch.bfloat16).cuda().eval()
+
+    ids = torch.randint(0, 50304, (args.batch, args.ctx), device="cuda")
+
+    def run(use_fused):
+        orig = A.decode_attention_available
+        if not use_fused:
+            A.decode_attention_available = lambda q, hd: False
+        try:
+            with torch.no_grad():
+                out = m(input_ids=ids, use_cache=True)
+                past = out["past_key_values"]
+                tok = ids[:, -1:]
+                for _ in range(4):  # warmup
+                    s = m(input_ids=tok, past_key_values=past, use_cache=True)
+                torch.cuda.synchronize()
+                t0 = time.perf_counter()
+                for _ in range(args.steps):
+                    s = m(input_ids=tok, past_key_values=past, use_cache=True)
+                    tok = s["prediction_scores"][:, -1:].argmax(-1)
+                torch.cuda.synchronize()
+                dt = (time.perf_counter() - t0) / args.steps
+        finally:
+            A.decode_attention_available = orig
+        return dt
+
+    t_unfused = run(False)
+    t_fused = run(True)
+    print(f"# decode bench: GPT-2 345M, batch {args.batch}, ctx {args.ctx}")
+    print(f"unfused bmm+softmax decode: {t_unfused * 1e3:.3f} ms/step "
+          f"({args.batch / t_unfused:.0f} tok/s)")
+    print(f"fused flash_decode:         {t_fused * 1e3:.3f} ms/step "
+          f"({args.batch / t_fused:.0f} tok/s)  "
+          f"[{100 * (t_unfused - t_fused) / t_unfused:+.1f}% step time]")
+
+
+if __name__ == "__main__":
+    main()
