@@ -215,6 +215,9 @@ class EagerTrainer(TrainerBase):
         self._sync_dp_grads()
         self.optimizer.step()
         self.optimizer.zero_grad()
+        if hasattr(self.model, "update_momentum_encoder"):
+            # contrastive SSL models (MoCo v3): EMA the key encoder per step
+            self.model.update_momentum_encoder()
         self.write_metrics(loss_dict, data_time)
 
 

@@ -1,5 +1,6 @@
 from .bert_model import BertForPreTraining, BertModel
 from .bloom import BloomForCausalLM, BloomModel
+from .contrastive import MoCoV3, SimCSEModel
 from .build import build_model
 from .gpt_model import GPTForPreTraining, GPTModel
 from .llama import LlamaForCausalLM, LlamaModel
@@ -30,5 +31,7 @@ __all__ = [
     "BloomModel",
     "BloomForCausalLM",
     "MAEForPreTraining",
+    "SimCSEModel",
+    "MoCoV3",
     "build_model",
 ]
