@@ -194,6 +194,12 @@ class GPTModel(nn.Module):
             past_key_values[0][0].shape[2] if past_key_values is not None else 0
         )
         h = self.embeddings(input_ids, past_length)
+        if self.sequence_parallel and (use_cache or past_key_values is not None):
+            raise RuntimeError(
+                "sequence_parallel is a training-time sharding; build the "
+                "generation model with sequence_parallel=False (checkpoints "
+                "are topology-independent and load either way)"
+            )
         if self.sequence_parallel and not use_cache and past_key_values is None:
             # SP region entry: [b, s, h] -> this rank's [b, s/tp, h] shard
             from ..parallel.comm import scatter_to_sequence_parallel_region

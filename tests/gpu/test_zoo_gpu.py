@@ -170,3 +170,26 @@ def test_t5_relative_position_bias_gpu_bf16():
     assert m.t5_model.embedding.position_embeddings is None
     assert m.t5_model.enc_rel_bias.weight.grad is not None
     assert m.t5_model.dec_rel_bias.weight.grad is not None
+
+
+def test_simcse_gpu_bf16():
+    from libai_amd.models import SimCSEModel
+
+    torch.manual_seed(0)
+    m = SimCSEModel(vocab_size=1024, hidden_size=256, hidden_layers=2,
+                    num_attention_heads=4, intermediate_size=512,
+                    max_position_embeddings=128, hidden_dropout_prob=0.1,
+                    attention_probs_dropout_prob=0.1).to(torch.bfloat16).cuda()
+    ids = torch.randint(0, 1024, (8, 64), device="cuda")
+    _step(m, dict(input_ids=ids))
+
+
+def test_moco_gpu_bf16():
+    from libai_amd.models import MoCoV3
+
+    torch.manual_seed(0)
+    m = MoCoV3(img_size=64, patch_size=16, embed_dim=256, depth=2, num_heads=4,
+               proj_dim=64, proj_hidden=128).to(torch.bfloat16).cuda()
+    imgs = torch.randn(4, 3, 64, 64, device="cuda", dtype=torch.bfloat16)
+    _step(m, dict(images=imgs, images2=imgs.flip(-1)))
+    m.update_momentum_encoder()
