@@ -7,6 +7,7 @@ from .embedding import Embedding, PatchEmbedding, SinePositionalEmbedding, Vocab
 from .layer_norm import LayerNorm, RMSLayerNorm, RMSNorm
 from .linear import Linear, Linear1D
 from .lm_logits import LMLogits
+from .moe import MoELayer
 from .mlp import MLP
 from .transformer_layer import TransformerLayer
 
@@ -29,5 +30,6 @@ __all__ = [
     "Linear1D",
     "LMLogits",
     "MLP",
+    "MoELayer",
     "TransformerLayer",
 ]

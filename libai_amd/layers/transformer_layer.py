@@ -38,6 +38,8 @@ class TransformerLayer(nn.Module):
         mlp_type="dense",
         activation="gelu",
         sequence_parallel=False,
+        moe_num_experts=8,
+        moe_top_k=2,
         *,
         layer_idx=0,
     ):
@@ -83,7 +85,18 @@ class TransformerLayer(nn.Module):
             self.post_cross_attention_layernorm = LayerNorm(
                 hidden_size, eps=layernorm_epsilon, layer_idx=layer_idx
             )
-        if mlp_type == "gated":
+        if mlp_type == "moe":
+            from .moe import MoELayer
+
+            assert not sequence_parallel, "MoE + SP not wired yet"
+            self.mlp = MoELayer(
+                hidden_size, ffn_hidden_size, num_experts=moe_num_experts,
+                top_k=moe_top_k, activation=activation,
+                init_method=init_method,
+                output_layer_init_method=output_layer_init_method,
+                layer_idx=layer_idx,
+            )
+        elif mlp_type == "gated":
             from .mlp import GatedMLP
 
             self.mlp = GatedMLP(

@@ -233,17 +233,27 @@ class FusedAdamW(torch.optim.Optimizer):
                 # stage 3 additionally splits buckets by module UNIT so the
                 # gather/release lifecycle is per-transformer-layer
                 unit = getattr(p, "_zero3_unit", -1) if zero == 3 else -1
-                key = (p.dtype, p.device, _is_tp_sharded(p), unit)
+                ep = bool(getattr(p, "expert_parallel", False))
+                key = (p.dtype, p.device, _is_tp_sharded(p), unit, ep)
                 by_key.setdefault(key, []).append(p)
             wd_on = group["weight_decay"] > 0
-            for (dtype, device, tp_sharded, unit), plist in by_key.items():
-                self._buckets.append(
-                    (gi, _Bucket(plist, dtype, device, wd_on, tp_sharded, dp, dpr,
-                                 zero, unit=unit))
-                )
+            for (dtype, device, tp_sharded, unit, ep), plist in by_key.items():
+                if ep and zero > 0:
+                    raise NotImplementedError(
+                        "ZeRO is not composed with expert-parallel params: "
+                        "the EP group IS the DP group, and each rank's expert "
+                        "state is already unique"
+                    )
+                b = _Bucket(plist, dtype, device, wd_on, tp_sharded, dp, dpr,
+                            zero, unit=unit)
+                b.expert_parallel = ep
+                self._buckets.append((gi, b))
         # DP replicas must start bit-identical; broadcast once from dp rank 0
+        # (expert buckets are INTENTIONALLY different per EP(=DP) rank)
         if dp > 1 and dist.is_initialized():
             for _, b in self._buckets:
+                if getattr(b, "expert_parallel", False):
+                    continue
                 src_rank = dist.get_global_rank(dutil.data_parallel_group, 0)
                 dist.broadcast(b.flat_param, src=src_rank,
                                group=dutil.data_parallel_group)
@@ -321,6 +331,12 @@ class FusedAdamW(torch.optim.Optimizer):
                     b.release_params()
             return
         for _, b in self.buckets:
+            if getattr(b, "expert_parallel", False):
+                # complete already (tokens arrived via all-to-all), but the
+                # per-rank losses are means — match the dense params' DP
+                # average normalization without a collective
+                b.flat_grad.div_(dp)
+                continue
             b.flat_grad.div_(dp)
             if self._zero_eff >= 2:
                 out = b._upd_grad()
@@ -361,6 +377,8 @@ class FusedAdamW(torch.optim.Optimizer):
         self._chunks = []  # (bucket, start, end, param_ids)
         self._param_chunk = {}
         for _, b in self.buckets:
+            if getattr(b, "expert_parallel", False):
+                continue  # no DP collective for expert grads
             off = 0
             cur_params, cur_start = [], 0
             for p in b.params:
@@ -468,14 +486,24 @@ class FusedAdamW(torch.optim.Optimizer):
         total = torch.zeros(1, dtype=torch.float32, device=device)
         use_hip = device.type == "cuda" and has_ext()
         chunk = ext().adamw_chunk_elems() if use_hip else 0
+        total_ep = torch.zeros(1, dtype=torch.float32, device=device)
         for _, b in self.buckets:
             if not b.tp_sharded and dutil.tensor_parallel_rank != 0:
                 continue  # replicated grads counted once per TP group
+            dst = total_ep if getattr(b, "expert_parallel", False) else total
             if use_hip:
-                ext().l2norm_sq(b.norm_desc(chunk), b.dtype == torch.bfloat16, total)
+                ext().l2norm_sq(b.norm_desc(chunk), b.dtype == torch.bfloat16, dst)
             else:
-                total += b._upd_grad().float().pow(2).sum()
+                dst += b._upd_grad().float().pow(2).sum()
         if dist.is_initialized():
+            # structural (rank-symmetric) decision — a value-dependent branch
+            # could diverge across ranks and wedge the collective
+            has_ep = any(getattr(b, "expert_parallel", False)
+                         for _, b in self.buckets)
+            if has_ep and dutil.data_parallel_size > 1:
+                # expert grads are per-EP-rank unique: sum their norms over DP
+                dist.all_reduce(total_ep, group=dutil.data_parallel_group)
+            total += total_ep
             if self._zero_eff > 0 and dutil.data_parallel_size > 1:
                 dist.all_reduce(total, group=dutil.data_parallel_group)
             if dutil.tensor_parallel_size > 1:

@@ -193,3 +193,16 @@ def test_moco_gpu_bf16():
     imgs = torch.randn(4, 3, 64, 64, device="cuda", dtype=torch.bfloat16)
     _step(m, dict(images=imgs, images2=imgs.flip(-1)))
     m.update_momentum_encoder()
+
+
+def test_moe_layer_gpu_bf16():
+    from libai_amd.layers import TransformerLayer
+
+    torch.manual_seed(0)
+    layer = TransformerLayer(256, 512, 4, mlp_type="moe", moe_num_experts=4,
+                             moe_top_k=2).to(torch.bfloat16).cuda()
+    x = torch.randn(2, 64, 256, device="cuda", dtype=torch.bfloat16)
+    y = layer(x)
+    (y.float().pow(2).mean() + layer.mlp.last_aux_loss).backward()
+    assert torch.isfinite(y.float()).all()
+    assert layer.mlp.w1.grad is not None
