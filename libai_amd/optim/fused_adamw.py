@@ -604,6 +604,37 @@ class FusedAdamW(torch.optim.Optimizer):
         return full.reshape(-1), list(full.shape)
 
     @staticmethod
+    def _ep_gather_param_state(p, flat, shape):
+        """Expert-parallel param state -> full expert dim over DP(=EP),
+        making the optimizer checkpoint EP-topology-independent."""
+        if not (getattr(p, "expert_parallel", False) and dist.is_initialized()):
+            return flat, shape
+        dutil = du.get_dist_util()
+        dp = dutil.data_parallel_size
+        if dp == 1:
+            return flat, shape
+        local = flat.view(shape)
+        shards = [torch.empty_like(local) for _ in range(dp)]
+        dist.all_gather(shards, local.contiguous(),
+                        group=dutil.data_parallel_group)
+        full = torch.cat(shards, dim=0)
+        return full.reshape(-1), list(full.shape)
+
+    @staticmethod
+    def _ep_slice_param_state(p, full_flat, full_shape):
+        if not (getattr(p, "expert_parallel", False) and dist.is_initialized()):
+            return full_flat, full_shape
+        dutil = du.get_dist_util()
+        dp = dutil.data_parallel_size
+        if dp == 1 or list(full_shape) == list(p.shape):
+            return full_flat, full_shape
+        if full_shape[0] == p.shape[0] * dp:
+            sl = full_flat.view(full_shape).chunk(dp, dim=0)[
+                dutil.data_parallel_rank]
+            return sl.reshape(-1), list(p.shape)
+        return full_flat, full_shape
+
+    @staticmethod
     def _tp_slice_param_state(p, full_flat, full_shape):
         if not (getattr(p, "tensor_parallel", False) and dist.is_initialized()):
             return full_flat
@@ -627,6 +658,9 @@ class FusedAdamW(torch.optim.Optimizer):
                 mast, full_shape = self._tp_gather_param_state(p, master[off:off + n])
                 ea, _ = self._tp_gather_param_state(p, m[off:off + n])
                 es, _ = self._tp_gather_param_state(p, v[off:off + n])
+                mast, full_shape = self._ep_gather_param_state(p, mast, full_shape)
+                ea, _ = self._ep_gather_param_state(p, ea, list(p.shape))
+                es, _ = self._ep_gather_param_state(p, es, list(p.shape))
                 per_param.append(
                     {
                         "master": mast,
@@ -684,12 +718,15 @@ class FusedAdamW(torch.optim.Optimizer):
                     off += n
                     continue
                 shp = entry.get("shape", list(p.shape))
+                em, shp_e = self._ep_slice_param_state(p, entry["master"], shp)
+                ea2, _ = self._ep_slice_param_state(p, entry["exp_avg"], shp)
+                es2, _ = self._ep_slice_param_state(p, entry["exp_avg_sq"], shp)
                 master[off : off + n].copy_(
-                    self._tp_slice_param_state(p, entry["master"], shp))
+                    self._tp_slice_param_state(p, em, shp_e))
                 m[off : off + n].copy_(
-                    self._tp_slice_param_state(p, entry["exp_avg"], shp))
+                    self._tp_slice_param_state(p, ea2, shp_e))
                 v[off : off + n].copy_(
-                    self._tp_slice_param_state(p, entry["exp_avg_sq"], shp))
+                    self._tp_slice_param_state(p, es2, shp_e))
                 idx += 1
                 off += n
             sl = slice(b.shard_off, b.shard_off + b.shard) if self._zero_eff > 0 \

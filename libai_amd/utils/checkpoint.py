@@ -41,6 +41,15 @@ def _consolidate_model_state(model):
             dist.all_gather(shards, t.contiguous(), group=dutil.tensor_parallel_group)
             out[name] = tp_merge(shards, dim,
                                  getattr(p, "tp_fused_chunks", 1)).cpu()
+        elif p is not None and getattr(p, "expert_parallel", False) \
+                and dutil.data_parallel_size > 1:
+            # expert-parallel params: every EP(=DP) rank owns DIFFERENT
+            # experts — gather the full expert dim so the checkpoint is
+            # EP-topology-independent like the TP shards
+            shards = [torch.empty_like(t) for _ in range(dutil.data_parallel_size)]
+            dist.all_gather(shards, t.contiguous(),
+                            group=dutil.data_parallel_group)
+            out[name] = torch.cat(shards, dim=0).cpu()
         else:
             t = t.detach()
             # clone on CPU: .cpu() would alias the flat bucket storage, which
@@ -65,6 +74,11 @@ def _shard_for_load(model, full_state):
 
             dim = getattr(p, "tp_shard_dim", 0)
             full = tp_slice(full, tp, tpr, dim, getattr(p, "tp_fused_chunks", 1))
+        elif p is not None and getattr(p, "expert_parallel", False) \
+                and full.shape[0] != cur.shape[0]:
+            ep = dutil.data_parallel_size
+            if full.shape[0] == cur.shape[0] * ep:
+                full = full.chunk(ep, dim=0)[dutil.data_parallel_rank]
         if tuple(full.shape) != tuple(cur.shape):
             logging.getLogger(__name__).warning(
                 f"checkpoint key {name}: shape {tuple(full.shape)} != model "

@@ -117,3 +117,52 @@ def test_moe_ep2_training_with_fused_adamw():
     """End-to-end: expert grads skip the DP all-reduce, the rest stays
     DP-synced, grad-norm clip runs without divergent collectives."""
     run_dist(_ep2_train_worker, 2)
+
+
+def _ep2_ckpt_worker(rank, world, tmpdir):
+    import os
+
+    import torch
+
+    from libai_amd.layers import TransformerLayer
+    from libai_amd.optim import FusedAdamW
+    from libai_amd.utils import distributed as du
+    from libai_amd.utils.checkpoint import Checkpointer
+
+    du.setup_dist_util({})
+    torch.manual_seed(0)
+    layer = TransformerLayer(H, FFN, 4, mlp_type="moe", moe_num_experts=E,
+                             moe_top_k=TOPK)
+    opt = FusedAdamW(layer.parameters(), lr=1e-3)
+    opt.set_param_names(layer.named_parameters())
+    torch.manual_seed(50 + rank)
+    for _ in range(2):
+        opt.zero_grad()
+        y = layer(torch.randn(2, 8, H))
+        (y.pow(2).mean() + layer.mlp.last_aux_loss).backward()
+        opt.grad_sync()
+        opt.step()
+    ck = Checkpointer(layer, tmpdir, optimizer=opt)
+    ck.save("moe")
+    my_w1 = layer.mlp.w1.detach().clone()
+
+    # resume into a FRESH ep2 replica: each rank's experts must come back
+    torch.manual_seed(777 + rank)
+    layer2 = TransformerLayer(H, FFN, 4, mlp_type="moe", moe_num_experts=E,
+                              moe_top_k=TOPK)
+    opt2 = FusedAdamW(layer2.parameters(), lr=1e-3)
+    opt2.set_param_names(layer2.named_parameters())
+    ck2 = Checkpointer(layer2, tmpdir, optimizer=opt2)
+    ck2.load(os.path.join(tmpdir, "moe"))
+    assert torch.allclose(layer2.mlp.w1.detach(), my_w1, atol=1e-6), \
+        "expert weights did not round-trip per EP rank"
+    # the saved model.pt holds the FULL expert dim (EP-independent)
+    full = torch.load(os.path.join(tmpdir, "moe", "model.pt"),
+                      map_location="cpu", weights_only=False)
+    assert full["mlp.w1"].shape[0] == E
+    return True
+
+
+@pytest.mark.timeout(300)
+def test_moe_checkpoint_roundtrip_ep2(tmp_path):
+    run_dist(_ep2_ckpt_worker, 2, args=(str(tmp_path),))
