@@ -60,7 +60,7 @@ class Transformer(nn.Module):
                  init_method, output_layer_init_method, bias_gelu_fusion,
                  bias_dropout_fusion, scale_mask_softmax_fusion,
                  apply_query_key_layer_scaling, apply_residual_post_layernorm,
-                 sequence_parallel=False):
+                 sequence_parallel=False, moe_num_experts=0, moe_top_k=2):
         super().__init__()
         self.num_layers = num_layers
         self.checkpoint_activations = False
@@ -80,6 +80,9 @@ class Transformer(nn.Module):
                     apply_residual_post_layernorm=apply_residual_post_layernorm,
                     attn_mask_type=AttnMaskType.causal,
                     sequence_parallel=sequence_parallel,
+                    mlp_type="moe" if moe_num_experts else "dense",
+                    moe_num_experts=moe_num_experts or 8,
+                    moe_top_k=moe_top_k,
                     layer_idx=i,
                 )
                 for i in range(num_layers)
@@ -128,6 +131,8 @@ class GPTModel(nn.Module):
         apply_query_key_layer_scaling=False,
         apply_residual_post_layernorm=False,
         sequence_parallel=False,
+        moe_num_experts=0,
+        moe_top_k=2,
         amp_enabled=False,
     ):
         super().__init__()
@@ -151,6 +156,7 @@ class GPTModel(nn.Module):
             bias_dropout_fusion, scale_mask_softmax_fusion,
             apply_query_key_layer_scaling, apply_residual_post_layernorm,
             sequence_parallel=sequence_parallel,
+            moe_num_experts=moe_num_experts, moe_top_k=moe_top_k,
         )
         self.lm_head = LMLogits(vocab_size, bias=False,
                                 sequence_parallel=sequence_parallel, layer_idx=-1)
@@ -186,6 +192,8 @@ class GPTModel(nn.Module):
                 "apply_residual_post_layernorm", False
             ),
             "sequence_parallel": cfg.get("sequence_parallel", False),
+            "moe_num_experts": cfg.get("moe_num_experts", 0),
+            "moe_top_k": cfg.get("moe_top_k", 2),
             "amp_enabled": cfg.get("amp_enabled", False),
         }
 
@@ -250,7 +258,12 @@ class GPTForPreTraining(nn.Module):
             return {"prediction_scores": out}
         logits = self.GPT_model(input_ids)
         if labels is not None:
-            return self.loss_func(logits, labels)
+            out = self.loss_func(logits, labels)
+            aux = [m.last_aux_loss for m in self.GPT_model.modules()
+                   if getattr(m, "last_aux_loss", None) is not None]
+            if aux:
+                out["moe_aux_loss"] = torch.stack(aux).sum()
+            return out
         return {"prediction_scores": logits}
 
     def set_activation_checkpoint(self, enabled=True):
