@@ -24,7 +24,7 @@ MODEL_KW = dict(
 STEPS = 3
 
 
-def _matrix_worker(rank, world, tp, pp, zero, acc):
+def _matrix_worker(rank, world, tp, pp, zero, acc, moe=0):
     import torch
     import torch.distributed as dist
 
@@ -42,7 +42,7 @@ def _matrix_worker(rank, world, tp, pp, zero, acc):
     assert dutil.data_parallel_size == world // (tp * pp)
 
     torch.manual_seed(123)  # same init everywhere; TP shards slice it
-    model = GPTForPreTraining(**MODEL_KW)
+    model = GPTForPreTraining(**MODEL_KW, moe_num_experts=moe)
     model.hidden_size = MODEL_KW["hidden_size"]
     sched = PipelineScheduler(model, dtype=torch.float32) if pp > 1 else None
     opt = FusedAdamW(model.parameters(), lr=1e-3, weight_decay=0.01,
@@ -74,11 +74,15 @@ def _matrix_worker(rank, world, tp, pp, zero, acc):
 
     assert all(l == l and l < 100 for l in losses), f"bad losses {losses}"
 
-    # DP replicas must hold identical parameters after the steps
+    # DP replicas must hold identical NON-expert parameters after the steps
+    # (expert-parallel params are intentionally different per EP(=DP) rank)
     if dutil.data_parallel_size > 1:
         if zero == 3:  # ZeRO-3 params are released at rest; gather them back
             opt.materialize_all_params()
-        flats = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+        flats = torch.cat([
+            p.detach().reshape(-1) for p in model.parameters()
+            if not getattr(p, "expert_parallel", False)
+        ])
         gathered = [torch.empty_like(flats)
                     for _ in range(dutil.data_parallel_size)]
         dist.all_gather(gathered, flats, group=dutil.data_parallel_group)
@@ -108,6 +112,13 @@ MATRIX_8 = [
 @pytest.mark.parametrize("world,tp,pp,zero,acc", MATRIX_4)
 def test_topology_matrix_4rank(world, tp, pp, zero, acc):
     run_dist(_matrix_worker, world, args=(tp, pp, zero, acc))
+
+
+@pytest.mark.timeout(600)
+def test_topology_moe_dp2xpp2():
+    """Expert parallelism composed with PP: EP == the stage-local DP
+    group; experts dispatch within each stage, dense grads DP-sync."""
+    run_dist(_matrix_worker, 4, args=(1, 2, 0, 2, 4))
 
 
 @pytest.mark.timeout(900)
