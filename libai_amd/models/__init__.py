@@ -1,5 +1,6 @@
 from .bert_model import BertForPreTraining, BertModel
 from .bloom import BloomForCausalLM, BloomModel
+from .clip import CLIPModel
 from .contrastive import MoCoV3, SimCSEModel
 from .build import build_model
 from .gpt_model import GPTForPreTraining, GPTModel
@@ -33,5 +34,6 @@ __all__ = [
     "MAEForPreTraining",
     "SimCSEModel",
     "MoCoV3",
+    "CLIPModel",
     "build_model",
 ]

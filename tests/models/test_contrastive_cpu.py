@@ -73,3 +73,32 @@ def test_trainer_calls_momentum_update():
     tr = EagerTrainer(m, data, opt, grad_acc_steps=1)
     tr.train(0, 2)
     assert len(calls) == 2
+
+
+def test_clip_learns_and_inference_shapes():
+    from libai_amd.models import CLIPModel
+
+    torch.manual_seed(0)
+    m = CLIPModel(embed_dim=32, img_size=32, patch_size=8, vision_width=32,
+                  vision_layers=2, vision_heads=4, vocab_size=128,
+                  context_length=16, text_width=32, text_layers=2, text_heads=4)
+    imgs = torch.randn(4, 3, 32, 32)
+    txt = torch.randint(0, 127, (4, 12))
+    txt[:, -1] = 127  # eot = max id
+    opt = torch.optim.AdamW(m.parameters(), lr=1e-3)
+    first = None
+    for _ in range(12):
+        opt.zero_grad()
+        loss = m(images=imgs, text_ids=txt)["clip_loss"]
+        loss.backward()
+        opt.step()
+        first = first if first is not None else float(loss)
+    assert float(loss) < first, (first, float(loss))
+    m.eval()
+    with torch.no_grad():
+        out = m(images=imgs, text_ids=txt)
+        assert out["logits_per_image"].shape == (4, 4)
+        zi = m(images=imgs)["image_embeds"]
+        zt = m(text_ids=txt)["text_embeds"]
+    assert torch.allclose(zi.norm(dim=-1), torch.ones(4), atol=1e-4)
+    assert torch.allclose(zt.norm(dim=-1), torch.ones(4), atol=1e-4)
