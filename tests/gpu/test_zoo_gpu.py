@@ -206,3 +206,43 @@ def test_moe_layer_gpu_bf16():
     (y.float().pow(2).mean() + layer.mlp.last_aux_loss).backward()
     assert torch.isfinite(y.float()).all()
     assert layer.mlp.w1.grad is not None
+
+
+def test_clip_gpu_bf16():
+    from libai_amd.models import CLIPModel
+
+    torch.manual_seed(0)
+    m = CLIPModel(embed_dim=64, img_size=64, patch_size=16, vision_width=256,
+                  vision_layers=2, vision_heads=4, vocab_size=1024,
+                  context_length=32, text_width=256, text_layers=2,
+                  text_heads=4).to(torch.bfloat16).cuda()
+    imgs = torch.randn(4, 3, 64, 64, device="cuda", dtype=torch.bfloat16)
+    txt = torch.randint(0, 1023, (4, 16), device="cuda")
+    txt[:, -1] = 1023
+    _step(m, dict(images=imgs, text_ids=txt))
+
+
+def test_llama_gqa_gpu_bf16():
+    from libai_amd.models import LlamaForCausalLM
+
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(hidden_layers=2, vocab_size=1024, hidden_size=512,
+                         intermediate_size=1024, num_attention_heads=8,
+                         num_key_value_heads=2,
+                         max_position_embeddings=128).to(torch.bfloat16).cuda()
+    ids = torch.randint(0, 1024, (2, 65), device="cuda")
+    _step(m, dict(input_ids=ids[:, :-1], labels=ids[:, 1:]))
+
+
+def test_gpt_moe_gpu_bf16():
+    from libai_amd.models import GPTForPreTraining
+
+    torch.manual_seed(0)
+    m = GPTForPreTraining(hidden_layers=2, vocab_size=1024, hidden_size=256,
+                          ffn_hidden_size=512, num_attention_heads=4,
+                          max_seq_length=64, moe_num_experts=4,
+                          embedding_dropout_prob=0.0,
+                          attention_dropout_prob=0.0,
+                          output_dropout_prob=0.0).to(torch.bfloat16).cuda()
+    ids = torch.randint(0, 1024, (2, 65), device="cuda")
+    _step(m, dict(input_ids=ids[:, :-1], labels=ids[:, 1:]))
