@@ -97,6 +97,7 @@ def build_bench_model(args):
         if args.model == "llama7b":
             shape = dict(hidden_layers=32, hidden_size=4096, intermediate_size=11008,
                          num_attention_heads=32)
+            args.act_ckpt = True  # part of the BASELINE #4 config (7B on 288 GB)
         else:
             shape = dict(hidden_layers=16, hidden_size=2048, intermediate_size=5504,
                          num_attention_heads=16)
