@@ -33,8 +33,10 @@ class LlamaAttention(nn.Module):
 
     def __init__(self, hidden_size, num_heads, max_position_embeddings,
                  init_method, output_init_method, rope_theta=10000.0,
-                 num_key_value_heads=None, *, layer_idx=0):
+                 num_key_value_heads=None, sequence_parallel=False, *,
+                 layer_idx=0):
         super().__init__()
+        self.sequence_parallel = sequence_parallel
         self.hidden_size = hidden_size
         self.num_heads = num_heads
         self.head_dim = hidden_size // num_heads
@@ -55,22 +57,27 @@ class LlamaAttention(nn.Module):
             self.query_key_value = Linear1D(hidden_size, 3 * hidden_size,
                                             bias=False, parallel="col",
                                             init_method=init_method,
+                                            sequence_parallel=sequence_parallel,
                                             layer_idx=layer_idx)
         else:
             self.q_proj = Linear1D(hidden_size, hidden_size, bias=False,
                                    parallel="col", init_method=init_method,
+                                   sequence_parallel=sequence_parallel,
                                    layer_idx=layer_idx)
             self.kv_proj = Linear1D(
                 hidden_size, 2 * self.num_kv_heads * self.head_dim, bias=False,
                 parallel="col", init_method=init_method, fused_chunks=2,
-                layer_idx=layer_idx)
+                sequence_parallel=sequence_parallel, layer_idx=layer_idx)
         self.o_proj = Linear1D(hidden_size, hidden_size, bias=False, parallel="row",
                                init_method=output_init_method, skip_bias_add=True,
+                               sequence_parallel=sequence_parallel,
                                layer_idx=layer_idx)
         self.scale = 1.0 / math.sqrt(self.head_dim)
 
     def _project(self, hidden_states):
         b, s, _ = hidden_states.shape
+        if self.sequence_parallel:  # col projections gather the seq shards
+            s = s * du.get_dist_util().tensor_parallel_size
         if self.num_kv_heads == self.num_heads:
             qkv = self.query_key_value(hidden_states)
             qkv5 = qkv.view(b, s, self.num_heads_local, 3, self.head_dim)
@@ -83,7 +90,12 @@ class LlamaAttention(nn.Module):
 
     def forward(self, hidden_states, past_key_value=None, use_cache=False,
                 residual=None):
+        if self.sequence_parallel and (use_cache or past_key_value is not None):
+            raise RuntimeError("sequence_parallel is training-only; build the "
+                               "generation model with sequence_parallel=False")
         b, s, _ = hidden_states.shape
+        if self.sequence_parallel:
+            s = s * du.get_dist_util().tensor_parallel_size
         q, k, v = self._project(hidden_states)
         pos0 = past_key_value[0].shape[2] if past_key_value is not None else 0
         q = apply_rotary_pos_emb(q, self.max_pos, self.rope_theta, pos0)
@@ -148,14 +160,18 @@ class LlamaMLP(nn.Module):
     """Gated MLP: fused [gate|up] col projection -> SwiGLU -> row projection."""
 
     def __init__(self, hidden_size, intermediate_size, init_method,
-                 output_init_method, *, layer_idx=0):
+                 output_init_method, sequence_parallel=False, *, layer_idx=0):
         super().__init__()
         self.gate_up_proj = Linear1D(hidden_size, 2 * intermediate_size, bias=False,
                                      parallel="col", init_method=init_method,
-                                     fused_chunks=2, layer_idx=layer_idx)
+                                     fused_chunks=2,
+                                     sequence_parallel=sequence_parallel,
+                                     layer_idx=layer_idx)
         self.down_proj = Linear1D(intermediate_size, hidden_size, bias=False,
                                   parallel="row", init_method=output_init_method,
-                                  skip_bias_add=True, layer_idx=layer_idx)
+                                  skip_bias_add=True,
+                                  sequence_parallel=sequence_parallel,
+                                  layer_idx=layer_idx)
 
     def forward(self, x, residual=None):
         gu = self.gate_up_proj(x)
@@ -168,7 +184,8 @@ class LlamaDecoderLayer(nn.Module):
     def __init__(self, hidden_size, intermediate_size, num_heads,
                  max_position_embeddings, rms_norm_eps, init_method,
                  output_init_method, rope_theta=10000.0,
-                 num_key_value_heads=None, *, layer_idx=0):
+                 num_key_value_heads=None, sequence_parallel=False, *,
+                 layer_idx=0):
         super().__init__()
         self.layer_idx = layer_idx
         self.input_layernorm = RMSLayerNorm(hidden_size, eps=rms_norm_eps,
@@ -177,11 +194,19 @@ class LlamaDecoderLayer(nn.Module):
                                         max_position_embeddings, init_method,
                                         output_init_method, rope_theta,
                                         num_key_value_heads,
+                                        sequence_parallel=sequence_parallel,
                                         layer_idx=layer_idx)
         self.post_attention_layernorm = RMSLayerNorm(hidden_size, eps=rms_norm_eps,
                                                      layer_idx=layer_idx)
         self.mlp = LlamaMLP(hidden_size, intermediate_size, init_method,
-                            output_init_method, layer_idx=layer_idx)
+                            output_init_method,
+                            sequence_parallel=sequence_parallel,
+                            layer_idx=layer_idx)
+        if sequence_parallel:
+            # RMSNorms run on seq shards: grads are partial over local tokens
+            for ln in (self.input_layernorm, self.post_attention_layernorm):
+                for p in ln.parameters():
+                    p.sequence_parallel_grad = True
 
     def forward(self, hidden_states, past_key_value=None, use_cache=False):
         ln1 = self.input_layernorm(hidden_states)
@@ -213,9 +238,11 @@ class LlamaModel(nn.Module):
         tie_word_embeddings=False,
         rope_theta=10000.0,
         num_key_value_heads=None,
+        sequence_parallel=False,
         amp_enabled=False,
     ):
         super().__init__()
+        self.sequence_parallel = sequence_parallel
         init_method = init_method_normal(initializer_range)
         output_init_method = (
             scaled_init_method_normal(initializer_range, hidden_layers)
@@ -230,12 +257,16 @@ class LlamaModel(nn.Module):
                     hidden_size, intermediate_size, num_attention_heads,
                     max_position_embeddings, rms_norm_eps, init_method,
                     output_init_method, rope_theta, num_key_value_heads,
+                    sequence_parallel=sequence_parallel,
                     layer_idx=i,
                 )
                 for i in range(hidden_layers)
             ]
         )
         self.norm = RMSLayerNorm(hidden_size, eps=rms_norm_eps, layer_idx=-1)
+        if sequence_parallel:
+            for p in self.norm.parameters():
+                p.sequence_parallel_grad = True
         self.hidden_layers = hidden_layers
         self.checkpoint_activations = False
 
@@ -256,6 +287,7 @@ class LlamaModel(nn.Module):
             "tie_word_embeddings": cfg.get("tie_word_embeddings", False),
             "rope_theta": cfg.get("rope_theta", 10000.0),
             "num_key_value_heads": cfg.get("num_key_value_heads", None),
+            "sequence_parallel": cfg.get("sequence_parallel", False),
             "amp_enabled": cfg.get("amp_enabled", False),
         }
 
@@ -266,6 +298,11 @@ class LlamaModel(nn.Module):
 
     def forward(self, input_ids, past_key_values=None, use_cache=False):
         h = self.embed_tokens(input_ids)
+        if self.sequence_parallel and not use_cache and past_key_values is None:
+            from ..parallel.comm import scatter_to_sequence_parallel_region
+
+            assert input_ids.shape[1] % du.get_dist_util().tensor_parallel_size == 0
+            h = scatter_to_sequence_parallel_region(h)
         presents = [] if use_cache else None
         for i, layer in enumerate(self.layers):
             past = past_key_values[i] if past_key_values is not None else None
@@ -322,7 +359,10 @@ class LlamaForCausalLM(nn.Module):
                           else kwargs.get("initializer_range", 0.02))
             self.lm_head = _LMHeadWeight(vocab, hidden,
                                          init_method_normal(init_range))
-        self.lm_logits = LMLogits(vocab, bias=False, layer_idx=-1)
+        sp = (cfg.get("sequence_parallel", False) if cfg is not None
+              else kwargs.get("sequence_parallel", False))
+        self.lm_logits = LMLogits(vocab, bias=False, sequence_parallel=sp,
+                                  layer_idx=-1)
         self.loss_func = LlamaLoss()
 
     @classmethod

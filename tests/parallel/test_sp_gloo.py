@@ -87,3 +87,75 @@ def test_sp_tp2_matches_single_process():
                 f"grad mismatch {name} (rank {rank}): "
                 f"{(g - want).abs().max()}"
             )
+
+
+LLAMA_KW = dict(
+    hidden_layers=2,
+    vocab_size=128,
+    hidden_size=64,
+    intermediate_size=128,
+    num_attention_heads=8,
+    num_key_value_heads=2,  # GQA + SP compose
+    max_position_embeddings=64,
+)
+
+
+def _llama_reference():
+    from libai_amd.models import LlamaForCausalLM
+    from libai_amd.utils import distributed as du
+
+    du._DIST_UTIL = None
+    du.setup_dist_util({})
+    torch.manual_seed(123)
+    model = LlamaForCausalLM(**LLAMA_KW)
+    torch.manual_seed(9)
+    ids = torch.randint(0, 128, (2, 33))
+    out = model(input_ids=ids[:, :-1], labels=ids[:, 1:])
+    out["lm_loss"].backward()
+    grads = {n: p.grad.clone() for n, p in model.named_parameters()
+             if p.grad is not None}
+    return float(out["lm_loss"]), grads
+
+
+def _llama_sp_worker(rank, world):
+    import torch
+
+    from libai_amd.models import LlamaForCausalLM
+    from libai_amd.optim import FusedAdamW
+    from libai_amd.utils import distributed as du
+
+    du.setup_dist_util({"tensor_parallel_size": 2})
+    torch.manual_seed(123)
+    model = LlamaForCausalLM(**LLAMA_KW, sequence_parallel=True)
+    opt = FusedAdamW(model.parameters(), lr=1e-3)
+    torch.manual_seed(9)
+    ids = torch.randint(0, 128, (2, 33))
+    opt.zero_grad()
+    out = model(input_ids=ids[:, :-1], labels=ids[:, 1:])
+    out["lm_loss"].backward()
+    opt.grad_sync()
+    grads = {n: p.grad.clone() for n, p in model.named_parameters()
+             if p.grad is not None}
+    return float(out["lm_loss"]), grads
+
+
+def test_llama_gqa_sp_tp2_matches_single_process():
+    """SP + GQA + gated MLP (fused chunk pairing) at tp2 vs single rank."""
+    ref_loss, ref_grads = _llama_reference()
+    results = run_dist(_llama_sp_worker, 2)
+    from libai_amd.layers.linear import tp_slice
+
+    for rank, (loss, grads) in enumerate(results):
+        assert loss == pytest.approx(ref_loss, abs=1e-4)
+        for name, g in grads.items():
+            full = ref_grads[name]
+            want = full
+            if g.shape != full.shape:
+                dim = next(d for d in range(g.ndim)
+                           if g.shape[d] != full.shape[d])
+                fused = 2 if ("gate_up" in name or "kv_proj" in name) else 1
+                want = tp_slice(full, 2, rank, dim, fused)
+            assert torch.allclose(g, want, atol=1e-4), (
+                f"grad mismatch {name} (rank {rank}): "
+                f"{(g - want).abs().max()}"
+            )
