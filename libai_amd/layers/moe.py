@@ -115,7 +115,8 @@ class MoELayer(nn.Module):
         x = hidden_states.reshape(-1, self.hidden_size)
         n = x.shape[0]
 
-        logits = self.router(x.float())  # [n, E]
+        # routing decisions in fp32 regardless of the model dtype
+        logits = F.linear(x.float(), self.router.weight.float())  # [n, E]
         probs = torch.softmax(logits, dim=-1)
         gates, idx = probs.topk(self.top_k, dim=-1)  # [n, k]
         gates = gates / gates.sum(dim=-1, keepdim=True)
