@@ -160,9 +160,22 @@ class EagerTrainer(TrainerBase):
         if isinstance(data, Instance):
             data = data.to_dict()
         device = du.get_device()
+        keep = None
+        if self.pipeline_scheduler is not None and hasattr(
+            self.model, "pipeline_stage_batch_keys"
+        ):
+            # middle pipeline stages often consume NO batch tensors (GPT) or
+            # only small masks (BERT) — skip the H2D copies of the rest
+            # (the reference loads the full batch on every stage; this was
+            # flagged as wasteful for large CV inputs)
+            keep = self.model.pipeline_stage_batch_keys(
+                self.pipeline_scheduler.is_first,
+                self.pipeline_scheduler.is_last,
+            )
         out = {
             k: (v.to(device, non_blocking=True) if torch.is_tensor(v) else v)
             for k, v in data.items()
+            if keep is None or k in keep
         }
         # CV mixup/cutmix, applied at batch time like the reference
         # (engine/default.py:509-515); produces soft labels which the default

@@ -269,6 +269,16 @@ class GPTForPreTraining(nn.Module):
     def set_activation_checkpoint(self, enabled=True):
         self.GPT_model.set_activation_checkpoint(enabled)
 
+    def pipeline_stage_batch_keys(self, is_first, is_last):
+        """Batch tensors this pipeline stage actually consumes (middle
+        stages read nothing; the activations arrive via P2P)."""
+        keys = set()
+        if is_first:
+            keys.add("input_ids")
+        if is_last:
+            keys.add("labels")
+        return keys
+
     # -- pipeline protocol --------------------------------------------------
 
     def pipeline_units(self):

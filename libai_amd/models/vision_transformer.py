@@ -56,10 +56,12 @@ class VisionTransformer(nn.Module):
         self.img_size = img_size
         self.num_classes = num_classes
         self.depth = depth
+        self.embed_dim = embed_dim
         init_method = init_method_normal(0.02)
 
         self.embedding = ViTEmbedding(img_size, patch_size, in_chans, embed_dim,
                                       drop_rate)
+        self.patch_embed_seq_len = self.embedding.num_patches + 1  # + CLS
 
         ffn_size = int(embed_dim * mlp_ratio)
         self.blocks = nn.ModuleList(
@@ -119,6 +121,19 @@ class VisionTransformer(nn.Module):
 
     def set_activation_checkpoint(self, enabled=True):
         self.checkpoint_activations = enabled
+
+    def pipeline_stage_batch_keys(self, is_first, is_last):
+        # labels ([b] ints) ride along everywhere as the batch-size witness
+        # for boundary shapes; the [b, 3, H, W] images only go to stage 0
+        keys = {"labels"}
+        if is_first:
+            keys.add("images")
+        return keys
+
+    def pipeline_boundary_shapes(self, batch, first_idx):
+        ref = batch.get("images", batch.get("labels"))
+        b = ref.shape[0]
+        return [(b, self.patch_embed_seq_len, self.embed_dim)]
 
     # -- pipeline protocol --------------------------------------------------
 

@@ -88,3 +88,13 @@ def test_pp2_matches_single_process(num_micro):
     assert seen == set(ref_grads.keys()), (
         f"missing grads: {set(ref_grads) - seen}"
     )
+
+
+def test_pipeline_stage_batch_keys():
+    """Middle stages consume no batch tensors; first/last declare theirs."""
+    from libai_amd.models import GPTForPreTraining
+
+    m = GPTForPreTraining(**MODEL_KW)
+    assert m.pipeline_stage_batch_keys(True, False) == {"input_ids"}
+    assert m.pipeline_stage_batch_keys(False, True) == {"labels"}
+    assert m.pipeline_stage_batch_keys(False, False) == set()
