@@ -285,6 +285,16 @@ class BertForPreTraining(nn.Module):
 
     # -- pipeline protocol --------------------------------------------------
 
+    def pipeline_stage_batch_keys(self, is_first, is_last):
+        # every stage rebuilds the padding mask; input_ids ride along as the
+        # [b, s] shape witness for the activation boundary
+        keys = {"input_ids", "attention_mask"}
+        if is_first:
+            keys.add("tokentype_ids")
+        if is_last:
+            keys.update({"lm_labels", "loss_mask", "ns_labels"})
+        return keys
+
     def pipeline_units(self):
         units = [
             (

@@ -167,6 +167,12 @@ class BloomForCausalLM(nn.Module):
 
     # -- pipeline protocol --------------------------------------------------
 
+    def pipeline_stage_batch_keys(self, is_first, is_last):
+        keys = {"input_ids"}  # shape witness on middle stages
+        if is_last:
+            keys.add("labels")
+        return keys
+
     def pipeline_units(self):
         bl = self.bloom
 

@@ -389,6 +389,12 @@ class LlamaForCausalLM(nn.Module):
 
     # -- pipeline protocol --------------------------------------------------
 
+    def pipeline_stage_batch_keys(self, is_first, is_last):
+        keys = {"input_ids"}
+        if is_last:
+            keys.add("labels")
+        return keys
+
     def pipeline_units(self):
         units = [(0, "embed_tokens", lambda h, b: self.model.embed_tokens(b["input_ids"]))]
         for i, layer in enumerate(self.model.layers):

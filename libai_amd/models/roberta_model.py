@@ -122,6 +122,14 @@ class RobertaForPreTraining(nn.Module):
             emb = emb + rb.embeddings.tokentype_embeddings(tt)
         return rb.embeddings.embedding_dropout(emb)
 
+    def pipeline_stage_batch_keys(self, is_first, is_last):
+        keys = {"input_ids", "attention_mask"}
+        if is_first:
+            keys.add("tokentype_ids")
+        if is_last:
+            keys.update({"lm_labels", "loss_mask"})
+        return keys
+
     def pipeline_units(self):
         from .bert_model import extended_attn_mask
 

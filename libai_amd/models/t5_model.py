@@ -317,6 +317,13 @@ class T5ForPreTraining(nn.Module):
                 b["_cross_mask"] = cross_attn_mask(dm, enc)
         return b["_cross_mask"]
 
+    def pipeline_stage_batch_keys(self, is_first, is_last):
+        keys = {"encoder_input_ids", "decoder_input_ids",
+                "encoder_attn_mask", "decoder_attn_mask"}
+        if is_last:
+            keys.update({"lm_labels", "loss_mask"})
+        return keys
+
     def pipeline_units(self):
         t5 = self.t5_model
         n = t5.hidden_layers
