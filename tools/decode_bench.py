@@ -21,6 +21,7 @@ def main():
     p.add_argument("--batch", type=int, default=32)
     p.add_argument("--ctx", type=int, default=1024)
     p.add_argument("--steps", type=int, default=64)
+    p.add_argument("--model", default="gpt2", choices=["gpt2", "llama1b"])
     args = p.parse_args()
 
     import bench as bench_mod
@@ -32,14 +33,25 @@ def main():
 
     du.setup_dist_util({})
     torch.manual_seed(0)
-    m = GPTForPreTraining(
-        hidden_layers=24, vocab_size=50304, hidden_size=1024,
-        ffn_hidden_size=4096, num_attention_heads=16, max_seq_length=2048,
-        embedding_dropout_prob=0.0, attention_dropout_prob=0.0,
-        output_dropout_prob=0.0,
-    ).to(torch.bfloat16).cuda().eval()
+    if args.model == "llama1b":
+        from libai_amd.models import LlamaForCausalLM
 
-    ids = torch.randint(0, 50304, (args.batch, args.ctx), device="cuda")
+        m = LlamaForCausalLM(
+            hidden_layers=16, vocab_size=32000, hidden_size=2048,
+            intermediate_size=5504, num_attention_heads=16,
+            num_key_value_heads=4, max_position_embeddings=4096,
+        ).to(torch.bfloat16).cuda().eval()
+        vocab = 32000
+    else:
+        m = GPTForPreTraining(
+            hidden_layers=24, vocab_size=50304, hidden_size=1024,
+            ffn_hidden_size=4096, num_attention_heads=16, max_seq_length=2048,
+            embedding_dropout_prob=0.0, attention_dropout_prob=0.0,
+            output_dropout_prob=0.0,
+        ).to(torch.bfloat16).cuda().eval()
+        vocab = 50304
+
+    ids = torch.randint(0, vocab, (args.batch, args.ctx), device="cuda")
 
     def run(use_fused):
         orig = A.decode_attention_available
@@ -65,7 +77,7 @@ def main():
 
     t_unfused = run(False)
     t_fused = run(True)
-    print(f"# decode bench: GPT-2 345M, batch {args.batch}, ctx {args.ctx}")
+    print(f"# decode bench: {args.model}, batch {args.batch}, ctx {args.ctx}")
     print(f"unfused bmm+softmax decode: {t_unfused * 1e3:.3f} ms/step "
           f"({args.batch / t_unfused:.0f} tok/s)")
     print(f"fused flash_decode:         {t_fused * 1e3:.3f} ms/step "
