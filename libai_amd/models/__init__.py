@@ -6,6 +6,7 @@ from .build import build_model
 from .gpt_model import GPTForPreTraining, GPTModel
 from .llama import LlamaForCausalLM, LlamaModel
 from .mae import MAEForPreTraining
+from .palm import PaLMForCausalLM, PaLMModel
 from .resmlp import ResMLP
 from .roberta_model import RobertaForCausalLM, RobertaForPreTraining, RobertaModel
 from .swin_transformer import SwinTransformer
@@ -35,5 +36,7 @@ __all__ = [
     "SimCSEModel",
     "MoCoV3",
     "CLIPModel",
+    "PaLMModel",
+    "PaLMForCausalLM",
     "build_model",
 ]

@@ -246,3 +246,14 @@ def test_gpt_moe_gpu_bf16():
                           output_dropout_prob=0.0).to(torch.bfloat16).cuda()
     ids = torch.randint(0, 1024, (2, 65), device="cuda")
     _step(m, dict(input_ids=ids[:, :-1], labels=ids[:, 1:]))
+
+
+def test_palm_gpu_bf16():
+    from libai_amd.models import PaLMForCausalLM
+
+    torch.manual_seed(0)
+    m = PaLMForCausalLM(hidden_layers=2, vocab_size=1024, hidden_size=512,
+                        intermediate_size=1024, num_attention_heads=8,
+                        max_position_embeddings=128).to(torch.bfloat16).cuda()
+    ids = torch.randint(0, 1024, (2, 65), device="cuda")
+    _step(m, dict(input_ids=ids[:, :-1], labels=ids[:, 1:]))

@@ -213,3 +213,26 @@ def test_llama_gqa_fwd_bwd_and_decode_parity():
                past_key_values=o["past_key_values"], use_cache=True)
     assert (st["prediction_scores"][:, 0] - full[:, 10]).abs().max() < 1e-4
     assert o["past_key_values"][0][0].shape[1] == 2  # kv heads only
+
+
+def test_palm_parallel_block_and_decode_parity():
+    """PaLM: parallel attn+MLP residual, multi-query attention (kv=1),
+    tied logits; incremental decode == full forward."""
+    from libai_amd.models import PaLMForCausalLM
+
+    torch.manual_seed(0)
+    m = PaLMForCausalLM(hidden_layers=2, vocab_size=128, hidden_size=64,
+                        intermediate_size=128, num_attention_heads=8,
+                        max_position_embeddings=64).eval()
+    ids = torch.randint(0, 128, (2, 18))
+    out = m(input_ids=ids[:, :-1], labels=ids[:, 1:])
+    out["lm_loss"].backward()
+    assert torch.isfinite(out["lm_loss"])
+    # MQA: one kv head
+    assert m.model.layers[0].attn.num_kv_heads == 1
+    with torch.no_grad():
+        full = m(input_ids=ids)["prediction_scores"]
+        o = m(input_ids=ids[:, :10], use_cache=True)
+        st = m(input_ids=ids[:, 10:11],
+               past_key_values=o["past_key_values"], use_cache=True)
+    assert (st["prediction_scores"][:, 0] - full[:, 10]).abs().max() < 1e-4
