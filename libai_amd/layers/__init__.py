@@ -8,6 +8,7 @@ from .layer_norm import LayerNorm, RMSLayerNorm, RMSNorm
 from .linear import Linear, Linear1D
 from .lm_logits import LMLogits
 from .moe import MoELayer
+from .position_bias import T5RelativePositionBias, build_alibi_bias
 from .mlp import MLP
 from .transformer_layer import TransformerLayer
 
@@ -31,5 +32,7 @@ __all__ = [
     "LMLogits",
     "MLP",
     "MoELayer",
+    "T5RelativePositionBias",
+    "build_alibi_bias",
     "TransformerLayer",
 ]
