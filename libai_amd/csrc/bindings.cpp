@@ -59,18 +59,19 @@ void ln_bwd_wgrad_f32(const void*, const void*, const float*, const float*, floa
                       float*, void*, void*, int64_t, int, int, bool, hipStream_t);
 void bias_gelu_fwd_bf16(const void*, const void*, void*, int64_t, int, hipStream_t);
 void bias_gelu_fwd_f32(const void*, const void*, void*, int64_t, int, hipStream_t);
-void bias_gelu_bwd_bf16(const void*, const void*, const void*, void*, int64_t, int,
-                        hipStream_t);
-void bias_gelu_bwd_f32(const void*, const void*, const void*, void*, int64_t, int,
-                       hipStream_t);
+void bias_gelu_bwd_bf16(const void*, const void*, const void*, void*, float*,
+                        int64_t, int, hipStream_t);
+void bias_gelu_bwd_f32(const void*, const void*, const void*, void*, float*,
+                       int64_t, int, hipStream_t);
+int bias_col_grid_rows(int64_t, int, int);
 void bias_dropout_res_fwd_bf16(const void*, const void*, const void*, void*, int64_t,
                                int, float, uint64_t, hipStream_t);
 void bias_dropout_res_fwd_f32(const void*, const void*, const void*, void*, int64_t,
                               int, float, uint64_t, hipStream_t);
-void bias_dropout_res_bwd_bf16(const void*, void*, int64_t, int, float, uint64_t,
-                               hipStream_t);
-void bias_dropout_res_bwd_f32(const void*, void*, int64_t, int, float, uint64_t,
-                              hipStream_t);
+void bias_dropout_res_bwd_bf16(const void*, void*, float*, int64_t, int, float,
+                               uint64_t, hipStream_t);
+void bias_dropout_res_bwd_f32(const void*, void*, float*, int64_t, int, float,
+                              uint64_t, hipStream_t);
 void colsum_bf16(const void*, float*, void*, int64_t, int, int, hipStream_t);
 void colsum_f32(const void*, float*, void*, int64_t, int, int, hipStream_t);
 void softmax_fwd_bf16(const void*, const uint8_t*, void*, int64_t, int, int, int, float,
@@ -219,17 +220,29 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, c10::optional<torch::Tensor> b) {
   return y;
 }
 
-torch::Tensor bias_gelu_bwd(torch::Tensor x, c10::optional<torch::Tensor> b,
-                            torch::Tensor dy) {
+std::tuple<torch::Tensor, c10::optional<torch::Tensor>> bias_gelu_bwd(
+    torch::Tensor x, c10::optional<torch::Tensor> b, torch::Tensor dy,
+    bool want_dbias) {
   CHECK_IN(x);
   CHECK_IN(dy);
   auto dx = torch::empty_like(x);
   const int W = b.has_value() ? (int)b->numel() : (int)x.size(-1);
+  const int V = is_bf16(x) ? 8 : 4;
+  float* pp = nullptr;
+  torch::Tensor partial;
+  int P = 0;
+  if (want_dbias && W % V == 0) {
+    P = bias_col_grid_rows(x.numel(), W, V);
+    partial = torch::empty({P, W}, x.options().dtype(torch::kFloat32));
+    pp = partial.data_ptr<float>();
+  }
   auto fn = is_bf16(x) ? bias_gelu_bwd_bf16 : bias_gelu_bwd_f32;
   fn(x.data_ptr(), b.has_value() ? b->data_ptr() : nullptr, dy.data_ptr(), dx.data_ptr(),
-     x.numel(), W, cur_stream());
+     pp, x.numel(), W, cur_stream());
   check_launch("bias_gelu_bwd");
-  return dx;
+  if (pp != nullptr)
+    return {dx, partial.sum(0).to(x.scalar_type())};
+  return {dx, c10::nullopt};
 }
 
 torch::Tensor colsum(torch::Tensor x, int64_t W) {
@@ -262,14 +275,26 @@ torch::Tensor bias_dropout_res_fwd(torch::Tensor x, c10::optional<torch::Tensor>
   return y;
 }
 
-torch::Tensor bias_dropout_res_bwd(torch::Tensor dy, double p, int64_t seed) {
+std::tuple<torch::Tensor, c10::optional<torch::Tensor>> bias_dropout_res_bwd(
+    torch::Tensor dy, double p, int64_t seed, bool want_dbias) {
   CHECK_IN(dy);
   auto dx = torch::empty_like(dy);
+  const int W = (int)dy.size(-1);
+  const int V = is_bf16(dy) ? 8 : 4;
+  float* pp = nullptr;
+  torch::Tensor partial;
+  if (want_dbias && W % V == 0) {
+    int P = bias_col_grid_rows(dy.numel(), W, V);
+    partial = torch::empty({P, W}, dy.options().dtype(torch::kFloat32));
+    pp = partial.data_ptr<float>();
+  }
   auto fn = is_bf16(dy) ? bias_dropout_res_bwd_bf16 : bias_dropout_res_bwd_f32;
-  fn(dy.data_ptr(), dx.data_ptr(), dy.numel(), (int)dy.size(-1), (float)p,
+  fn(dy.data_ptr(), dx.data_ptr(), pp, dy.numel(), W, (float)p,
      (uint64_t)seed, cur_stream());
   check_launch("bias_dropout_res_bwd");
-  return dx;
+  if (pp != nullptr)
+    return {dx, partial.sum(0).to(dy.scalar_type())};
+  return {dx, c10::nullopt};
 }
 
 // ---------------------------------------------------------------------------

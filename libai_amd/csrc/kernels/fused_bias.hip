@@ -18,7 +18,9 @@ template <class E, bool GRAD>
 __global__ void bias_gelu_kernel(const typename E::T* __restrict__ x,
                                  const typename E::T* __restrict__ b,
                                  const typename E::T* __restrict__ dy,
-                                 typename E::T* __restrict__ out, int64_t n, int W) {
+                                 typename E::T* __restrict__ out,
+                                 float* __restrict__ db_partial, int64_t n,
+                                 int W) {
   using VecT = typename E::VecT;
   constexpr int V = E::VEC;
   const int wvec = W / V;
@@ -27,6 +29,7 @@ __global__ void bias_gelu_kernel(const typename E::T* __restrict__ x,
   if (cv >= wvec) return;
   VecT vb;
   if (b) vb = ((const VecT*)b)[cv];
+  float db[GRAD ? V : 1] = {0.f};
   for (int64_t r = blockIdx.y; r < R; r += gridDim.y) {
     const int64_t i = r * wvec + cv;
     VecT vx = ((const VecT*)x)[i];
@@ -36,7 +39,9 @@ __global__ void bias_gelu_kernel(const typename E::T* __restrict__ x,
 #pragma unroll
       for (int j = 0; j < V; ++j) {
         float xv = E::to_f(vx[j]) + (b ? E::to_f(vb[j]) : 0.f);
-        o[j] = E::from_f(E::to_f(vdy[j]) * gelu_erf_grad(xv));
+        float g = E::to_f(vdy[j]) * gelu_erf_grad(xv);
+        o[j] = E::from_f(g);
+        db[j] += g;  // bias grad rides along: no second pass over dx
       }
     } else {
 #pragma unroll
@@ -46,6 +51,11 @@ __global__ void bias_gelu_kernel(const typename E::T* __restrict__ x,
       }
     }
     ((VecT*)out)[i] = o;
+  }
+  if (GRAD && db_partial != nullptr) {
+#pragma unroll
+    for (int j = 0; j < V; ++j)
+      db_partial[(int64_t)blockIdx.y * W + cv * V + j] = db[j];
   }
 }
 
@@ -61,7 +71,8 @@ template <class E, bool GRAD>
 __global__ void bias_dropout_res_kernel(const typename E::T* __restrict__ x,
                                         const typename E::T* __restrict__ b,
                                         const typename E::T* __restrict__ res,
-                                        typename E::T* __restrict__ out, int64_t n,
+                                        typename E::T* __restrict__ out,
+                                        float* __restrict__ db_partial, int64_t n,
                                         int W, float p, uint64_t seed) {
   using VecT = typename E::VecT;
   constexpr int V = E::VEC;
@@ -73,6 +84,7 @@ __global__ void bias_dropout_res_kernel(const typename E::T* __restrict__ x,
   if (cv >= wvec) return;
   VecT vb;
   if (b) vb = ((const VecT*)b)[cv];
+  float db[GRAD ? V : 1] = {0.f};
   for (int64_t r = blockIdx.y; r < R; r += gridDim.y) {
     const int64_t i = r * wvec + cv;
     VecT vx = ((const VecT*)x)[i];
@@ -93,7 +105,11 @@ __global__ void bias_dropout_res_kernel(const typename E::T* __restrict__ x,
     VecT o;
     if (GRAD) {
 #pragma unroll
-      for (int j = 0; j < V; ++j) o[j] = E::from_f(E::to_f(vx[j]) * keep[j]);
+      for (int j = 0; j < V; ++j) {
+        float g = E::to_f(vx[j]) * keep[j];
+        o[j] = E::from_f(g);
+        db[j] += g;
+      }
     } else {
       VecT vr;
       if (res) vr = ((const VecT*)res)[i];
@@ -104,6 +120,11 @@ __global__ void bias_dropout_res_kernel(const typename E::T* __restrict__ x,
       }
     }
     ((VecT*)out)[i] = o;
+  }
+  if (GRAD && db_partial != nullptr) {
+#pragma unroll
+    for (int j = 0; j < V; ++j)
+      db_partial[(int64_t)blockIdx.y * W + cv * V + j] = db[j];
   }
 }
 
@@ -168,21 +189,29 @@ inline dim3 col_grid(int64_t n, int W, int V, int* block_out) {
 
 }  // namespace
 
+extern "C" int bias_col_grid_rows(int64_t n, int W, int V) {
+  int blk;
+  dim3 g = col_grid(n, W, V, &blk);
+  return (int)g.y;
+}
+
 #define BIAS_LAUNCHERS(SUFF, ETYPE)                                                      \
   extern "C" void bias_gelu_fwd_##SUFF(const void* x, const void* b, void* y, int64_t n, \
                                        int W, hipStream_t stream) {                      \
     int blk;                                                                             \
     dim3 g = col_grid(n, W, ETYPE::VEC, &blk);                                           \
     bias_gelu_kernel<ETYPE, false><<<g, dim3(blk), 0, stream>>>(                         \
-        (const ETYPE::T*)x, (const ETYPE::T*)b, nullptr, (ETYPE::T*)y, n, W);            \
+        (const ETYPE::T*)x, (const ETYPE::T*)b, nullptr, (ETYPE::T*)y, nullptr, n,       \
+        W);                                                                              \
   }                                                                                      \
   extern "C" void bias_gelu_bwd_##SUFF(const void* x, const void* b, const void* dy,     \
-                                       void* dx, int64_t n, int W, hipStream_t stream) { \
+                                       void* dx, float* db_partial, int64_t n, int W,    \
+                                       hipStream_t stream) {                             \
     int blk;                                                                             \
     dim3 g = col_grid(n, W, ETYPE::VEC, &blk);                                           \
     bias_gelu_kernel<ETYPE, true><<<g, dim3(blk), 0, stream>>>(                          \
-        (const ETYPE::T*)x, (const ETYPE::T*)b, (const ETYPE::T*)dy, (ETYPE::T*)dx, n,   \
-        W);                                                                              \
+        (const ETYPE::T*)x, (const ETYPE::T*)b, (const ETYPE::T*)dy, (ETYPE::T*)dx,      \
+        db_partial, n, W);                                                               \
   }                                                                                      \
   extern "C" void bias_dropout_res_fwd_##SUFF(const void* x, const void* b,              \
                                               const void* res, void* y, int64_t n,       \
@@ -191,16 +220,18 @@ inline dim3 col_grid(int64_t n, int W, int V, int* block_out) {
     int blk;                                                                             \
     dim3 g = col_grid(n, W, ETYPE::VEC, &blk);                                           \
     bias_dropout_res_kernel<ETYPE, false><<<g, dim3(blk), 0, stream>>>(                  \
-        (const ETYPE::T*)x, (const ETYPE::T*)b, (const ETYPE::T*)res, (ETYPE::T*)y, n,   \
-        W, p, seed);                                                                     \
+        (const ETYPE::T*)x, (const ETYPE::T*)b, (const ETYPE::T*)res, (ETYPE::T*)y,      \
+        nullptr, n, W, p, seed);                                                         \
   }                                                                                      \
-  extern "C" void bias_dropout_res_bwd_##SUFF(const void* dy, void* dx, int64_t n,       \
-                                              int W, float p, uint64_t seed,             \
+  extern "C" void bias_dropout_res_bwd_##SUFF(const void* dy, void* dx,                  \
+                                              float* db_partial, int64_t n, int W,       \
+                                              float p, uint64_t seed,                    \
                                               hipStream_t stream) {                      \
     int blk;                                                                             \
     dim3 g = col_grid(n, W, ETYPE::VEC, &blk);                                           \
     bias_dropout_res_kernel<ETYPE, true><<<g, dim3(blk), 0, stream>>>(                   \
-        (const ETYPE::T*)dy, nullptr, nullptr, (ETYPE::T*)dx, n, W, p, seed);            \
+        (const ETYPE::T*)dy, nullptr, nullptr, (ETYPE::T*)dx, db_partial, n, W, p,       \
+        seed);                                                                           \
   }                                                                                      \
   extern "C" void colsum_##SUFF(const void* in, float* partial, void* out, int64_t R,    \
                                 int W, int P, hipStream_t stream) {                      \

@@ -32,8 +32,11 @@ class _BiasGeluFn(torch.autograd.Function):
             (x,) = ctx.saved_tensors
             b = None
         dy = dy.contiguous()
-        dx = ext().bias_gelu_bwd(x, b, dy)
-        dbias = ext().colsum(dx, x.shape[-1]) if ctx.has_bias else None
+        # the bias grad partials ride in the same kernel pass (no re-read
+        # of dx); falls back to the separate colsum for ragged widths
+        dx, dbias = ext().bias_gelu_bwd(x, b, dy, ctx.has_bias)
+        if ctx.has_bias and dbias is None:
+            dbias = ext().colsum(dx, x.shape[-1])
         return dx, dbias
 
 
@@ -62,11 +65,14 @@ class _BiasDropoutAddFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         dy = dy.contiguous()
+        dbias = None
         if ctx.p > 0:
-            dx = ext().bias_dropout_res_bwd(dy, ctx.p, ctx.seed)
+            dx, dbias = ext().bias_dropout_res_bwd(dy, ctx.p, ctx.seed,
+                                                   ctx.has_bias)
         else:
             dx = dy
-        dbias = ext().colsum(dx, ctx.width) if ctx.has_bias else None
+        if ctx.has_bias and dbias is None:
+            dbias = ext().colsum(dx, ctx.width)
         dres = dy if ctx.has_res else None
         return dx, dbias, dres, None
 

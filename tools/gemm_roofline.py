@@ -98,8 +98,7 @@ def main():
             dy4h = torch.matmul(dy, w2)  # dX GEMM of the row linear
             from libai_amd.ops.fused_bias import bias_gelu as _bg  # noqa
 
-            dpre = ext().bias_gelu_bwd(pre, torch.zeros_like(b1), dy4h)
-            db = dpre.float().sum(0)
+            dpre, db = ext().bias_gelu_bwd(pre, torch.zeros_like(b1), dy4h, True)
             return dpre, db
 
         def fused():
