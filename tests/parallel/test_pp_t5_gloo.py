@@ -304,3 +304,28 @@ def test_vit_pp2_end_to_end():
     """ViT under PP through the TRAINER (catches the boundary-shape
     derivation and the stage batch-key filtering)."""
     run_dist(_vit_pp2_worker, 2)
+
+
+def _t5_eval_pp2_worker(rank, world):
+    import torch
+
+    from libai_amd.evaluation import inference_on_dataset
+    from libai_amd.evaluation.ppl_evaluator import PPLEvaluator
+    from libai_amd.models.t5_model import T5ForPreTraining
+    from libai_amd.parallel.pipeline import PipelineScheduler
+    from libai_amd.utils import distributed as du
+
+    du.setup_dist_util({"pipeline_parallel_size": 2, "pipeline_num_layers": 4})
+    torch.manual_seed(123)
+    model = T5ForPreTraining(**T5_KW)
+    sched = PipelineScheduler(model, dtype=torch.float32)
+    res = inference_on_dataset(model, _t5_batches(3), PPLEvaluator(),
+                               pipeline_scheduler=sched)
+    return res
+
+
+def test_t5_eval_under_pp2():
+    """Eval through the TUPLE-boundary pipeline (enc-dec): both ranks get
+    identical, finite results."""
+    results = run_dist(_t5_eval_pp2_worker, 2)
+    assert results[0] == results[1] and results[0]
