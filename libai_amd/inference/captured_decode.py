@@ -1,0 +1,145 @@
+"""hipGraph-captured greedy decode (serving hot loop).
+
+The per-token decode step is LAUNCH-BOUND: ~9 kernels per layer x 24 layers
+at ~5 us each of launch overhead.  This module captures ONE fully static
+decode step — embedding lookup at a device position tensor, per-layer
+in-place KV insert + fused flash_decode (K16), greedy argmax, token/position
+advance — into a single hipGraph and replays it per token with ZERO host
+work and zero synchronisation inside the loop.
+
+Static-shape contract (what makes the step capturable):
+  - KV caches preallocated at [b, nh, max_len, hs]; writes via index_copy_
+    at a DEVICE position tensor (no host ints anywhere in the step)
+  - per-batch kv_len is a device int32 tensor the flash_decode kernel masks
+    by, so the grid is sized once for max_len
+  - the argmax result is copied back into the static input token buffer and
+    recorded into a preallocated output buffer at a device step counter, so
+    the graph is fully self-advancing: N replays = N tokens
+
+Reference capability: libai/inference/generator/generation_utils.py greedy
+search + oneflow nn.Graph capture of the decode step; rebuilt here on
+torch.cuda.CUDAGraph (hipGraph on ROCm) + the K16 decode kernel.
+
+GPU-only (bf16, head_size in {64, 128}, TP=1): raises on anything else —
+no silent eager fallback.
+"""
+
+import torch
+
+__all__ = ["CapturedGPTDecoder"]
+
+
+class CapturedGPTDecoder:
+    """Capture-once / replay-per-token greedy decoder for GPTForPreTraining.
+
+    Usage::
+
+        dec = CapturedGPTDecoder(model, max_batch=32, max_seq_len=1024)
+        tokens = dec.generate(prompt_ids, max_new_tokens=64)  # [b, new]
+    """
+
+    def __init__(self, model, max_batch, max_seq_len):
+        gpt = model.GPT_model if hasattr(model, "GPT_model") else model
+        self.model = model
+        self.gpt = gpt
+        attn = gpt.transformer.layers[0].self_attention
+        from ..utils import distributed as du
+
+        if du.get_dist_util().tensor_parallel_size != 1:
+            raise RuntimeError("CapturedGPTDecoder supports TP=1 (shard the "
+                               "batch over DP ranks for captured serving)")
+        if attn.head_size not in (64, 128):
+            raise RuntimeError(f"flash_decode needs head_size in (64,128), "
+                               f"got {attn.head_size}")
+        self.n_layers = len(gpt.transformer.layers)
+        self.n_heads = attn.num_heads_local
+        self.head_size = attn.head_size
+        self.max_batch = max_batch
+        self.max_seq_len = max_seq_len
+        dev = next(model.parameters()).device
+        if dev.type != "cuda":
+            raise RuntimeError("CapturedGPTDecoder is GPU-only")
+        self.device = dev
+        b, h, d = max_batch, self.n_heads, self.head_size
+        self.caches = [
+            (torch.zeros(b, h, max_seq_len, d, device=dev, dtype=torch.bfloat16),
+             torch.zeros(b, h, max_seq_len, d, device=dev, dtype=torch.bfloat16),
+             torch.zeros(b, dtype=torch.int32, device=dev))
+            for _ in range(self.n_layers)
+        ]
+        self.pos = torch.zeros(1, dtype=torch.int64, device=dev)
+        self.step_idx = torch.zeros(1, dtype=torch.int64, device=dev)
+        self.static_tok = torch.zeros(b, 1, dtype=torch.int64, device=dev)
+        self.out_tokens = None  # sized per generate() call
+        self.graph = None
+        self._graph_new = 0
+
+    # ---- the one decode step (the thing that gets captured) -------------
+    def _step(self):
+        logits = self.gpt(self.static_tok, static_caches=self.caches,
+                          position=self.pos)
+        nxt = logits[:, -1, :].argmax(dim=-1, keepdim=True)  # [b, 1]
+        self.out_tokens.index_copy_(1, self.step_idx, nxt)
+        self.static_tok.copy_(nxt)
+        self.pos.add_(1)
+        self.step_idx.add_(1)
+        for _, _, kv32 in self.caches:
+            kv32.add_(1)
+
+    def _reset_to(self, prompt_len, first_tok):
+        self.pos.fill_(prompt_len)
+        self.step_idx.fill_(1)
+        self.static_tok.copy_(first_tok)
+        for _, _, kv32 in self.caches:
+            kv32.fill_(prompt_len + 1)  # length AFTER this step's insert
+
+    @torch.no_grad()
+    def generate(self, prompt_ids, max_new_tokens):
+        """Greedy-decode ``max_new_tokens`` tokens after ``prompt_ids``
+        ([b, L] int64, b <= max_batch, L + new <= max_seq_len).
+        Returns the generated tokens [b, max_new_tokens]."""
+        b, L = prompt_ids.shape
+        assert b == self.max_batch, "captured graph is shape-static: pad the batch"
+        assert L + max_new_tokens <= self.max_seq_len
+        self.model.eval()
+
+        # 1) eager prefill (one big flash_fwd pass), fill the static caches
+        out = self.model(input_ids=prompt_ids.to(self.device), use_cache=True)
+        past = out["past_key_values"]
+        logits = out["prediction_scores"]
+        for (ck, cv, _), (pk, pv) in zip(self.caches, past):
+            ck[:, :, :L].copy_(pk)
+            cv[:, :, :L].copy_(pv)
+        first = logits[:, -1, :].argmax(dim=-1, keepdim=True)  # token at pos L
+
+        if self.out_tokens is None or self.out_tokens.shape[1] < max_new_tokens:
+            self.out_tokens = torch.zeros(self.max_batch, max_new_tokens,
+                                          dtype=torch.int64, device=self.device)
+            self.graph = None  # out buffer is graph-captured state
+        self.out_tokens[:, 0:1].copy_(first)
+
+        if max_new_tokens == 1:
+            return self.out_tokens[:, :1].clone()
+
+        if self.graph is None:
+            # warmup twice on a side stream (allocator + library init),
+            # then reset the advance state and capture
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                self._reset_to(L, first)
+                self._step()
+                self._step()
+            torch.cuda.current_stream().wait_stream(s)
+            self._reset_to(L, first)
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                self._step()
+            self._graph_new = max_new_tokens
+        else:
+            self._reset_to(L, first)
+
+        # 2) replay: each replay appends one token, no host work at all
+        for _ in range(max_new_tokens - 1):
+            self.graph.replay()
+        return self.out_tokens[:, :max_new_tokens].clone()

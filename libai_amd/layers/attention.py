@@ -108,6 +108,8 @@ class MultiheadAttention(nn.Module):
         use_cache=False,
         residual=None,
         position_bias=None,
+        static_cache=None,
+        position=None,
     ):
         """attention_mask: [b, sq, sk] bool/uint8, 1 = MASKED (reference semantics).
 
@@ -120,6 +122,25 @@ class MultiheadAttention(nn.Module):
         If ``residual`` is given, returns hidden + dropout(out + bias) (the
         fused TransformerLayer path); otherwise applies bias+dropout only.
         """
+        if static_cache is not None:
+            # hipGraph-capturable decode: preallocated [b, nh, MAX, hs] caches,
+            # device position/length tensors, in-place KV insert, fused
+            # flash_decode with per-batch kv_len (all shapes static)
+            from ..ops.attention import flash_decode_attn
+
+            ck, cv, kv_len32 = static_cache
+            q, k, v = self._split_heads(self.query_key_value(hidden_states), 3)
+            ck.index_copy_(2, position, k)
+            cv.index_copy_(2, position, v)
+            scale_c = self.norm_factor * (self.coeff if self.coeff else 1.0)
+            ctx = flash_decode_attn(q.contiguous(), ck, cv, scale_c,
+                                    kv_len=kv_len32)
+            b, nh, _, hs = ctx.shape
+            context = ctx.permute(0, 2, 1, 3).reshape(b, 1, nh * hs)
+            out, bias = self.dense(context)
+            return bias_dropout_add(out, bias=bias, residual=residual,
+                                    p=0.0, training=False)
+
         # fused flash path: consumes the qkv buffer with ZERO copies (strided
         # [b, s, nh, hs] views), O lands directly in [b, s, h] layout.
         if (

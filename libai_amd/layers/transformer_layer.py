@@ -135,7 +135,19 @@ class TransformerLayer(nn.Module):
         past_key_value=None,
         use_cache=False,
         position_bias=None,
+        static_cache=None,
+        position=None,
     ):
+        if static_cache is not None:
+            # hipGraph-capturable decode step: fixed-shape path, no cache
+            # tuples returned (KV written in-place into the static buffers)
+            ln1 = self.input_layernorm(hidden_states)
+            residual = ln1 if self.apply_residual_post_layernorm else hidden_states
+            h = self.self_attention(ln1, residual=residual,
+                                    static_cache=static_cache, position=position)
+            ln2 = self.post_attention_layernorm(h)
+            residual = ln2 if self.apply_residual_post_layernorm else h
+            return self.mlp(ln2, residual=residual)
         if past_key_value is not None:
             if self.is_decoder:
                 self_past, cross_past = past_key_value

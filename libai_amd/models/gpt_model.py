@@ -49,7 +49,12 @@ class GPTEmbedding(nn.Module):
 
     def forward(self, input_ids, past_length=0):
         seq_len = input_ids.size(1)
-        pos_ids = self.position_ids[:, past_length : past_length + seq_len]
+        if torch.is_tensor(past_length):
+            # device-tensor position (hipGraph-captured decode: no host ints)
+            pos_ids = past_length.view(1, 1) + torch.arange(
+                seq_len, device=input_ids.device).view(1, -1)
+        else:
+            pos_ids = self.position_ids[:, past_length : past_length + seq_len]
         embeds = self.token_embeddings(input_ids) + self.position_embeddings(pos_ids)
         return self.dropout(embeds)
 
@@ -95,7 +100,13 @@ class Transformer(nn.Module):
             return act_checkpoint(layer, hidden_states, use_reentrant=False)
         return layer(hidden_states, past_key_value=past_key_value, use_cache=use_cache)
 
-    def forward(self, hidden_states, past_key_values=None, use_cache=False):
+    def forward(self, hidden_states, past_key_values=None, use_cache=False,
+                static_caches=None, position=None):
+        if static_caches is not None:
+            for layer, sc in zip(self.layers, static_caches):
+                hidden_states = layer(hidden_states, static_cache=sc,
+                                      position=position)
+            return self.layernorm_f(hidden_states)
         presents = [] if use_cache else None
         for i, layer in enumerate(self.layers):
             past = past_key_values[i] if past_key_values is not None else None
@@ -197,7 +208,15 @@ class GPTModel(nn.Module):
             "amp_enabled": cfg.get("amp_enabled", False),
         }
 
-    def forward(self, input_ids, past_key_values=None, use_cache=False):
+    def forward(self, input_ids, past_key_values=None, use_cache=False,
+                static_caches=None, position=None):
+        if static_caches is not None:
+            # hipGraph-captured decode: all shapes static, position/kv_len are
+            # DEVICE tensors, KV written in place into preallocated buffers
+            h = self.embeddings(input_ids, position)
+            h = self.transformer(h, static_caches=static_caches,
+                                 position=position)
+            return self.lm_head(h, self.embeddings.token_embeddings.weight)
         past_length = (
             past_key_values[0][0].shape[2] if past_key_values is not None else 0
         )
