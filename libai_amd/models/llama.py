@@ -89,7 +89,26 @@ class LlamaAttention(nn.Module):
         return q, kv[:, :, 0], kv[:, :, 1]
 
     def forward(self, hidden_states, past_key_value=None, use_cache=False,
-                residual=None):
+                residual=None, static_cache=None, position=None):
+        if static_cache is not None:
+            # hipGraph-capturable decode: RoPE at a device position tensor,
+            # in-place KV insert into [b, kvh, MAX, hd] buffers, fused
+            # flash_decode masked by device int32 kv_len (GQA in-kernel)
+            from ..ops.attention import flash_decode_attn
+
+            ck, cv, kv32 = static_cache
+            b, _, _ = hidden_states.shape
+            q, k, v = self._project(hidden_states)
+            q = apply_rotary_pos_emb(q, self.max_pos, self.rope_theta, position)
+            k = apply_rotary_pos_emb(k, self.max_pos, self.rope_theta, position)
+            ck.index_copy_(2, position, k.permute(0, 2, 1, 3))
+            cv.index_copy_(2, position, v.permute(0, 2, 1, 3))
+            ctx = flash_decode_attn(q.permute(0, 2, 1, 3).contiguous(), ck, cv,
+                                    self.scale, kv_len=kv32)
+            context = ctx.permute(0, 2, 1, 3).reshape(
+                b, 1, self.num_heads_local * self.head_dim)
+            out, _ = self.o_proj(context)
+            return out + residual if residual is not None else out
         if self.sequence_parallel and (use_cache or past_key_value is not None):
             raise RuntimeError("sequence_parallel is training-only; build the "
                                "generation model with sequence_parallel=False")
@@ -208,7 +227,14 @@ class LlamaDecoderLayer(nn.Module):
                 for p in ln.parameters():
                     p.sequence_parallel_grad = True
 
-    def forward(self, hidden_states, past_key_value=None, use_cache=False):
+    def forward(self, hidden_states, past_key_value=None, use_cache=False,
+                static_cache=None, position=None):
+        if static_cache is not None:
+            ln1 = self.input_layernorm(hidden_states)
+            h = self.self_attn(ln1, residual=hidden_states,
+                               static_cache=static_cache, position=position)
+            ln2 = self.post_attention_layernorm(h)
+            return self.mlp(ln2, residual=h)
         ln1 = self.input_layernorm(hidden_states)
         attn_out = self.self_attn(ln1, past_key_value=past_key_value,
                                   use_cache=use_cache, residual=hidden_states)
@@ -296,7 +322,13 @@ class LlamaModel(nn.Module):
             return act_checkpoint(layer, h, use_reentrant=False)
         return layer(h, past_key_value=past, use_cache=use_cache)
 
-    def forward(self, input_ids, past_key_values=None, use_cache=False):
+    def forward(self, input_ids, past_key_values=None, use_cache=False,
+                static_caches=None, position=None):
+        if static_caches is not None:
+            h = self.embed_tokens(input_ids)
+            for layer, sc in zip(self.layers, static_caches):
+                h = layer(h, static_cache=sc, position=position)
+            return self.norm(h)
         h = self.embed_tokens(input_ids)
         if self.sequence_parallel and not use_cache and past_key_values is None:
             from ..parallel.comm import scatter_to_sequence_parallel_region
@@ -373,7 +405,12 @@ class LlamaForCausalLM(nn.Module):
         return (self.model.embed_tokens.weight if self.tie_word_embeddings
                 else self.lm_head.weight)
 
-    def forward(self, input_ids, labels=None, past_key_values=None, use_cache=False):
+    def forward(self, input_ids, labels=None, past_key_values=None, use_cache=False,
+                static_caches=None, position=None):
+        if static_caches is not None:
+            h = self.model(input_ids, static_caches=static_caches,
+                           position=position)
+            return self.lm_logits(h, self._head_weight())
         h = self.model(input_ids, past_key_values=past_key_values, use_cache=use_cache)
         if use_cache:
             h, presents = h

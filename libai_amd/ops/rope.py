@@ -52,8 +52,22 @@ def _ref_rope(x, cos_t, sin_t, pos0):
 
 
 def apply_rotary_pos_emb(x, max_seq, theta=10000.0, pos0=0):
-    """x: [b, s, nh, hs] (strided views ok) -> rotated contiguous tensor."""
+    """x: [b, s, nh, hs] (strided views ok) -> rotated contiguous tensor.
+
+    ``pos0`` may be a DEVICE int64 tensor (hipGraph-captured decode: the
+    position advances on-device, so the table rows are gathered with
+    index_select instead of a host slice)."""
     cos_t, sin_t = _tables(max_seq, x.shape[-1], theta, x.device)
+    if torch.is_tensor(pos0):
+        s = x.shape[1]
+        idx = pos0.view(1) + torch.arange(s, device=x.device)
+        cos = cos_t.index_select(0, idx)[None, :, None, :]
+        sin = sin_t.index_select(0, idx)[None, :, None, :]
+        half = x.shape[-1] // 2
+        # fp32 math + one final cast: matches the HIP kernel's rounding
+        x1, x2 = x[..., :half].float(), x[..., half:].float()
+        return torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin],
+                         dim=-1).to(x.dtype)
     if use_hip(x):
         return _RopeFn.apply(x, cos_t, sin_t, pos0)
     return _ref_rope(x, cos_t, sin_t, pos0)

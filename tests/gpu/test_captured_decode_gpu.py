@@ -97,3 +97,32 @@ def test_captured_decode_speed():
           f"ms/tok ({t_eag / t_cap:.2f}x)")
     # the captured step must not be SLOWER than eager on a launch-bound model
     assert t_cap < t_eag
+
+
+def _tiny_llama():
+    from libai_amd.models.llama import LlamaForCausalLM
+
+    torch.manual_seed(1)
+    m = LlamaForCausalLM(
+        hidden_layers=4, vocab_size=1024, hidden_size=256,
+        intermediate_size=512, num_attention_heads=4,
+        num_key_value_heads=2, max_position_embeddings=256,
+    )
+    return m.to("cuda", torch.bfloat16).eval()
+
+
+def test_captured_llama_decode_matches_eager():
+    """GQA KV caches + RoPE at a device position under hipGraph capture."""
+    from libai_amd.inference.captured_decode import CapturedLlamaDecoder
+
+    model = _tiny_llama()
+    b, L, n_new = 4, 32, 24
+    prompt = torch.randint(0, 1024, (b, L), device="cuda")
+    ref = _eager_greedy(model, prompt, n_new)
+    dec = CapturedLlamaDecoder(model, max_batch=b, max_seq_len=256)
+    got = dec.generate(prompt, n_new)
+    assert torch.equal(got, ref), f"mismatch:\n{got}\nvs\n{ref}"
+
+    prompt2 = torch.randint(0, 1024, (b, L), device="cuda")
+    assert torch.equal(dec.generate(prompt2, n_new),
+                       _eager_greedy(model, prompt2, n_new))

@@ -84,28 +84,30 @@ def main():
           f"({args.batch / t_fused:.0f} tok/s)  "
           f"[{100 * (t_unfused - t_fused) / t_unfused:+.1f}% step time]")
 
-    if args.model == "gpt2":
-        # hipGraph-captured step (GPT path): one replay per token
-        from libai_amd.inference.captured_decode import CapturedGPTDecoder
+    # hipGraph-captured step: one replay per token
+    from libai_amd.inference.captured_decode import (
+        CapturedGPTDecoder,
+        CapturedLlamaDecoder,
+    )
 
-        dec = CapturedGPTDecoder(m, max_batch=args.batch,
-                                 max_seq_len=args.ctx + args.steps + 8)
-        dec.generate(ids, args.steps)  # capture + warm
+    cls = CapturedGPTDecoder if args.model == "gpt2" else CapturedLlamaDecoder
+    dec = cls(m, max_batch=args.batch, max_seq_len=args.ctx + args.steps + 8)
+    dec.generate(ids, args.steps)  # capture + warm
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    dec.generate(ids, args.steps)
+    torch.cuda.synchronize()
+    t_cap = (time.perf_counter() - t0) / args.steps
+    # subtract the prefill (measured separately) to isolate the step
+    with torch.no_grad():
+        torch.cuda.synchronize(); t0 = time.perf_counter()
+        m(input_ids=ids, use_cache=True)
         torch.cuda.synchronize()
-        t0 = time.perf_counter()
-        dec.generate(ids, args.steps)
-        torch.cuda.synchronize()
-        t_cap = (time.perf_counter() - t0) / args.steps
-        # subtract the prefill (measured separately) to isolate the step
-        with torch.no_grad():
-            torch.cuda.synchronize(); t0 = time.perf_counter()
-            m(input_ids=ids, use_cache=True)
-            torch.cuda.synchronize()
-        t_prefill = time.perf_counter() - t0
-        t_step = (t_cap * args.steps - t_prefill) / args.steps
-        print(f"hipGraph-captured decode:   {t_step * 1e3:.3f} ms/step "
-              f"({args.batch / t_step:.0f} tok/s)  "
-              f"[{100 * (t_fused - t_step) / t_fused:+.1f}% vs fused eager]")
+    t_prefill = time.perf_counter() - t0
+    t_step = (t_cap * args.steps - t_prefill) / args.steps
+    print(f"hipGraph-captured decode:   {t_step * 1e3:.3f} ms/step "
+          f"({args.batch / t_step:.0f} tok/s)  "
+          f"[{100 * (t_fused - t_step) / t_fused:+.1f}% vs fused eager]")
 
 
 if __name__ == "__main__":
