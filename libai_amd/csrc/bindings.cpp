@@ -133,6 +133,10 @@ void rope_bwd_bf16(const void*, void*, const float*, const float*, int64_t, int6
 void rope_bwd_f32(const void*, void*, const float*, const float*, int64_t, int64_t,
                   int64_t, int64_t, int64_t, int64_t, int, int, int, int, int,
                   hipStream_t);
+void rope_kv_insert_bf16(const void*, const void*, const void*, void*, void*,
+                         void*, const float*, const float*, const long long*,
+                         int, int, int, int, int64_t, int64_t, int64_t, int64_t,
+                         int64_t, int64_t, int64_t, int, hipStream_t);
 void swiglu_fwd_bf16(const void*, void*, int64_t, int, hipStream_t);
 void swiglu_fwd_f32(const void*, void*, int64_t, int, hipStream_t);
 void swiglu_bwd_bf16(const void*, const void*, void*, int64_t, int, hipStream_t);
@@ -587,6 +591,35 @@ torch::Tensor rope(torch::Tensor x, torch::Tensor cos_t, torch::Tensor sin_t,
   return out;
 }
 
+// Fused decode-step RoPE + KV insert (hipGraph-captured serving; see
+// kernels/rope_swiglu.hip).  q/k/v are [B, 1, H, HD] strided views; pos is a
+// DEVICE int64 [1] tensor; ck/cv are the [B, KVH, max_len, HD] static caches.
+// Returns rotated q in [B, NH, 1, HD] (flash_decode input layout).
+torch::Tensor rope_kv_insert(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                             torch::Tensor ck, torch::Tensor cv,
+                             torch::Tensor cos_t, torch::Tensor sin_t,
+                             torch::Tensor pos, bool rotate) {
+  TORCH_CHECK(q.dim() == 4 && q.size(1) == 1 && q.stride(3) == 1);
+  TORCH_CHECK(k.dim() == 4 && k.size(1) == 1 && k.stride(3) == 1);
+  TORCH_CHECK(v.dim() == 4 && v.size(1) == 1 && v.stride(3) == 1);
+  TORCH_CHECK(ck.is_contiguous() && cv.is_contiguous());
+  TORCH_CHECK(pos.scalar_type() == torch::kInt64 && pos.is_cuda());
+  TORCH_CHECK(is_bf16(q) && is_bf16(ck), "rope_kv_insert is bf16-only");
+  const int B = (int)q.size(0), NH = (int)q.size(2), HD = (int)q.size(3);
+  const int KVH = (int)k.size(2);
+  TORCH_CHECK(ck.size(0) == B && ck.size(1) == KVH && ck.size(3) == HD);
+  auto qo = torch::empty({B, NH, 1, HD}, q.options());
+  rope_kv_insert_bf16(q.data_ptr(), k.data_ptr(), v.data_ptr(), qo.data_ptr(),
+                      ck.data_ptr(), cv.data_ptr(), cos_t.data_ptr<float>(),
+                      sin_t.data_ptr<float>(),
+                      (const long long*)pos.data_ptr<int64_t>(), B, NH, KVH,
+                      HD, ck.size(2), q.stride(0), q.stride(2), k.stride(0),
+                      k.stride(2), v.stride(0), v.stride(2), rotate ? 1 : 0,
+                      cur_stream());
+  check_launch("rope_kv_insert");
+  return qo;
+}
+
 torch::Tensor swiglu_fwd(torch::Tensor x) {
   CHECK_IN(x);
   const int64_t F = x.size(-1) / 2;
@@ -656,6 +689,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("rope", &rope);
+  m.def("rope_kv_insert", &rope_kv_insert);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("adamw_step", &adamw_step);

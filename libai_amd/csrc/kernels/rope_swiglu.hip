@@ -59,6 +59,68 @@ __global__ void rope_kernel(const typename E::T* __restrict__ x,
 }
 
 // ---------------------------------------------------------------------------
+// Fused decode-step RoPE + KV-cache insert (hipGraph-captured serving step):
+// one launch replaces per-layer {rope(q), rope(k), 2x index_copy, permutes}.
+//   q   [B, 1, NH, HD]  (strided view, last dim contiguous) -> rotated into
+//   qo  [B, NH, 1, HD]  contiguous (flash_decode input layout)
+//   k   [B, 1, KVH, HD] -> rotated into ck[b, h, pos, :]
+//   v   [B, 1, KVH, HD] -> copied  into cv[b, h, pos, :]
+// pos is a DEVICE int64 pointer (position advances on-device under capture).
+// One wavefront per (b, head) row; rows ordered q-heads, k-heads, v-heads.
+// ---------------------------------------------------------------------------
+template <class E, bool ROT>
+__global__ void rope_kv_insert_kernel(
+    const typename E::T* __restrict__ q, const typename E::T* __restrict__ k,
+    const typename E::T* __restrict__ v, typename E::T* __restrict__ qo,
+    typename E::T* __restrict__ ck, typename E::T* __restrict__ cv,
+    const float* __restrict__ cos_t, const float* __restrict__ sin_t,
+    const long long* __restrict__ pos_p, int B, int NH, int KVH, int HD,
+    int64_t max_len, int64_t q_sb, int64_t q_sh, int64_t k_sb, int64_t k_sh,
+    int64_t v_sb, int64_t v_sh) {
+  const int64_t pos = pos_p[0];
+  const int half = HD / 2;
+  const int rows_per_b = NH + 2 * KVH;
+  const int wave = (int)((blockIdx.x * (int64_t)blockDim.x + threadIdx.x) >> 6);
+  const int lane = threadIdx.x & 63;
+  if (wave >= B * rows_per_b) return;
+  const int b = wave / rows_per_b;
+  int r = wave % rows_per_b;
+  const float* cp = cos_t + pos * half;
+  const float* sp = sin_t + pos * half;
+  if (r < NH) {  // q head: rotate (or copy) -> qo[b, r, 0, :]
+    const typename E::T* xp = q + b * q_sb + r * q_sh;
+    typename E::T* op = qo + ((int64_t)b * NH + r) * HD;
+    if constexpr (ROT) {
+      for (int j = lane; j < half; j += 64) {
+        float a = E::to_f(xp[j]), bb = E::to_f(xp[half + j]);
+        op[j] = E::from_f(a * cp[j] - bb * sp[j]);
+        op[half + j] = E::from_f(bb * cp[j] + a * sp[j]);
+      }
+    } else {
+      for (int j = lane; j < HD; j += 64) op[j] = xp[j];
+    }
+  } else if (r < NH + KVH) {  // k head: rotate (or copy) -> ck[b, h, pos, :]
+    const int h = r - NH;
+    const typename E::T* xp = k + b * k_sb + h * k_sh;
+    typename E::T* op = ck + (((int64_t)b * KVH + h) * max_len + pos) * HD;
+    if constexpr (ROT) {
+      for (int j = lane; j < half; j += 64) {
+        float a = E::to_f(xp[j]), bb = E::to_f(xp[half + j]);
+        op[j] = E::from_f(a * cp[j] - bb * sp[j]);
+        op[half + j] = E::from_f(bb * cp[j] + a * sp[j]);
+      }
+    } else {
+      for (int j = lane; j < HD; j += 64) op[j] = xp[j];
+    }
+  } else {  // v head: copy -> cv[b, h, pos, :]
+    const int h = r - NH - KVH;
+    const typename E::T* xp = v + b * v_sb + h * v_sh;
+    typename E::T* op = cv + (((int64_t)b * KVH + h) * max_len + pos) * HD;
+    for (int j = lane; j < HD; j += 64) op[j] = xp[j];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // SwiGLU: y = silu(g) * u, with g = x[..., :F], u = x[..., F:2F]
 // ---------------------------------------------------------------------------
 template <class E, bool BWD>
@@ -149,3 +211,23 @@ inline int64_t grid_for(int64_t n) {
 
 ROPE_LAUNCHERS(bf16, BF16Elem)
 ROPE_LAUNCHERS(f32, F32Elem)
+
+extern "C" void rope_kv_insert_bf16(
+    const void* q, const void* k, const void* v, void* qo, void* ck, void* cv,
+    const float* cos_t, const float* sin_t, const long long* pos_p, int B,
+    int NH, int KVH, int HD, int64_t max_len, int64_t q_sb, int64_t q_sh,
+    int64_t k_sb, int64_t k_sh, int64_t v_sb, int64_t v_sh,
+    int rotate, hipStream_t stream) {
+  const int waves = B * (NH + 2 * KVH);
+  const int blocks = CDIV(waves, 4);  // 4 waves (256 threads) per block
+  if (rotate)
+    rope_kv_insert_kernel<BF16Elem, true><<<dim3(blocks), 256, 0, stream>>>(
+        (const BF16Elem::T*)q, (const BF16Elem::T*)k, (const BF16Elem::T*)v,
+        (BF16Elem::T*)qo, (BF16Elem::T*)ck, (BF16Elem::T*)cv, cos_t, sin_t,
+        pos_p, B, NH, KVH, HD, max_len, q_sb, q_sh, k_sb, k_sh, v_sb, v_sh);
+  else
+    rope_kv_insert_kernel<BF16Elem, false><<<dim3(blocks), 256, 0, stream>>>(
+        (const BF16Elem::T*)q, (const BF16Elem::T*)k, (const BF16Elem::T*)v,
+        (BF16Elem::T*)qo, (BF16Elem::T*)ck, (BF16Elem::T*)cv, cos_t, sin_t,
+        pos_p, B, NH, KVH, HD, max_len, q_sb, q_sh, k_sb, k_sh, v_sb, v_sh);
+}

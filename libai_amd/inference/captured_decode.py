@@ -60,12 +60,15 @@ class _CapturedDecoderBase:
             raise RuntimeError("captured decode is GPU-only")
         self.device = dev
         b = max_batch
+        # ONE shared kv_len tensor (all layers decode in lockstep): its
+        # advance is a single captured kernel, not one per layer
+        self.kv32 = torch.zeros(b, dtype=torch.int32, device=dev)
         self.caches = [
             (torch.zeros(b, n_kv, max_seq_len, head_size, device=dev,
                          dtype=torch.bfloat16),
              torch.zeros(b, n_kv, max_seq_len, head_size, device=dev,
                          dtype=torch.bfloat16),
-             torch.zeros(b, dtype=torch.int32, device=dev))
+             self.kv32)
             for _ in range(self._n_layers(model))
         ]
         self.pos = torch.zeros(1, dtype=torch.int64, device=dev)
@@ -97,15 +100,13 @@ class _CapturedDecoderBase:
         self.static_tok.copy_(nxt)
         self.pos.add_(1)
         self.step_idx.add_(1)
-        for _, _, kv32 in self.caches:
-            kv32.add_(1)
+        self.kv32.add_(1)
 
     def _reset_to(self, prompt_len, first_tok):
         self.pos.fill_(prompt_len)
         self.step_idx.fill_(1)
         self.static_tok.copy_(first_tok)
-        for _, _, kv32 in self.caches:
-            kv32.fill_(prompt_len + 1)  # length AFTER this step's insert
+        self.kv32.fill_(prompt_len + 1)  # length AFTER this step's insert
 
     @torch.no_grad()
     def generate(self, prompt_ids, max_new_tokens):

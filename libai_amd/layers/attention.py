@@ -23,6 +23,17 @@ from torch import nn
 
 from ..ops.fused_bias import bias_dropout_add
 from ..ops.softmax import fused_scale_mask_softmax
+
+_DUMMY_TABS = {}
+
+
+def _dummy_tab(device):
+    """1-element fp32 placeholder for the no-rotate kv_insert call (the
+    kernel never reads cos/sin when rotate=False)."""
+    key = str(device)
+    if key not in _DUMMY_TABS:
+        _DUMMY_TABS[key] = torch.zeros(1, dtype=torch.float32, device=device)
+    return _DUMMY_TABS[key]
 from ..utils import distributed as du
 from .linear import Linear1D
 
@@ -124,18 +135,24 @@ class MultiheadAttention(nn.Module):
         """
         if static_cache is not None:
             # hipGraph-capturable decode: preallocated [b, nh, MAX, hs] caches,
-            # device position/length tensors, in-place KV insert, fused
-            # flash_decode with per-batch kv_len (all shapes static)
+            # device position/length tensors; ONE kv_insert kernel (no-rotate
+            # variant of K17's decode fusion) moves q to flash layout and
+            # writes k/v into the cache row at the device position; fused
+            # flash_decode masks by per-batch kv_len (all shapes static)
+            from ..ops._ext import ext
             from ..ops.attention import flash_decode_attn
 
             ck, cv, kv_len32 = static_cache
-            q, k, v = self._split_heads(self.query_key_value(hidden_states), 3)
-            ck.index_copy_(2, position, k)
-            cv.index_copy_(2, position, v)
+            qkv = self.query_key_value(hidden_states)
+            b, s, _ = hidden_states.shape
+            nh, hs = self.num_heads_local, self.head_size
+            qkv4 = qkv.view(b, s, nh, 3 * hs)
+            q, k, v = (qkv4[..., :hs], qkv4[..., hs:2 * hs], qkv4[..., 2 * hs:])
+            dummy = _dummy_tab(hidden_states.device)
+            qo = ext().rope_kv_insert(q, k, v, ck, cv, dummy, dummy,
+                                      position, False)
             scale_c = self.norm_factor * (self.coeff if self.coeff else 1.0)
-            ctx = flash_decode_attn(q.contiguous(), ck, cv, scale_c,
-                                    kv_len=kv_len32)
-            b, nh, _, hs = ctx.shape
+            ctx = flash_decode_attn(qo, ck, cv, scale_c, kv_len=kv_len32)
             context = ctx.permute(0, 2, 1, 3).reshape(b, 1, nh * hs)
             out, bias = self.dense(context)
             return bias_dropout_add(out, bias=bias, residual=residual,

@@ -94,17 +94,19 @@ class LlamaAttention(nn.Module):
             # hipGraph-capturable decode: RoPE at a device position tensor,
             # in-place KV insert into [b, kvh, MAX, hd] buffers, fused
             # flash_decode masked by device int32 kv_len (GQA in-kernel)
+            from ..ops._ext import ext
             from ..ops.attention import flash_decode_attn
+            from ..ops.rope import _tables
 
             ck, cv, kv32 = static_cache
             b, _, _ = hidden_states.shape
             q, k, v = self._project(hidden_states)
-            q = apply_rotary_pos_emb(q, self.max_pos, self.rope_theta, position)
-            k = apply_rotary_pos_emb(k, self.max_pos, self.rope_theta, position)
-            ck.index_copy_(2, position, k.permute(0, 2, 1, 3))
-            cv.index_copy_(2, position, v.permute(0, 2, 1, 3))
-            ctx = flash_decode_attn(q.permute(0, 2, 1, 3).contiguous(), ck, cv,
-                                    self.scale, kv_len=kv32)
+            # ONE kernel: RoPE(q) -> flash layout, RoPE(k)/copy(v) -> cache
+            # row at the device position (replaces ~15 eager-shaped launches)
+            cos_t, sin_t = _tables(self.max_pos, self.head_dim,
+                                   self.rope_theta, hidden_states.device)
+            qo = ext().rope_kv_insert(q, k, v, ck, cv, cos_t, sin_t, position, True)
+            ctx = flash_decode_attn(qo, ck, cv, self.scale, kv_len=kv32)
             context = ctx.permute(0, 2, 1, 3).reshape(
                 b, 1, self.num_heads_local * self.head_dim)
             out, _ = self.o_proj(context)
