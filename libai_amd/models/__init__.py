@@ -5,6 +5,7 @@ from .contrastive import MoCoV3, SimCSEModel
 from .build import build_model
 from .gpt_model import GPTForPreTraining, GPTModel
 from .llama import LlamaForCausalLM, LlamaModel
+from .convnext import ConvNeXt
 from .mae import MAEForPreTraining
 from .palm import PaLMForCausalLM, PaLMModel
 from .resmlp import ResMLP
@@ -32,6 +33,7 @@ __all__ = [
     "ResMLP",
     "BloomModel",
     "BloomForCausalLM",
+    "ConvNeXt",
     "MAEForPreTraining",
     "SimCSEModel",
     "MoCoV3",

@@ -257,3 +257,15 @@ def test_palm_gpu_bf16():
                         max_position_embeddings=128).to(torch.bfloat16).cuda()
     ids = torch.randint(0, 1024, (2, 65), device="cuda")
     _step(m, dict(input_ids=ids[:, :-1], labels=ids[:, 1:]))
+
+
+def test_convnext_gpu_bf16():
+    from libai_amd.models import ConvNeXt
+
+    torch.manual_seed(0)
+    m = ConvNeXt(img_size=64, num_classes=16, depths=(1, 1, 2, 1),
+                 dims=(32, 64, 128, 256),
+                 drop_path_rate=0.1).to(torch.bfloat16).cuda()
+    imgs = torch.randn(2, 3, 64, 64, device="cuda", dtype=torch.bfloat16)
+    labels = torch.randint(0, 16, (2,), device="cuda")
+    _step(m, dict(images=imgs, labels=labels))
