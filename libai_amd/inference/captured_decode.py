@@ -31,7 +31,8 @@ eager; Llama-1B (GQA+RoPE via the fused rope_kv_insert kernel) 3.83 vs
 
 import torch
 
-__all__ = ["CapturedGPTDecoder", "CapturedLlamaDecoder"]
+__all__ = ["CapturedGPTDecoder", "CapturedLlamaDecoder",
+           "SamplingMixin", "CapturedGPTSampler", "CapturedLlamaSampler"]
 
 
 class _CapturedDecoderBase:
@@ -94,10 +95,14 @@ class _CapturedDecoderBase:
         """Run one static step on self.static_tok -> [b, 1, vocab]."""
         raise NotImplementedError
 
+    # -- token selection (overridden by the sampling mixin) ---------------
+    def _select(self, logits):
+        return logits.argmax(dim=-1, keepdim=True)  # greedy
+
     # -- the one decode step (the thing that gets captured) ---------------
     def _step(self):
         logits = self._logits()
-        nxt = logits[:, -1, :].argmax(dim=-1, keepdim=True)  # [b, 1]
+        nxt = self._select(logits[:, -1, :])  # [b, 1]
         self.out_tokens.index_copy_(1, self.step_idx, nxt)
         self.static_tok.copy_(nxt)
         self.pos.add_(1)
@@ -127,7 +132,7 @@ class _CapturedDecoderBase:
         for (ck, cv, _), (pk, pv) in zip(self.caches, past):
             ck[:, :, :L].copy_(pk)
             cv[:, :, :L].copy_(pv)
-        first = logits[:, -1, :].argmax(dim=-1, keepdim=True)  # token at pos L
+        first = self._select(logits[:, -1, :])  # token at pos L
 
         if self.out_tokens is None or self.out_tokens.shape[1] < max_new_tokens:
             self.out_tokens = torch.zeros(self.max_batch, max_new_tokens,
@@ -196,3 +201,42 @@ class CapturedLlamaDecoder(_CapturedDecoderBase):
     def _logits(self):
         return self.model(self.static_tok, static_caches=self.caches,
                           position=self.pos)
+
+
+class SamplingMixin:
+    """Temperature / top-k sampling INSIDE the captured graph.
+
+    torch's philox generator is hipGraph-aware (the RNG offset lives in
+    graph-owned device memory), so `torch.multinomial` replays with fresh
+    randomness each replay — the whole sampled decode loop stays on-device.
+    Set temperature/top_k before the first generate() (they are captured).
+    """
+
+    temperature = 1.0
+    top_k = 0  # 0 = no top-k filter
+
+    def _select(self, logits):
+        logits = logits.float()
+        if self.temperature != 1.0:
+            logits = logits / self.temperature
+        if self.top_k:
+            kth = logits.topk(self.top_k, dim=-1).values[..., -1:]
+            logits = logits.masked_fill(logits < kth, float("-inf"))
+        probs = torch.softmax(logits, dim=-1)
+        return torch.multinomial(probs, 1)
+
+
+class CapturedGPTSampler(SamplingMixin, CapturedGPTDecoder):
+    def __init__(self, model, max_batch, max_seq_len, temperature=1.0,
+                 top_k=0):
+        super().__init__(model, max_batch, max_seq_len)
+        self.temperature = temperature
+        self.top_k = top_k
+
+
+class CapturedLlamaSampler(SamplingMixin, CapturedLlamaDecoder):
+    def __init__(self, model, max_batch, max_seq_len, temperature=1.0,
+                 top_k=0):
+        super().__init__(model, max_batch, max_seq_len)
+        self.temperature = temperature
+        self.top_k = top_k

@@ -372,3 +372,56 @@ def test_gemm_dw_matches_reference():
             got_s = ext().gemm_dw(dy, x, s)
             rel = (got_s.float() - ref).abs().max() / ref.abs().max()
             assert rel.item() < 2e-2, f"splits={s}: rel {rel.item()}"
+
+
+def test_rope_kv_insert_matches_reference():
+    """Fused decode-step RoPE+KV-insert (K17 decode fusion) vs fp32 python
+    rotation on STRIDED q/k/v views; untouched cache rows stay untouched."""
+    from libai_amd.ops._ext import ext
+    from libai_amd.ops.rope import _tables
+
+    torch.manual_seed(0)
+    B, NH, KVH, HD, MAX = 3, 8, 2, 64, 50
+    pos_i = 7
+    # strided views carved from a fused-projection-like buffer
+    buf = torch.randn(B, 1, NH * HD + 2 * KVH * HD, device="cuda",
+                      dtype=torch.bfloat16)
+    q = buf[..., : NH * HD].view(B, 1, NH, HD)
+    k = buf[..., NH * HD : NH * HD + KVH * HD].view(B, 1, KVH, HD)
+    v = buf[..., NH * HD + KVH * HD :].view(B, 1, KVH, HD)
+    ck = torch.randn(B, KVH, MAX, HD, device="cuda", dtype=torch.bfloat16)
+    cv = torch.randn_like(ck)
+    ck0, cv0 = ck.clone(), cv.clone()
+    pos = torch.tensor([pos_i], device="cuda", dtype=torch.int64)
+    cos_t, sin_t = _tables(MAX, HD, 10000.0, torch.device("cuda"))
+
+    qo = ext().rope_kv_insert(q, k, v, ck, cv, cos_t, sin_t, pos, True)
+
+    def rot(x):  # fp32 rotate_half at pos_i
+        cos = cos_t[pos_i].float()
+        sin = sin_t[pos_i].float()
+        x1, x2 = x[..., : HD // 2].float(), x[..., HD // 2 :].float()
+        return torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin],
+                         dim=-1).to(torch.bfloat16)
+
+    # rotation compared with tight tolerance (kernel may contract to FMA;
+    # 1-ulp fp32 difference can flip the bf16 rounding on exact ties)
+    assert torch.allclose(qo.float(), rot(q).permute(0, 2, 1, 3).float(),
+                          atol=2e-2, rtol=1e-2)
+    assert torch.allclose(ck[:, :, pos_i].float(), rot(k).squeeze(1).float(),
+                          atol=2e-2, rtol=1e-2)
+    assert torch.equal(cv[:, :, pos_i], v.squeeze(1))
+    # every other row untouched
+    mask = torch.ones(MAX, dtype=torch.bool)
+    mask[pos_i] = False
+    assert torch.equal(ck[:, :, mask], ck0[:, :, mask])
+    assert torch.equal(cv[:, :, mask], cv0[:, :, mask])
+
+    # no-rotate variant (GPT path): q transposed, k/v copied verbatim
+    ck.copy_(ck0)
+    cv.copy_(cv0)
+    dummy = torch.zeros(1, device="cuda", dtype=torch.float32)
+    qo2 = ext().rope_kv_insert(q, k, v, ck, cv, dummy, dummy, pos, False)
+    assert torch.equal(qo2, q.permute(0, 2, 1, 3))
+    assert torch.equal(ck[:, :, pos_i], k.squeeze(1))
+    assert torch.equal(cv[:, :, pos_i], v.squeeze(1))
