@@ -293,6 +293,14 @@ class DefaultTrainer(TrainerBase):
                                      max_iter=self.max_iter, max_to_keep=max_to_keep)
             ),
         ]
+        prof = try_get_key(cfg, "train.profiler", default=None)
+        if prof:
+            ret.append(hooks.TorchProfilerHook(
+                os.path.join(cfg.train.output_dir, "profiler"),
+                start_iter=prof.get("start_iter", 10),
+                end_iter=prof.get("end_iter", 13),
+                with_stack=prof.get("with_stack", False),
+            ))
         eval_period = try_get_key(cfg, "train.evaluation.eval_period", default=0)
         if eval_period and try_get_key(cfg, "train.evaluation.enabled", default=True):
             def _eval():

@@ -86,3 +86,25 @@ def test_periodic_checkpointer_gc(tmp_path):
     # at most 2 periodic checkpoints retained
     assert len(kept) == 2, kept
     assert kept[-1] == "model_0000003"
+
+
+def test_torch_profiler_hook(tmp_path):
+    """TorchProfilerHook captures [start, end) and writes a chrome trace."""
+    import torch
+    from libai_amd.engine.hooks import TorchProfilerHook
+    from libai_amd.engine.trainer import HookBase  # noqa: F401
+
+    class _T:
+        iter = 0
+
+    hook = TorchProfilerHook(str(tmp_path), start_iter=1, end_iter=3)
+    hook.trainer = _T()
+    x = torch.randn(8, 8)
+    for it in range(5):
+        hook.trainer.iter = it
+        hook.before_step()
+        (x @ x).sum()
+        hook.after_step()
+    traces = list(tmp_path.glob("rank*_trace.json"))
+    assert len(traces) == 1 and traces[0].stat().st_size > 0
+    assert hook._prof is None  # closed, one-shot
