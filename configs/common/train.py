@@ -34,4 +34,7 @@ train = ConfigDict(
     ),
     rdma_enabled=False,
     scheduler=None,
+    # torch.profiler window: dict(start_iter=10, end_iter=13[, with_stack])
+    # -> chrome traces in <output_dir>/profiler/ (None = off)
+    profiler=None,
 )
