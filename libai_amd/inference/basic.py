@@ -95,7 +95,31 @@ class TextGenerationPipeline(BasePipeline):
         return {"input_ids": batch.to(self.device)}
 
     def forward(self, model_inputs, max_length=64, do_sample=False, top_k=0,
-                top_p=1.0, temperature=1.0, num_beams=1, **kwargs):
+                top_p=1.0, temperature=1.0, num_beams=1, captured=False,
+                **kwargs):
+        if captured:
+            # hipGraph-captured serving loop (GPU, TP=1, greedy or top-k
+            # sampling; 2x the eager step — profiles/decode_captured.md)
+            from .captured_decode import (
+                CapturedGPTDecoder,
+                CapturedGPTSampler,
+                CapturedLlamaDecoder,
+                CapturedLlamaSampler,
+            )
+
+            ids = model_inputs["input_ids"]
+            is_gpt = hasattr(self.model, "GPT_model")
+            if do_sample:
+                cls = CapturedGPTSampler if is_gpt else CapturedLlamaSampler
+                dec = cls(self.model, max_batch=ids.shape[0],
+                          max_seq_len=max_length,
+                          temperature=temperature, top_k=top_k)
+            else:
+                cls = CapturedGPTDecoder if is_gpt else CapturedLlamaDecoder
+                dec = cls(self.model, max_batch=ids.shape[0],
+                          max_seq_len=max_length)
+            new_toks = dec.generate(ids, max_length - ids.shape[1])
+            return {"sequences": torch.cat([ids, new_toks], dim=1)}
         gen = Generator(self.model)
         eos = (
             self.tokenizer.convert_tokens_to_ids(self.tokenizer.eos_token)
