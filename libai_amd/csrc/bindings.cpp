@@ -136,7 +136,7 @@ void rope_bwd_f32(const void*, void*, const float*, const float*, int64_t, int64
 void rope_kv_insert_bf16(const void*, const void*, const void*, void*, void*,
                          void*, const float*, const float*, const long long*,
                          int, int, int, int, int64_t, int64_t, int64_t, int64_t,
-                         int64_t, int64_t, int64_t, int, hipStream_t);
+                         int64_t, int64_t, int64_t, int, int, hipStream_t);
 void swiglu_fwd_bf16(const void*, void*, int64_t, int, hipStream_t);
 void swiglu_fwd_f32(const void*, void*, int64_t, int, hipStream_t);
 void swiglu_bwd_bf16(const void*, const void*, void*, int64_t, int, hipStream_t);
@@ -604,6 +604,9 @@ torch::Tensor rope_kv_insert(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   TORCH_CHECK(v.dim() == 4 && v.size(1) == 1 && v.stride(3) == 1);
   TORCH_CHECK(ck.is_contiguous() && cv.is_contiguous());
   TORCH_CHECK(pos.scalar_type() == torch::kInt64 && pos.is_cuda());
+  TORCH_CHECK(pos.numel() == 1 || pos.numel() == q.size(0),
+              "pos must be scalar or per-batch");
+  TORCH_CHECK(pos.is_contiguous());
   TORCH_CHECK(is_bf16(q) && is_bf16(ck), "rope_kv_insert is bf16-only");
   const int B = (int)q.size(0), NH = (int)q.size(2), HD = (int)q.size(3);
   const int KVH = (int)k.size(2);
@@ -615,7 +618,7 @@ torch::Tensor rope_kv_insert(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                       (const long long*)pos.data_ptr<int64_t>(), B, NH, KVH,
                       HD, ck.size(2), q.stride(0), q.stride(2), k.stride(0),
                       k.stride(2), v.stride(0), v.stride(2), rotate ? 1 : 0,
-                      cur_stream());
+                      pos.numel() > 1 ? 1 : 0, cur_stream());
   check_launch("rope_kv_insert");
   return qo;
 }

@@ -76,8 +76,7 @@ __global__ void rope_kv_insert_kernel(
     const float* __restrict__ cos_t, const float* __restrict__ sin_t,
     const long long* __restrict__ pos_p, int B, int NH, int KVH, int HD,
     int64_t max_len, int64_t q_sb, int64_t q_sh, int64_t k_sb, int64_t k_sh,
-    int64_t v_sb, int64_t v_sh) {
-  const int64_t pos = pos_p[0];
+    int64_t v_sb, int64_t v_sh, int per_batch_pos) {
   const int half = HD / 2;
   const int rows_per_b = NH + 2 * KVH;
   const int wave = (int)((blockIdx.x * (int64_t)blockDim.x + threadIdx.x) >> 6);
@@ -85,6 +84,10 @@ __global__ void rope_kv_insert_kernel(
   if (wave >= B * rows_per_b) return;
   const int b = wave / rows_per_b;
   int r = wave % rows_per_b;
+  // per-slot position (continuous batching: each sequence decodes at its
+  // own offset); clamp guards an idle slot that over-advanced
+  const int64_t pos = pos_p[per_batch_pos ? b : 0];
+  if (pos < 0 || pos >= max_len) return;
   const float* cp = cos_t + pos * half;
   const float* sp = sin_t + pos * half;
   if (r < NH) {  // q head: rotate (or copy) -> qo[b, r, 0, :]
@@ -217,17 +220,19 @@ extern "C" void rope_kv_insert_bf16(
     const float* cos_t, const float* sin_t, const long long* pos_p, int B,
     int NH, int KVH, int HD, int64_t max_len, int64_t q_sb, int64_t q_sh,
     int64_t k_sb, int64_t k_sh, int64_t v_sb, int64_t v_sh,
-    int rotate, hipStream_t stream) {
+    int rotate, int per_batch_pos, hipStream_t stream) {
   const int waves = B * (NH + 2 * KVH);
   const int blocks = CDIV(waves, 4);  // 4 waves (256 threads) per block
   if (rotate)
     rope_kv_insert_kernel<BF16Elem, true><<<dim3(blocks), 256, 0, stream>>>(
         (const BF16Elem::T*)q, (const BF16Elem::T*)k, (const BF16Elem::T*)v,
         (BF16Elem::T*)qo, (BF16Elem::T*)ck, (BF16Elem::T*)cv, cos_t, sin_t,
-        pos_p, B, NH, KVH, HD, max_len, q_sb, q_sh, k_sb, k_sh, v_sb, v_sh);
+        pos_p, B, NH, KVH, HD, max_len, q_sb, q_sh, k_sb, k_sh, v_sb, v_sh,
+        per_batch_pos);
   else
     rope_kv_insert_kernel<BF16Elem, false><<<dim3(blocks), 256, 0, stream>>>(
         (const BF16Elem::T*)q, (const BF16Elem::T*)k, (const BF16Elem::T*)v,
         (BF16Elem::T*)qo, (BF16Elem::T*)ck, (BF16Elem::T*)cv, cos_t, sin_t,
-        pos_p, B, NH, KVH, HD, max_len, q_sb, q_sh, k_sb, k_sh, v_sb, v_sh);
+        pos_p, B, NH, KVH, HD, max_len, q_sb, q_sh, k_sb, k_sh, v_sb, v_sh,
+        per_batch_pos);
 }

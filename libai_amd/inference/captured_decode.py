@@ -32,7 +32,8 @@ eager; Llama-1B (GQA+RoPE via the fused rope_kv_insert kernel) 3.83 vs
 import torch
 
 __all__ = ["CapturedGPTDecoder", "CapturedLlamaDecoder",
-           "SamplingMixin", "CapturedGPTSampler", "CapturedLlamaSampler"]
+           "SamplingMixin", "CapturedGPTSampler", "CapturedLlamaSampler",
+           "ContinuousGPTDecoder", "ContinuousLlamaDecoder"]
 
 
 class _CapturedDecoderBase:
@@ -240,3 +241,124 @@ class CapturedLlamaSampler(SamplingMixin, CapturedLlamaDecoder):
         super().__init__(model, max_batch, max_seq_len)
         self.temperature = temperature
         self.top_k = top_k
+
+
+class _ContinuousMixin:
+    """Continuous batching on the captured loop: each slot decodes at its
+    OWN position (per-batch device pos/kv_len; the rope_kv_insert kernel and
+    flash_decode both index them per slot), so requests of different lengths
+    share one replayed graph and new requests are admitted between replays
+    without recapturing.
+
+    Protocol::
+
+        dec = ContinuousGPTDecoder(model, max_batch=8, max_seq_len=1024)
+        dec.add_request(0, prompt_a)           # eager prefill into slot 0
+        dec.step(16)                           # 16 graph replays
+        dec.add_request(1, prompt_b)           # admit mid-flight
+        dec.step(16)
+        toks_a = dec.tokens(0)                 # generated tokens so far
+        dec.release(0)                         # free the slot
+
+    Generated tokens land in a per-slot ring buffer (``ring_cap`` entries):
+    harvest at least every ring_cap steps.  EOS/length policy is the
+    caller's (check tokens(), then release()).  Idle slots decode a parked
+    dummy (pos 0, kv_len 1) whose outputs are ignored.
+    """
+
+    ring_cap = 64
+
+    def _init_cont(self):
+        B, dev = self.max_batch, self.device
+        self.pos = torch.zeros(B, dtype=torch.int64, device=dev)  # per-slot
+        self.slot_step = torch.zeros(B, 1, dtype=torch.int64, device=dev)
+        self.ring = torch.zeros(B, self.ring_cap, dtype=torch.int64,
+                                device=dev)
+        self.kv32.fill_(1)
+        self.graph = None
+
+    def _step(self):
+        logits = self._logits()
+        nxt = self._select(logits[:, -1, :])  # [b, 1]
+        self.ring.scatter_(1, self.slot_step.remainder(self.ring_cap), nxt)
+        self.static_tok.copy_(nxt)
+        self.pos.add_(1)
+        self.kv32.add_(1)
+        self.slot_step.add_(1)
+
+    def _park_all(self):
+        self.pos.zero_()
+        self.kv32.fill_(1)
+        self.slot_step.zero_()
+        self.static_tok.zero_()
+
+    def _capture(self):
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            self._park_all()
+            self._step()
+            self._step()
+        torch.cuda.current_stream().wait_stream(s)
+        self._park_all()
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self._step()
+
+    @torch.no_grad()
+    def add_request(self, slot, prompt_ids):
+        """Eager-prefill ``prompt_ids`` (1-D int64) into ``slot``; the
+        prompt's first generated token lands in the ring at index 0."""
+        self.model.eval()
+        prompt_ids = prompt_ids.view(1, -1).to(self.device)
+        L = prompt_ids.shape[1]
+        assert L + 1 < self.max_seq_len
+        out = self.model(input_ids=prompt_ids, use_cache=True)
+        for (ck, cv, _), (pk, pv) in zip(self.caches, out["past_key_values"]):
+            ck[slot, :, :L].copy_(pk[0])
+            cv[slot, :, :L].copy_(pv[0])
+        first = self._select(out["prediction_scores"][:, -1, :])
+        self.ring[slot, 0] = first[0, 0]
+        self.static_tok[slot, 0] = first[0, 0]
+        self.pos[slot] = L
+        self.kv32[slot] = L + 1
+        self.slot_step[slot] = 1
+
+    def release(self, slot):
+        self.pos[slot] = 0
+        self.kv32[slot] = 1
+        self.slot_step[slot] = 0
+
+    @torch.no_grad()
+    def step(self, n=1):
+        """Replay the captured decode step ``n`` times (every active slot
+        gains ``n`` tokens; idle slots burn a parked dummy row)."""
+        if self.graph is None:
+            self._capture()
+        for _ in range(n):
+            self.graph.replay()
+
+    def tokens(self, slot):
+        """Generated tokens for ``slot`` so far (host sync; ring must not
+        have wrapped — harvest at least every ring_cap steps)."""
+        n = int(self.slot_step[slot])
+        if n > self.ring_cap:
+            raise RuntimeError(
+                f"slot {slot} generated {n} > ring_cap={self.ring_cap} "
+                "tokens since admission; harvest more often")
+        return self.ring[slot, :n].clone()
+
+    def generate(self, *a, **k):
+        raise RuntimeError("continuous decoder: use add_request/step/tokens")
+
+
+class ContinuousGPTDecoder(_ContinuousMixin, CapturedGPTDecoder):
+    def __init__(self, model, max_batch, max_seq_len):
+        super().__init__(model, max_batch, max_seq_len)
+        self._init_cont()
+
+
+class ContinuousLlamaDecoder(_ContinuousMixin, CapturedLlamaDecoder):
+    def __init__(self, model, max_batch, max_seq_len):
+        super().__init__(model, max_batch, max_seq_len)
+        self._init_cont()

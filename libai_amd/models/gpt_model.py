@@ -51,7 +51,7 @@ class GPTEmbedding(nn.Module):
         seq_len = input_ids.size(1)
         if torch.is_tensor(past_length):
             # device-tensor position (hipGraph-captured decode: no host ints)
-            pos_ids = past_length.view(1, 1) + torch.arange(
+            pos_ids = past_length.view(-1, 1) + torch.arange(
                 seq_len, device=input_ids.device).view(1, -1)
         else:
             pos_ids = self.position_ids[:, past_length : past_length + seq_len]
