@@ -293,14 +293,21 @@ class _ContinuousMixin:
         self.static_tok.zero_()
 
     def _capture(self):
+        # warm up on the LIVE state (admissions may already have happened):
+        # the two warmup steps scribble k/v at each slot's current pos and
+        # pos+1, which the first two real replays rewrite before kv_len
+        # covers them — so only the small advance tensors need restoring
+        snap = [t.clone() for t in (self.pos, self.kv32, self.slot_step,
+                                    self.static_tok, self.ring)]
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
-            self._park_all()
             self._step()
             self._step()
         torch.cuda.current_stream().wait_stream(s)
-        self._park_all()
+        for t, sv in zip((self.pos, self.kv32, self.slot_step,
+                          self.static_tok, self.ring), snap):
+            t.copy_(sv)
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
             self._step()
