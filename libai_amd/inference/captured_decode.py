@@ -360,12 +360,14 @@ class _ContinuousMixin:
 
 
 class ContinuousGPTDecoder(_ContinuousMixin, CapturedGPTDecoder):
-    def __init__(self, model, max_batch, max_seq_len):
+    def __init__(self, model, max_batch, max_seq_len, ring_cap=64):
         super().__init__(model, max_batch, max_seq_len)
+        self.ring_cap = ring_cap
         self._init_cont()
 
 
 class ContinuousLlamaDecoder(_ContinuousMixin, CapturedLlamaDecoder):
-    def __init__(self, model, max_batch, max_seq_len):
+    def __init__(self, model, max_batch, max_seq_len, ring_cap=64):
         super().__init__(model, max_batch, max_seq_len)
+        self.ring_cap = ring_cap
         self._init_cont()
