@@ -37,4 +37,7 @@ train = ConfigDict(
     # torch.profiler window: dict(start_iter=10, end_iter=13[, with_stack])
     # -> chrome traces in <output_dir>/profiler/ (None = off)
     profiler=None,
+    # experimental fp8 e4m3 FORWARD GEMMs (bwd stays bf16); ~1.5-2x on the
+    # Linear1D matmuls (profiles/gemm_roofline.md)
+    fp8=dict(enabled=False),
 )

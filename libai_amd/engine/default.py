@@ -93,6 +93,14 @@ def default_setup(cfg, args=None):
 
     _check_batch_size(cfg)
 
+    fp8_cfg = try_get_key(cfg, "train.fp8", default=None)
+    if fp8_cfg and fp8_cfg.get("enabled", False):
+        from ..ops import fp8 as fp8_ops
+
+        fp8_ops.set_fp8_gemms(True)
+        logger.info("fp8 e4m3 forward GEMMs enabled (experimental; "
+                    "backward stays bf16)")
+
     if du.is_main_process():
         from ..config import LazyConfig
 

@@ -145,6 +145,13 @@ class Linear1D(nn.Module):
             else:
                 self.register_parameter("bias", None)
 
+    def _gemm(self, x, bias=None):
+        from ..ops import fp8
+
+        if fp8.fp8_eligible(x, self.weight):
+            return fp8.fp8_linear(x, self.weight, bias)
+        return F.linear(x, self.weight, bias)
+
     def forward(self, x):
         if self.parallel == "col":
             if self.sequence_parallel:
@@ -156,12 +163,12 @@ class Linear1D(nn.Module):
             else:
                 x = copy_to_tensor_parallel_region(x)
             if self.skip_bias_add:
-                return F.linear(x, self.weight), self.bias
+                return self._gemm(x), self.bias
             # bias fused into the GEMM epilogue (a separate add costs a full
             # HBM round-trip over the 3h-wide qkv activations)
-            return F.linear(x, self.weight, self.bias)
+            return self._gemm(x, self.bias)
         if self.parallel == "row":
-            out = F.linear(x, self.weight)
+            out = self._gemm(x)
             if self.sequence_parallel:
                 # SP: reduce the TP partial sums AND scatter back to shards
                 from ..parallel.comm import (
@@ -177,8 +184,8 @@ class Linear1D(nn.Module):
                 out = out + self.bias
             return out
         if self.skip_bias_add:
-            return F.linear(x, self.weight), self.bias
-        return F.linear(x, self.weight, self.bias)
+            return self._gemm(x), self.bias
+        return self._gemm(x, self.bias)
 
     def extra_repr(self):
         return (

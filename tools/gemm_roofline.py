@@ -42,6 +42,9 @@ def main():
     p.add_argument("--seq", type=int, default=1024)
     p.add_argument("--hidden", type=int, default=1024)
     p.add_argument("--vocab", type=int, default=50304)
+    p.add_argument("--fp8", action="store_true",
+                   help="also measure fp8 e4m3 via torch._scaled_mm "
+                        "(gfx950 fp8 dense peak ~5 PF/s)")
     args = p.parse_args()
 
     import bench as bench_mod  # noqa: F401  (loads the tunableop table)
@@ -74,6 +77,15 @@ def main():
             tf = gemm_tf(m, n, k, t)
             print(f"| {name} | {g} | {m}x{n}x{k} | {t * 1e3:.3f} | "
                   f"{tf:.0f} | {100 * tf / PEAK_TFLOPS:.1f}% |")
+        if args.fp8:
+            a8 = x.to(torch.float8_e4m3fn)
+            b8 = w.to(torch.float8_e4m3fn).t()
+            sc = torch.tensor(1.0, device="cuda")
+            t8 = bench(lambda: torch._scaled_mm(
+                a8, b8, scale_a=sc, scale_b=sc, out_dtype=torch.bfloat16))
+            tf8 = gemm_tf(m, n, k, t8)
+            print(f"| {name} | fwd FP8 e4m3 | {m}x{n}x{k} | {t8 * 1e3:.3f} | "
+                  f"{tf8:.0f} | {100 * tf8 / (2 * PEAK_TFLOPS):.1f}% of 5PF |")
         del x, w, dy
 
     # epilogue-fused MLP A/B (bias+gelu in the GEMM vs separate kernels)
