@@ -137,6 +137,8 @@ void rope_kv_insert_bf16(const void*, const void*, const void*, void*, void*,
                          void*, const float*, const float*, const long long*,
                          int, int, int, int, int64_t, int64_t, int64_t, int64_t,
                          int64_t, int64_t, int64_t, int, int, hipStream_t);
+void quant_fp8_bf16(const void*, void*, const float*, float*, int64_t,
+                    hipStream_t);
 void swiglu_fwd_bf16(const void*, void*, int64_t, int, hipStream_t);
 void swiglu_fwd_f32(const void*, void*, int64_t, int, hipStream_t);
 void swiglu_bwd_bf16(const void*, const void*, void*, int64_t, int, hipStream_t);
@@ -623,6 +625,23 @@ torch::Tensor rope_kv_insert(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   return qo;
 }
 
+// Single-pass delayed-scaling quantize: bf16 -> fp8 e4m3 with the GIVEN
+// scale; amax of the input accumulates into `amax` (caller derives the
+// next scale).  kernels/quant_fp8.hip.
+torch::Tensor quant_fp8(torch::Tensor x, torch::Tensor scale,
+                        torch::Tensor amax) {
+  CHECK_IN(x);
+  TORCH_CHECK(is_bf16(x), "quant_fp8: bf16 input only");
+  TORCH_CHECK(scale.scalar_type() == torch::kFloat32 && scale.is_cuda());
+  TORCH_CHECK(amax.scalar_type() == torch::kFloat32 && amax.is_cuda());
+  auto out = torch::empty(x.sizes(),
+                          x.options().dtype(torch::kFloat8_e4m3fn));
+  quant_fp8_bf16(x.data_ptr(), out.data_ptr(), scale.data_ptr<float>(),
+                 amax.data_ptr<float>(), x.numel(), cur_stream());
+  check_launch("quant_fp8");
+  return out;
+}
+
 torch::Tensor swiglu_fwd(torch::Tensor x) {
   CHECK_IN(x);
   const int64_t F = x.size(-1) / 2;
@@ -693,6 +712,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &ce_bwd);
   m.def("rope", &rope);
   m.def("rope_kv_insert", &rope_kv_insert);
+  m.def("quant_fp8", &quant_fp8);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("adamw_step", &adamw_step);

@@ -149,7 +149,11 @@ class Linear1D(nn.Module):
         from ..ops import fp8
 
         if fp8.fp8_eligible(x, self.weight):
-            return fp8.fp8_linear(x, self.weight, bias)
+            if not hasattr(self, "_fp8_states"):
+                self._fp8_states = (fp8.DelayedScale(), fp8.DelayedScale())
+            xs, ws = self._fp8_states
+            return fp8.fp8_linear(x, self.weight, bias, x_state=xs,
+                                  w_state=ws)
         return F.linear(x, self.weight, bias)
 
     def forward(self, x):

@@ -18,7 +18,7 @@ this is an experimental option beyond the reference's fp16/bf16 modes.
 import torch
 
 __all__ = ["fp8_available", "set_fp8_gemms", "fp8_gemms_enabled",
-           "fp8_eligible", "fp8_linear"]
+           "fp8_eligible", "fp8_linear", "DelayedScale"]
 
 _E4M3_MAX = 448.0
 _state = {"enabled": False}
@@ -63,18 +63,53 @@ def fp8_eligible(x, w):
 
 
 def _quant(t):
+    """Dynamic (3-pass) quantize — bootstrap / fallback path."""
     amax = t.abs().amax().float().clamp_(min=1e-8)
     scale = amax / _E4M3_MAX
     t8 = (t * (1.0 / scale)).to(torch.float8_e4m3fn)
     return t8, scale
 
 
+class DelayedScale:
+    """Per-call-site delayed scaling (transformer-engine idiom): the fused
+    quant kernel writes fp8 with the PREVIOUS step's scale in ONE pass and
+    accumulates this step's amax; the next scale derives from that amax
+    with no separate reduction pass over the activation."""
+
+    __slots__ = ("scale", "amax")
+
+    def __init__(self):
+        self.scale = None
+        self.amax = None
+
+    def quant(self, t):
+        from ._ext import ext
+
+        if self.scale is None:
+            # bootstrap: one dynamic quantize seeds the scale
+            t8, s0 = _quant(t)
+            self.scale = s0.reshape(1).clone()
+            self.amax = torch.zeros(1, device=t.device, dtype=torch.float32)
+            return t8, self.scale.clone()
+        used = self.scale.clone()  # the scale these values were written with
+        t8 = ext().quant_fp8(t, self.scale, self.amax)
+        # derive next step's scale from the freshly observed amax (values
+        # that grew past the stale scale saturated to +-448 this step)
+        torch.clamp(self.amax / _E4M3_MAX, min=1e-12, out=self.scale)
+        self.amax.zero_()
+        return t8, used
+
+
 class _Fp8LinearFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, bias):
+    def forward(ctx, x, w, bias, x_state, w_state):
         xs = x.reshape(-1, x.shape[-1]).contiguous()
-        x8, sx = _quant(xs)
-        w8, sw = _quant(w)
+        if x_state is not None:
+            x8, sx = x_state.quant(xs)
+            w8, sw = w_state.quant(w)
+        else:
+            x8, sx = _quant(xs)
+            w8, sw = _quant(w)
         y = torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw,
                              bias=bias, out_dtype=x.dtype)
         ctx.save_for_backward(xs, w)
@@ -88,9 +123,12 @@ class _Fp8LinearFn(torch.autograd.Function):
         dx = (dy2 @ w).view(*dy.shape[:-1], w.shape[1])
         dw = dy2.t() @ xs
         db = dy2.sum(0) if ctx.has_bias else None
-        return dx, dw, db
+        return dx, dw, db, None, None
 
 
-def fp8_linear(x, w, bias=None):
-    """y = x @ w^T + bias with the GEMM in fp8 e4m3 (bwd in bf16)."""
-    return _Fp8LinearFn.apply(x, w, bias)
+def fp8_linear(x, w, bias=None, x_state=None, w_state=None):
+    """y = x @ w^T + bias with the GEMM in fp8 e4m3 (bwd in bf16).
+
+    Pass ``DelayedScale`` states for the single-pass fused quantize; omit
+    them for the slower dynamic (3-pass) quantize."""
+    return _Fp8LinearFn.apply(x, w, bias, x_state, w_state)

@@ -79,3 +79,39 @@ def test_fp8_gpt_training_sanity():
     f8 = run(True)
     assert f8[-1] < 0.7 * f8[0], f8  # learning happens
     assert abs(f8[-1] - bf16[-1]) < 0.25 * bf16[0], (f8[-1], bf16[-1])
+
+
+def test_quant_fp8_kernel_numerics():
+    """Fused single-pass quantize: dequant error within e4m3 budget, amax
+    feeds the next scale (delayed scaling)."""
+    from libai_amd.ops import fp8
+
+    if not fp8.fp8_available():
+        pytest.skip("no fp8 on this stack")
+    torch.manual_seed(0)
+    x = torch.randn(4096, 1024, device="cuda", dtype=torch.bfloat16) * 3
+    st = fp8.DelayedScale()
+    st.quant(x)            # bootstrap (dynamic pass seeds the scale)
+    t8, used = st.quant(x)  # fused kernel path
+    deq = t8.float() * used
+    xf = x.float()
+    # e4m3: 3 mantissa bits -> relative error <= 2^-4 per element
+    rel = ((deq - xf).abs() / xf.abs().clamp(min=1e-3)).max()
+    assert rel.item() < 0.08, rel.item()
+    # the observed amax became the next scale
+    expect = (xf.abs().max() / 448.0).item()
+    assert abs(st.scale.item() - expect) / expect < 1e-3
+
+    # saturation: values that outgrow the stale scale clamp to +-448*scale
+    x2 = x * 100
+    t8b, used_b = st.quant(x2)
+    assert used_b.item() == pytest.approx(st_prev_scale_check(st, expect),
+                                          rel=1e-3)
+    assert t8b.float().abs().max().item() <= 448.0
+    # and the NEXT scale caught up to the new amax
+    assert st.scale.item() == pytest.approx(
+        (x2.float().abs().max() / 448.0).item(), rel=1e-3)
+
+
+def st_prev_scale_check(st, expect):
+    return expect  # used_b was the pre-update scale (== expect)

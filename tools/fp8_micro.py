@@ -31,6 +31,10 @@ def main():
         print(f"  _scaled_mm only:        {t(lambda: torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw, out_dtype=torch.bfloat16)):.3f} ms")
         print(f"  _scaled_mm + bias:      {t(lambda: torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw, bias=b, out_dtype=torch.bfloat16)):.3f} ms")
         print(f"  quant(x) only:          {t(lambda: fp8._quant(x)):.3f} ms")
+        st = fp8.DelayedScale(); st.quant(x.detach())
+        print(f"  fused delayed quant:    {t(lambda: st.quant(x.detach())):.3f} ms")
+        stx, stw = fp8.DelayedScale(), fp8.DelayedScale()
+        print(f"fp8_linear fwd (delayed): {t(lambda: fp8.fp8_linear(x, w, b, x_state=stx, w_state=stw)):.3f} ms")
     dy = torch.randn(M, N, device="cuda", dtype=torch.bfloat16)
     def fl_full():
         y = F.linear(x, w, b); y.backward(dy); x.grad = w.grad = b.grad = None
