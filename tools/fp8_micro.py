@@ -38,8 +38,10 @@ def main():
     dy = torch.randn(M, N, device="cuda", dtype=torch.bfloat16)
     def fl_full():
         y = F.linear(x, w, b); y.backward(dy); x.grad = w.grad = b.grad = None
+    stx2, stw2 = fp8.DelayedScale(), fp8.DelayedScale()
     def f8_full():
-        y = fp8.fp8_linear(x, w, b); y.backward(dy); x.grad = w.grad = b.grad = None
+        y = fp8.fp8_linear(x, w, b, x_state=stx2, w_state=stw2)
+        y.backward(dy); x.grad = w.grad = b.grad = None
     print(f"F.linear fwd+bwd:         {t(fl_full):.3f} ms")
     print(f"fp8_linear fwd+bwd:       {t(f8_full):.3f} ms")
     dy2 = dy
