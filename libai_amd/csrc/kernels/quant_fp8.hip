@@ -40,11 +40,13 @@ __global__ void quant_fp8_kernel(const uint16_t* __restrict__ x,
         float v2 = bf2f(xv[4 * q + 2]), v3 = bf2f(xv[4 * q + 3]);
         local = fmaxf(local, fmaxf(fmaxf(fabsf(v0), fabsf(v1)),
                                    fmaxf(fabsf(v2), fabsf(v3))));
-        // native packed cvt (v_cvt_pk_fp8_f32, saturating OCP e4m3):
-        // 2 elems/instruction vs the library's scalar bit-twiddling
+        // native packed cvt (v_cvt_pk_fp8_f32): 2 elems/instruction.
+        // The instruction does NOT saturate (overflow -> NaN), so clamp to
+        // +-448 first (measured: unclamped cvt NaN'd the training test)
+        auto sat = [&](float v) { return fminf(fmaxf(v * inv, -448.f), 448.f); };
         uint32_t r = 0;
-        r = __builtin_amdgcn_cvt_pk_fp8_f32(v0 * inv, v1 * inv, r, false);
-        r = __builtin_amdgcn_cvt_pk_fp8_f32(v2 * inv, v3 * inv, r, true);
+        r = __builtin_amdgcn_cvt_pk_fp8_f32(sat(v0), sat(v1), r, false);
+        r = __builtin_amdgcn_cvt_pk_fp8_f32(sat(v2), sat(v3), r, true);
         o[q] = r;
       }
       *reinterpret_cast<uint64_t*>(out + base) =
