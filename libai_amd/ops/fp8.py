@@ -13,6 +13,9 @@ Enable with ``train.fp8 = dict(enabled=True)`` (or
 ``libai_amd.ops.fp8.set_fp8_gemms(True)``); every Linear1D GEMM whose
 shapes divide 16 routes through here.  The HEADLINE bench stays bf16 —
 this is an experimental option beyond the reference's fp16/bf16 modes.
+Measured end-to-end at GPT-2 345M it is ~3% slower than bf16 (the GEMMs
+win, the per-site bookkeeping costs; profiles/gemm_roofline.md has the
+full iterated story) — it is aimed at larger hidden sizes.
 """
 
 import torch
