@@ -358,6 +358,41 @@ class _ContinuousMixin:
     def generate(self, *a, **k):
         raise RuntimeError("continuous decoder: use add_request/step/tokens")
 
+    @torch.no_grad()
+    def serve(self, prompts, max_new_tokens, eos_id=None, chunk=16):
+        """Drive a full request list through the slots: admit as slots
+        free up, harvest every ``chunk`` replays, trim at ``eos_id``.
+        Returns a list of 1-D token tensors aligned with ``prompts``."""
+        assert max_new_tokens <= self.ring_cap, "raise ring_cap"
+        results = [None] * len(prompts)
+        pending = list(range(len(prompts)))[::-1]
+        active = {}  # slot -> request index
+
+        def admit():
+            for slot in range(self.max_batch):
+                if slot not in active and pending:
+                    r = pending.pop()
+                    self.add_request(slot, prompts[r])
+                    active[slot] = r
+
+        admit()
+        while active:
+            self.step(chunk)
+            for slot, r in list(active.items()):
+                toks = self.tokens(slot)
+                done = toks.numel() >= max_new_tokens
+                if eos_id is not None:
+                    hit = (toks == eos_id).nonzero()
+                    if hit.numel():
+                        toks = toks[: int(hit[0, 0]) + 1]
+                        done = True
+                if done or int(self.pos[slot]) + chunk + 2 >= self.max_seq_len:
+                    results[r] = toks[:max_new_tokens]
+                    self.release(slot)
+                    del active[slot]
+            admit()
+        return results
+
 
 class ContinuousGPTDecoder(_ContinuousMixin, CapturedGPTDecoder):
     def __init__(self, model, max_batch, max_seq_len, ring_cap=64):

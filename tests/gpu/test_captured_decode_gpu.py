@@ -207,3 +207,25 @@ def test_continuous_batching_llama():
     dec.step(12)
     assert torch.equal(dec.tokens(0), _eager_greedy(model, p0, 13)[0])
     assert torch.equal(dec.tokens(2), _eager_greedy(model, p1, 13)[0])
+
+
+def test_continuous_serve_driver():
+    """serve(): more requests than slots, chunked harvest, EOS trim."""
+    from libai_amd.inference.captured_decode import ContinuousGPTDecoder
+
+    model = _tiny_gpt()
+    torch.manual_seed(3)
+    prompts = [torch.randint(0, 1024, (int(l),), device="cuda")
+               for l in (12, 30, 7, 21, 16, 9, 25)]
+    dec = ContinuousGPTDecoder(model, max_batch=3, max_seq_len=256)
+    outs = dec.serve(prompts, max_new_tokens=10, chunk=4)
+    assert len(outs) == 7
+    for p, o in zip(prompts, outs):
+        assert o.shape == (10,)
+        ref = _eager_greedy(model, p.view(1, -1), 10)[0]
+        assert torch.equal(o, ref), (p.shape, o, ref)
+
+    # EOS: use the known first generated token of prompt 0 -> 1-token result
+    eos = int(outs[0][0])
+    outs2 = dec.serve(prompts[:1], max_new_tokens=10, eos_id=eos, chunk=4)
+    assert outs2[0].numel() >= 1 and int(outs2[0][-1]) == eos
