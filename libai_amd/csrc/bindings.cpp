@@ -121,6 +121,12 @@ void flash_decode_bf16(const void*, const void*, const void*, void*, const int*,
                        int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
                        int64_t, int64_t, int64_t, int64_t, int, int, int, int,
                        float, int, hipStream_t);
+int flash_decode_num_splits(int, int, int);
+void flash_decode_split_bf16(const void*, const void*, const void*, void*,
+                             float*, float*, float*, const int*, int64_t,
+                             int64_t, int64_t, int64_t, int64_t, int64_t,
+                             int64_t, int64_t, int64_t, int64_t, int, int, int,
+                             int, int, float, int, hipStream_t);
 void rope_fwd_bf16(const void*, void*, const float*, const float*, int64_t, int64_t,
                    int64_t, int64_t, int64_t, int64_t, int, int, int, int, int,
                    hipStream_t);
@@ -425,11 +431,28 @@ torch::Tensor flash_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
     kvl = kv_len->data_ptr<int>();
   }
   auto o = torch::empty({B, H, 1, D}, q.options());
-  flash_decode_bf16(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(), kvl,
-                    q.stride(0), q.stride(1), k.stride(0), k.stride(2),
-                    k.stride(1), v.stride(0), v.stride(2), v.stride(1),
-                    o.stride(0), o.stride(1), B, H, Skv, D, (float)scale,
-                    H / Hkv, cur_stream());
+  const int S = flash_decode_num_splits(B, H, Skv);
+  if (S > 1) {
+    // split-KV (flash-decoding): B*H WGs underfill 256 CUs at serving
+    // batch sizes; fixed 512-key splits keep outputs bitwise-independent
+    // of the cache capacity (see kernels/flash_decode.hip)
+    auto fopt = q.options().dtype(torch::kFloat32);
+    auto pm = torch::empty({(int64_t)B * H * S}, fopt);
+    auto pl = torch::empty({(int64_t)B * H * S}, fopt);
+    auto pa = torch::empty({(int64_t)B * H * S * D}, fopt);
+    flash_decode_split_bf16(
+        q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+        pm.data_ptr<float>(), pl.data_ptr<float>(), pa.data_ptr<float>(), kvl,
+        q.stride(0), q.stride(1), k.stride(0), k.stride(2), k.stride(1),
+        v.stride(0), v.stride(2), v.stride(1), o.stride(0), o.stride(1), B, H,
+        S, Skv, D, (float)scale, H / Hkv, cur_stream());
+  } else {
+    flash_decode_bf16(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                      kvl, q.stride(0), q.stride(1), k.stride(0), k.stride(2),
+                      k.stride(1), v.stride(0), v.stride(2), v.stride(1),
+                      o.stride(0), o.stride(1), B, H, Skv, D, (float)scale,
+                      H / Hkv, cur_stream());
+  }
   check_launch("flash_decode");
   return o;
 }

@@ -320,3 +320,48 @@ def test_model_decode_uses_fused_kernel_and_matches_full():
                use_cache=True)
     err = (st["prediction_scores"][:, 0].float() - full[:, 32].float()).abs().max()
     assert err.item() < 0.1, err.item()  # bf16 cache round-trips
+
+
+def test_flash_decode_split_kv_capacity_independent():
+    """Split-KV decode (fixed 512-key splits): outputs are bitwise
+    IDENTICAL whether the cache capacity selects the single-WG kernel
+    (<=512) or the split path (>512) — the property the captured-vs-eager
+    greedy parity rests on — and match the fp32 reference."""
+    from libai_amd.ops.attention import flash_decode_attn
+
+    torch.manual_seed(0)
+    B, H, D, L = 4, 16, 64, 450
+    q = torch.randn(B, H, 1, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, H, L, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn_like(k)
+    kvl = torch.full((B,), L, dtype=torch.int32, device="cuda")
+    scale = D ** -0.5
+
+    def with_capacity(cap):
+        kc = torch.randn(B, H, cap, D, device="cuda", dtype=torch.bfloat16)
+        vc = torch.randn_like(kc)
+        kc[:, :, :L].copy_(k)
+        vc[:, :, :L].copy_(v)
+        return flash_decode_attn(q, kc, vc, scale, kv_len=kvl)
+
+    o_small = with_capacity(512)    # single-WG kernel
+    o_big = with_capacity(1536)     # split path (S=3), 2 null splits
+    assert torch.equal(o_small, o_big)
+
+    ref = torch.softmax(
+        (q.float() @ k.float().transpose(-1, -2)) * scale, dim=-1
+    ) @ v.float()
+    assert (o_big.float() - ref).abs().max().item() < 2e-2
+
+    # long cache: several REAL splits vs reference
+    B2, L2, cap2 = 2, 1300, 1536
+    q2 = torch.randn(B2, H, 1, D, device="cuda", dtype=torch.bfloat16)
+    k2 = torch.randn(B2, H, cap2, D, device="cuda", dtype=torch.bfloat16)
+    v2 = torch.randn_like(k2)
+    kvl2 = torch.tensor([L2, 700], dtype=torch.int32, device="cuda")
+    o2 = flash_decode_attn(q2, k2, v2, scale, kv_len=kvl2)
+    for bi, n in enumerate((L2, 700)):
+        r = torch.softmax(
+            (q2[bi].float() @ k2[bi, :, :n].float().transpose(-1, -2)) * scale,
+            dim=-1) @ v2[bi, :, :n].float()
+        assert (o2[bi].float() - r).abs().max().item() < 2e-2
