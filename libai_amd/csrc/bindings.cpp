@@ -145,6 +145,9 @@ void rope_kv_insert_bf16(const void*, const void*, const void*, void*, void*,
                          int64_t, int64_t, int64_t, int, int, hipStream_t);
 void quant_fp8_bf16(const void*, void*, const float*, float*, int64_t,
                     hipStream_t);
+void res_ln_fwd_bf16(const void*, const void*, const void*, const void*,
+                     const void*, void*, void*, int64_t, int, float, bool,
+                     hipStream_t);
 void swiglu_fwd_bf16(const void*, void*, int64_t, int, hipStream_t);
 void swiglu_fwd_f32(const void*, void*, int64_t, int, hipStream_t);
 void swiglu_bwd_bf16(const void*, const void*, void*, int64_t, int, hipStream_t);
@@ -651,6 +654,30 @@ torch::Tensor rope_kv_insert(torch::Tensor q, torch::Tensor k, torch::Tensor v,
 // Single-pass delayed-scaling quantize: bf16 -> fp8 e4m3 with the GIVEN
 // scale; amax of the input accumulates into `amax` (caller derives the
 // next scale).  kernels/quant_fp8.hip.
+// Fused residual-add + norm (decode/eval forward only):
+// h = x (+bias) + res, y = norm(h).  Wave-per-row, H <= 2048 * vec.
+std::vector<torch::Tensor> res_norm_fwd(torch::Tensor x,
+                                        c10::optional<torch::Tensor> bias,
+                                        torch::Tensor res, torch::Tensor gamma,
+                                        c10::optional<torch::Tensor> beta,
+                                        double eps, bool rms) {
+  CHECK_IN(x);
+  CHECK_IN(res);
+  TORCH_CHECK(is_bf16(x), "res_norm_fwd: bf16 only");
+  const int H = (int)x.size(-1);
+  TORCH_CHECK(H % 8 == 0 && H / 8 <= 256, "H must be vec8 and <= 2048");
+  const int64_t R = x.numel() / H;
+  auto h = torch::empty_like(x);
+  auto y = torch::empty_like(x);
+  const void* bp = bias.has_value() ? bias->contiguous().data_ptr() : nullptr;
+  const void* btp = beta.has_value() ? beta->contiguous().data_ptr() : nullptr;
+  res_ln_fwd_bf16(x.data_ptr(), bp, res.contiguous().data_ptr(),
+                  gamma.contiguous().data_ptr(), btp, h.data_ptr(),
+                  y.data_ptr(), R, H, (float)eps, rms, cur_stream());
+  check_launch("res_norm_fwd");
+  return {h, y};
+}
+
 torch::Tensor quant_fp8(torch::Tensor x, torch::Tensor scale,
                         torch::Tensor amax) {
   CHECK_IN(x);
@@ -736,6 +763,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope", &rope);
   m.def("rope_kv_insert", &rope_kv_insert);
   m.def("quant_fp8", &quant_fp8);
+  m.def("res_norm_fwd", &res_norm_fwd);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("adamw_step", &adamw_step);

@@ -147,6 +147,85 @@ __global__ void norm_fwd_wave_kernel(const typename E::T* __restrict__ x,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused residual-add + norm (decode/eval path): h = x + bias + res is
+// written out (the next residual) AND normalized in the same registers —
+// replaces the bias_dropout_res + norm_fwd kernel pair per layer in the
+// hipGraph-captured serving step (both latency-bound at decode batch
+// sizes).  Forward-only: the captured decode never needs its backward.
+// ---------------------------------------------------------------------------
+template <class E, bool RMS, int NV>
+__global__ void res_norm_fwd_wave_kernel(
+    const typename E::T* __restrict__ x, const typename E::T* __restrict__ bias,
+    const typename E::T* __restrict__ res, const typename E::T* __restrict__ gamma,
+    const typename E::T* __restrict__ beta, typename E::T* __restrict__ h,
+    typename E::T* __restrict__ y, int H, float eps, int64_t R) {
+  using T = typename E::T;
+  using VecT = typename E::VecT;
+  constexpr int V = E::VEC;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int64_t row = (int64_t)blockIdx.x * (blockDim.x >> 6) + wave;
+  if (row >= R) return;
+  const T* xr = x + row * (int64_t)H;
+  const T* rr = res + row * (int64_t)H;
+  T* hr = h + row * (int64_t)H;
+  T* yr = y + row * (int64_t)H;
+  const int nvec = H / V;
+
+  VecT vh[NV];
+  float sum = 0.f, sumsq = 0.f;
+#pragma unroll
+  for (int n = 0; n < NV; ++n) {
+    const int i = lane + n * 64;
+    if (i < nvec) {
+      VecT vx = ((const VecT*)xr)[i];
+      VecT vr = ((const VecT*)rr)[i];
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        // association matches the bias_dropout_res kernel exactly
+        // ((x + bias) + res) so the eager and captured decode paths
+        // produce bitwise-identical residual streams
+        float f = E::to_f(vx[j]);
+        if (bias != nullptr) f += E::to_f(((const VecT*)bias)[i][j]);
+        f += E::to_f(vr[j]);
+        vh[n][j] = E::from_f(f);
+        const float g = E::to_f(vh[n][j]);  // stats on the STORED value
+        sum += g;
+        sumsq += g * g;
+      }
+      ((VecT*)hr)[i] = vh[n];
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    sum += __shfl_xor(sum, off, 64);
+    sumsq += __shfl_xor(sumsq, off, 64);
+  }
+  const float mean = RMS ? 0.f : sum / H;
+  const float rstd = rsqrtf(sumsq / H - mean * mean + eps);
+#pragma unroll
+  for (int n = 0; n < NV; ++n) {
+    const int i = lane + n * 64;
+    if (i < nvec) {
+      VecT g = ((const VecT*)gamma)[i];
+      VecT o;
+      if (!RMS && beta != nullptr) {
+        VecT b = ((const VecT*)beta)[i];
+#pragma unroll
+        for (int j = 0; j < V; ++j)
+          o[j] = E::from_f((E::to_f(vh[n][j]) - mean) * rstd * E::to_f(g[j]) +
+                           E::to_f(b[j]));
+      } else {
+#pragma unroll
+        for (int j = 0; j < V; ++j)
+          o[j] = E::from_f((E::to_f(vh[n][j]) - mean) * rstd * E::to_f(g[j]));
+      }
+      ((VecT*)yr)[i] = o;
+    }
+  }
+}
+
 template <class E, bool RMS, int NV>
 __global__ void norm_bwd_dx_wave_kernel(const typename E::T* __restrict__ dy,
                                         const typename E::T* __restrict__ x,
@@ -366,6 +445,27 @@ void launch_norm_fwd(const void* x, const void* gamma, const void* beta, void* y
 }
 
 template <class E, bool RMS>
+void launch_res_norm_fwd(const void* x, const void* bias, const void* res,
+                         const void* gamma, const void* beta, void* h, void* y,
+                         int64_t R, int H, float eps, hipStream_t stream) {
+  const int nvl = CDIV(H / E::VEC, 64);
+  dim3 g((uint32_t)CDIV(R, 4));
+#define RES_FWD_WAVE(NV)                                                      \
+  res_norm_fwd_wave_kernel<E, RMS, NV><<<g, dim3(256), 0, stream>>>(          \
+      (const typename E::T*)x, (const typename E::T*)bias,                    \
+      (const typename E::T*)res, (const typename E::T*)gamma,                 \
+      (const typename E::T*)beta, (typename E::T*)h, (typename E::T*)y, H,    \
+      eps, R)
+  if (nvl <= 1)
+    RES_FWD_WAVE(1);
+  else if (nvl == 2)
+    RES_FWD_WAVE(2);
+  else
+    RES_FWD_WAVE(4);
+#undef RES_FWD_WAVE
+}
+
+template <class E, bool RMS>
 void launch_norm_bwd_dx(const void* dy, const void* x, const void* gamma,
                         const float* mean, const float* rstd, void* dx, int64_t R,
                         int H, hipStream_t stream, int th) {
@@ -454,3 +554,16 @@ inline int row_block_threads(int H, int vec) {
 
 NORM_LAUNCHERS(bf16, BF16Elem)
 NORM_LAUNCHERS(f32, F32Elem)
+
+extern "C" void res_ln_fwd_bf16(const void* x, const void* bias,
+                                const void* res, const void* gamma,
+                                const void* beta, void* h, void* y, int64_t R,
+                                int H, float eps, bool rms,
+                                hipStream_t stream) {
+  if (rms)
+    launch_res_norm_fwd<BF16Elem, true>(x, bias, res, gamma, beta, h, y, R, H,
+                                        eps, stream);
+  else
+    launch_res_norm_fwd<BF16Elem, false>(x, bias, res, gamma, beta, h, y, R, H,
+                                         eps, stream);
+}

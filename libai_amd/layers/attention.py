@@ -154,9 +154,8 @@ class MultiheadAttention(nn.Module):
             scale_c = self.norm_factor * (self.coeff if self.coeff else 1.0)
             ctx = flash_decode_attn(qo, ck, cv, scale_c, kv_len=kv_len32)
             context = ctx.permute(0, 2, 1, 3).reshape(b, 1, nh * hs)
-            out, bias = self.dense(context)
-            return bias_dropout_add(out, bias=bias, residual=residual,
-                                    p=0.0, training=False)
+            # returns RAW (out, bias): the layer fuses bias+residual+norm
+            return self.dense(context)
 
         # fused flash path: consumes the qkv buffer with ZERO copies (strided
         # [b, s, nh, hs] views), O lands directly in [b, s, h] layout.

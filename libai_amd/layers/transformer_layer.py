@@ -140,12 +140,18 @@ class TransformerLayer(nn.Module):
     ):
         if static_cache is not None:
             # hipGraph-capturable decode step: fixed-shape path, no cache
-            # tuples returned (KV written in-place into the static buffers)
+            # tuples returned (KV written in-place into the static buffers);
+            # bias + residual + post-attention norm run as ONE kernel
+            from ..ops._ext import ext
+
             ln1 = self.input_layernorm(hidden_states)
             residual = ln1 if self.apply_residual_post_layernorm else hidden_states
-            h = self.self_attention(ln1, residual=residual,
-                                    static_cache=static_cache, position=position)
-            ln2 = self.post_attention_layernorm(h)
+            out, bias = self.self_attention(ln1, residual=residual,
+                                            static_cache=static_cache,
+                                            position=position)
+            ln = self.post_attention_layernorm
+            h, ln2 = ext().res_norm_fwd(out, bias, residual, ln.weight,
+                                        ln.bias, ln.eps, False)
             residual = ln2 if self.apply_residual_post_layernorm else h
             return self.mlp(ln2, residual=residual)
         if past_key_value is not None:

@@ -110,7 +110,7 @@ class LlamaAttention(nn.Module):
             context = ctx.permute(0, 2, 1, 3).reshape(
                 b, 1, self.num_heads_local * self.head_dim)
             out, _ = self.o_proj(context)
-            return out + residual if residual is not None else out
+            return out  # RAW: the layer fuses residual + RMSNorm
         if self.sequence_parallel and (use_cache or past_key_value is not None):
             raise RuntimeError("sequence_parallel is training-only; build the "
                                "generation model with sequence_parallel=False")
@@ -232,10 +232,15 @@ class LlamaDecoderLayer(nn.Module):
     def forward(self, hidden_states, past_key_value=None, use_cache=False,
                 static_cache=None, position=None):
         if static_cache is not None:
+            # residual + post-attention RMSNorm fused into one kernel
+            from ..ops._ext import ext
+
             ln1 = self.input_layernorm(hidden_states)
-            h = self.self_attn(ln1, residual=hidden_states,
-                               static_cache=static_cache, position=position)
-            ln2 = self.post_attention_layernorm(h)
+            a = self.self_attn(ln1, static_cache=static_cache,
+                               position=position)
+            ln = self.post_attention_layernorm
+            h, ln2 = ext().res_norm_fwd(a, None, hidden_states, ln.weight,
+                                        None, ln.eps, True)
             return self.mlp(ln2, residual=h)
         ln1 = self.input_layernorm(hidden_states)
         attn_out = self.self_attn(ln1, past_key_value=past_key_value,

@@ -3,6 +3,11 @@ precision: forward matmuls run on the gfx950 fp8 MFMA pipe (dense peak
 ~5 PFLOP/s, 2x bf16; measured 1.5-2x at the bench shapes,
 profiles/gemm_roofline.md), backward stays bf16.
 
+Note: under activation checkpointing the recompute pass quantizes again
+with the (slightly advanced) delayed scale, so recomputed activations can
+differ from the original forward by one scale step — bounded and standard
+for delayed-scaling fp8, but part of why this stays experimental.
+
 Per-tensor DYNAMIC scaling: each forward quantizes activations and weights
 with amax/448 scales and lets hipBLASLt dequantize in the epilogue
 (`torch._scaled_mm`).  The backward uses the saved bf16 tensors, so

@@ -425,3 +425,31 @@ def test_rope_kv_insert_matches_reference():
     assert torch.equal(qo2, q.permute(0, 2, 1, 3))
     assert torch.equal(ck[:, :, pos_i], k.squeeze(1))
     assert torch.equal(cv[:, :, pos_i], v.squeeze(1))
+
+
+def test_res_norm_fwd_matches_separate_ops():
+    """Fused residual+norm == the bias_dropout_res(p=0) + norm pair it
+    replaces in the captured decode step, BITWISE (same fp32 association)."""
+    from libai_amd.ops._ext import ext
+
+    torch.manual_seed(0)
+    R, H = 32, 1024
+    x = torch.randn(R, 1, H, device="cuda", dtype=torch.bfloat16)
+    bias = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    res = torch.randn_like(x)
+    g = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    b2 = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+
+    h_ref = ext().bias_dropout_res_fwd(x.contiguous(), bias, res.contiguous(),
+                                       0.0, 0)
+    y_ref, _, _ = ext().ln_fwd(h_ref, g, b2, False, 1e-5)
+    h, y = ext().res_norm_fwd(x, bias, res, g, b2, 1e-5, False)
+    assert torch.equal(h, h_ref)
+    assert torch.equal(y, y_ref)
+
+    # RMS variant, no bias (the Llama decode shape)
+    h_ref2 = x + res
+    y_ref2, _, _ = ext().ln_fwd(h_ref2.contiguous(), g, None, True, 1e-6)
+    h2, y2 = ext().res_norm_fwd(x, None, res, g, None, 1e-6, True)
+    assert torch.equal(h2, h_ref2)
+    assert torch.equal(y2, y_ref2)
