@@ -143,6 +143,7 @@ class TransformerLayer(nn.Module):
             # tuples returned (KV written in-place into the static buffers);
             # bias + residual + post-attention norm run as ONE kernel
             from ..ops._ext import ext
+            from ..ops.fused_bias import bias_dropout_add
 
             ln1 = self.input_layernorm(hidden_states)
             residual = ln1 if self.apply_residual_post_layernorm else hidden_states
@@ -150,8 +151,13 @@ class TransformerLayer(nn.Module):
                                             static_cache=static_cache,
                                             position=position)
             ln = self.post_attention_layernorm
-            h, ln2 = ext().res_norm_fwd(out, bias, residual, ln.weight,
-                                        ln.bias, ln.eps, False)
+            if out.shape[-1] <= 2048:
+                h, ln2 = ext().res_norm_fwd(out, bias, residual, ln.weight,
+                                            ln.bias, ln.eps, False)
+            else:  # wide rows exceed the wave kernel's register budget
+                h = bias_dropout_add(out, bias=bias, residual=residual,
+                                     p=0.0, training=False)
+                ln2 = ln(h)
             residual = ln2 if self.apply_residual_post_layernorm else h
             return self.mlp(ln2, residual=residual)
         if past_key_value is not None:

@@ -239,8 +239,12 @@ class LlamaDecoderLayer(nn.Module):
             a = self.self_attn(ln1, static_cache=static_cache,
                                position=position)
             ln = self.post_attention_layernorm
-            h, ln2 = ext().res_norm_fwd(a, None, hidden_states, ln.weight,
-                                        None, ln.eps, True)
+            if a.shape[-1] <= 2048:
+                h, ln2 = ext().res_norm_fwd(a, None, hidden_states, ln.weight,
+                                            None, ln.eps, True)
+            else:  # wide rows (e.g. 7B h4096): separate add + RMSNorm
+                h = a + hidden_states
+                ln2 = ln(h)
             return self.mlp(ln2, residual=h)
         ln1 = self.input_layernorm(hidden_states)
         attn_out = self.self_attn(ln1, past_key_value=past_key_value,

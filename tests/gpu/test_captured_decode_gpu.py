@@ -229,3 +229,22 @@ def test_continuous_serve_driver():
     eos = int(outs[0][0])
     outs2 = dec.serve(prompts[:1], max_new_tokens=10, eos_id=eos, chunk=4)
     assert outs2[0].numel() >= 1 and int(outs2[0][-1]) == eos
+
+
+def test_captured_llama_wide_hidden():
+    """h > 2048 takes the separate add+norm fallback (the 7B shape class);
+    parity must still hold."""
+    from libai_amd.inference.captured_decode import CapturedLlamaDecoder
+    from libai_amd.models.llama import LlamaForCausalLM
+
+    torch.manual_seed(2)
+    m = LlamaForCausalLM(
+        hidden_layers=2, vocab_size=512, hidden_size=2560,
+        intermediate_size=1024, num_attention_heads=20,
+        num_key_value_heads=4, max_position_embeddings=128,
+    ).to("cuda", torch.bfloat16).eval()
+    b, L, n_new = 2, 16, 12
+    prompt = torch.randint(0, 512, (b, L), device="cuda")
+    ref = _eager_greedy(m, prompt, n_new)
+    dec = CapturedLlamaDecoder(m, max_batch=b, max_seq_len=128)
+    assert torch.equal(dec.generate(prompt, n_new), ref)
