@@ -66,9 +66,13 @@ class ZeRO3Manager:
 
         dutil = du.get_dist_util()
         if dutil.pipeline_parallel_size > 1:
+            # parity note: the reference exercises stage 3 only at dp4/tp1/
+            # pp1 (reference tests/models/test_gpt.py:185-198); PP already
+            # partitions parameters, so stage 1/2 is the composition that
+            # makes sense with PP here too
             raise NotImplementedError(
-                "ZeRO stage 3 is not composed with pipeline parallelism yet; "
-                "use stage 1/2 with PP"
+                "ZeRO stage 3 is not composed with pipeline parallelism "
+                "(PP already partitions parameters); use stage 1/2 with PP"
             )
         assert optimizer._buckets is None, (
             "setup_zero3 must run before the optimizer builds its buckets "
