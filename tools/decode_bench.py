@@ -21,7 +21,8 @@ def main():
     p.add_argument("--batch", type=int, default=32)
     p.add_argument("--ctx", type=int, default=1024)
     p.add_argument("--steps", type=int, default=64)
-    p.add_argument("--model", default="gpt2", choices=["gpt2", "llama1b"])
+    p.add_argument("--model", default="gpt2",
+                   choices=["gpt2", "llama1b", "llama7b"])
     args = p.parse_args()
 
     import bench as bench_mod
@@ -33,13 +34,18 @@ def main():
 
     du.setup_dist_util({})
     torch.manual_seed(0)
-    if args.model == "llama1b":
+    if args.model in ("llama1b", "llama7b"):
         from libai_amd.models import LlamaForCausalLM
 
+        shape = (dict(hidden_layers=16, hidden_size=2048,
+                      intermediate_size=5504, num_attention_heads=16,
+                      num_key_value_heads=4)
+                 if args.model == "llama1b" else
+                 dict(hidden_layers=32, hidden_size=4096,
+                      intermediate_size=11008, num_attention_heads=32,
+                      num_key_value_heads=32))
         m = LlamaForCausalLM(
-            hidden_layers=16, vocab_size=32000, hidden_size=2048,
-            intermediate_size=5504, num_attention_heads=16,
-            num_key_value_heads=4, max_position_embeddings=4096,
+            vocab_size=32000, max_position_embeddings=4096, **shape,
         ).to(torch.bfloat16).cuda().eval()
         vocab = 32000
     else:
