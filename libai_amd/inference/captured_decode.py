@@ -33,7 +33,8 @@ import torch
 
 __all__ = ["CapturedGPTDecoder", "CapturedLlamaDecoder",
            "SamplingMixin", "CapturedGPTSampler", "CapturedLlamaSampler",
-           "ContinuousGPTDecoder", "ContinuousLlamaDecoder"]
+           "ContinuousGPTDecoder", "ContinuousLlamaDecoder",
+           "ContinuousGPTSampler", "ContinuousLlamaSampler"]
 
 
 class _CapturedDecoderBase:
@@ -406,3 +407,20 @@ class ContinuousLlamaDecoder(_ContinuousMixin, CapturedLlamaDecoder):
         super().__init__(model, max_batch, max_seq_len)
         self.ring_cap = ring_cap
         self._init_cont()
+
+
+class ContinuousGPTSampler(SamplingMixin, ContinuousGPTDecoder):
+    # continuous batching with in-graph top-k/temperature sampling
+    def __init__(self, model, max_batch, max_seq_len, ring_cap=64,
+                 temperature=1.0, top_k=0):
+        super().__init__(model, max_batch, max_seq_len, ring_cap=ring_cap)
+        self.temperature = temperature
+        self.top_k = top_k
+
+
+class ContinuousLlamaSampler(SamplingMixin, ContinuousLlamaDecoder):
+    def __init__(self, model, max_batch, max_seq_len, ring_cap=64,
+                 temperature=1.0, top_k=0):
+        super().__init__(model, max_batch, max_seq_len, ring_cap=ring_cap)
+        self.temperature = temperature
+        self.top_k = top_k

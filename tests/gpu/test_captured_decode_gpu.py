@@ -248,3 +248,18 @@ def test_captured_llama_wide_hidden():
     ref = _eager_greedy(m, prompt, n_new)
     dec = CapturedLlamaDecoder(m, max_batch=b, max_seq_len=128)
     assert torch.equal(dec.generate(prompt, n_new), ref)
+
+
+def test_continuous_sampler_smoke():
+    """Continuous batching + in-graph top-k sampling composes."""
+    from libai_amd.inference.captured_decode import ContinuousGPTSampler
+
+    model = _tiny_gpt()
+    dec = ContinuousGPTSampler(model, max_batch=2, max_seq_len=128,
+                               temperature=0.9, top_k=16)
+    p0 = torch.randint(0, 1024, (14,), device="cuda")
+    p1 = torch.randint(0, 1024, (23,), device="cuda")
+    outs = dec.serve([p0, p1], max_new_tokens=12, chunk=4)
+    for o in outs:
+        assert o.shape == (12,)
+        assert int(o.min()) >= 0 and int(o.max()) < 1024
