@@ -55,9 +55,10 @@ def _default_units(model):
         return list(model.zero3_unit_modules())
     from ..layers.transformer_layer import TransformerLayer
     from ..models.llama import LlamaDecoderLayer
+    from ..models.palm import PaLMBlock
 
     return [m for m in model.modules()
-            if isinstance(m, (TransformerLayer, LlamaDecoderLayer))]
+            if isinstance(m, (TransformerLayer, LlamaDecoderLayer, PaLMBlock))]
 
 
 class ZeRO3Manager:
